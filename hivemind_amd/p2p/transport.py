@@ -1,0 +1,490 @@
+"""Asyncio TCP transport: persistent multiplexed peer connections with
+unary and bidirectional-streaming RPC.
+
+This replaces the reference's out-of-tree Go libp2p daemon (``p2pd``) and its
+unix-socket control protocol (``hivemind/p2p/p2p_daemon.py``,
+``p2p_daemon_bindings/control.py``). Design differences, on purpose:
+
+* One asyncio TCP server per ``P2P`` instance, run on the caller's event loop
+  (no subprocess): tensor bytes never travel through this layer on the GPU
+  node -- RCCL over xGMI moves them -- so the control plane stays simple.
+* Persistent connections are multiplexed by 8-byte call ids; both directions
+  of a single TCP connection carry calls (like libp2p stream reuse).
+* Frames are length-prefixed msgpack envelopes; payloads are raw bytes.
+
+Public surface mirrors the reference semantics: ``P2P.create``,
+``add_unary_handler`` / ``add_stream_handler`` (reference
+``add_protobuf_handler`` with stream flags), ``call_unary`` /
+``call_stream`` (reference ``call_protobuf_handler`` /
+``iterate_protobuf_handler``), ``P2PHandlerError``, ``P2PDaemonError``.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import struct
+from typing import AsyncIterator, Awaitable, Callable, Dict, List, Optional, Tuple, Union
+
+from ..utils.crypto import PrivateKey
+from ..utils.logging import get_logger
+from ..utils.networking import LOCALHOST, make_endpoint, split_endpoint
+from ..utils.serializer import MSGPackSerializer
+from .peer_id import PeerID, PeerInfo
+
+logger = get_logger(__name__)
+
+# frame types
+_T_HELLO = 0
+_T_REQUEST = 1  # unary request
+_T_RESPONSE = 2  # unary response
+_T_STREAM_OPEN = 3  # open a streaming call
+_T_STREAM_ITEM = 4  # item caller->callee or callee->caller
+_T_STREAM_HALF_CLOSE = 5  # caller finished sending
+_T_STREAM_END = 6  # callee finished (stream done)
+_T_ERROR = 7
+_T_CANCEL = 8
+_T_PING = 9
+
+MAX_FRAME_SIZE = 256 * 1024 * 1024  # control plane sanity bound
+STREAM_CHUNK_SIZE = 512 * 1024  # aligns with averaging part size
+
+HandlerType = Union[
+    Callable[[bytes, "RpcContext"], Awaitable[bytes]],
+    Callable[[AsyncIterator[bytes], "RpcContext"], AsyncIterator[bytes]],
+]
+
+
+class P2PHandlerError(Exception):
+    """Raised on the caller side when the remote handler raised."""
+
+
+class P2PDaemonError(Exception):
+    """Transport-level failure (peer unreachable, connection lost)."""
+
+
+class RpcContext:
+    """Information about the remote side of an in-flight RPC."""
+
+    __slots__ = ("remote_id", "local_id")
+
+    def __init__(self, remote_id: PeerID, local_id: PeerID):
+        self.remote_id = remote_id
+        self.local_id = local_id
+
+
+class _Connection:
+    """One TCP connection to a peer; carries multiplexed calls in both directions."""
+
+    def __init__(self, p2p: "P2P", reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        self.p2p = p2p
+        self.reader = reader
+        self.writer = writer
+        self.remote_id: Optional[PeerID] = None
+        self.send_lock = asyncio.Lock()
+        # caller-side state for calls we initiated over this connection
+        self.pending_unary: Dict[bytes, asyncio.Future] = {}
+        self.inbound_streams: Dict[bytes, asyncio.Queue] = {}  # call_id -> queue of items/end
+        # callee-side state for calls the remote initiated
+        self.serving_streams: Dict[bytes, asyncio.Queue] = {}
+        self.serving_tasks: Dict[bytes, asyncio.Task] = {}
+        self.reader_task: Optional[asyncio.Task] = None
+        self.closed = asyncio.Event()
+
+    async def send_frame(self, ftype: int, call_id: bytes, handler: str = "", payload: bytes = b""):
+        env = MSGPackSerializer.dumps((ftype, call_id, handler, payload))
+        async with self.send_lock:
+            self.writer.write(struct.pack(">I", len(env)) + env)
+            await self.writer.drain()
+
+    async def recv_frame(self) -> Tuple[int, bytes, str, bytes]:
+        header = await self.reader.readexactly(4)
+        (length,) = struct.unpack(">I", header)
+        if length > MAX_FRAME_SIZE:
+            raise P2PDaemonError(f"frame too large: {length}")
+        env = await self.reader.readexactly(length)
+        ftype, call_id, handler, payload = MSGPackSerializer.loads(env)
+        return ftype, call_id, handler, payload
+
+    def fail_all(self, exc: Exception):
+        for fut in self.pending_unary.values():
+            if not fut.done():
+                fut.set_exception(exc)
+        self.pending_unary.clear()
+        for queue in self.inbound_streams.values():
+            queue.put_nowait(exc)
+        for task in self.serving_tasks.values():
+            task.cancel()
+        self.closed.set()
+
+    async def close(self):
+        try:
+            self.writer.close()
+            await self.writer.wait_closed()
+        except Exception:
+            pass
+        self.fail_all(P2PDaemonError("connection closed"))
+
+
+class P2P:
+    """A peer: TCP listener + outgoing connection cache + RPC handler registry.
+
+    Reference counterpart: ``P2P.create`` (p2p_daemon.py:83). ``replicate`` in
+    the reference shares one daemon between processes; here components in one
+    process simply share the P2P object.
+    """
+
+    def __init__(self):
+        self._identity: Optional[PrivateKey] = None
+        self.peer_id: Optional[PeerID] = None
+        self._server: Optional[asyncio.AbstractServer] = None
+        self._listen_host: str = LOCALHOST
+        self._port: int = 0
+        self._handlers: Dict[str, Tuple[HandlerType, bool, bool]] = {}
+        self._balanced: Dict[str, List[Tuple[HandlerType, bool, bool]]] = {}
+        self._balanced_rr: Dict[str, int] = {}
+        self._connections: Dict[PeerID, _Connection] = {}
+        self._conn_lock = asyncio.Lock()
+        self._endpoint_book: Dict[PeerID, str] = {}  # last known endpoint per peer
+        self._alive = True
+        self._listen = True
+
+    # ------------------------------------------------------------------ setup
+
+    @classmethod
+    async def create(
+        cls,
+        listen_host: str = LOCALHOST,
+        port: int = 0,
+        identity: Optional[PrivateKey] = None,
+        listen: bool = True,
+    ) -> "P2P":
+        self = cls()
+        self._identity = identity if identity is not None else PrivateKey()
+        self.peer_id = PeerID.from_identity(self._identity)
+        self._listen_host = listen_host
+        self._listen = listen
+        if listen:
+            self._server = await asyncio.start_server(self._on_accept, listen_host, port)
+            self._port = self._server.sockets[0].getsockname()[1]
+        return self
+
+    @property
+    def endpoint(self) -> str:
+        return make_endpoint(self._listen_host, self._port)
+
+    @property
+    def peer_info(self) -> PeerInfo:
+        return PeerInfo(self.peer_id, (self.endpoint,) if self._listen else ())
+
+    def get_visible_maddrs(self) -> List[str]:
+        """API-compat: list of announce addresses (host:port strings)."""
+        return [self.endpoint] if self._listen else []
+
+    # -------------------------------------------------------------- handlers
+
+    def add_unary_handler(self, name: str, handler, balanced: bool = False):
+        self._add_handler(name, handler, False, False, balanced)
+
+    def add_stream_handler(self, name: str, handler, stream_input: bool = True, stream_output: bool = True, balanced: bool = False):
+        self._add_handler(name, handler, stream_input, stream_output, balanced)
+
+    def _add_handler(self, name: str, handler, stream_input: bool, stream_output: bool, balanced: bool):
+        entry = (handler, stream_input, stream_output)
+        if balanced:
+            self._balanced.setdefault(name, []).append(entry)
+            self._handlers.setdefault(name, entry)
+        elif name in self._handlers and name not in self._balanced:
+            raise ValueError(f"handler {name} already registered")
+        else:
+            self._handlers[name] = entry
+
+    def remove_handler(self, name: str):
+        self._handlers.pop(name, None)
+        self._balanced.pop(name, None)
+
+    def _resolve_handler(self, name: str) -> Tuple[HandlerType, bool, bool]:
+        if name in self._balanced:
+            entries = self._balanced[name]
+            idx = self._balanced_rr.get(name, 0) % len(entries)
+            self._balanced_rr[name] = idx + 1
+            return entries[idx]
+        if name not in self._handlers:
+            raise KeyError(f"no handler named {name}")
+        return self._handlers[name]
+
+    # ----------------------------------------------------------- connections
+
+    async def _on_accept(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
+        conn = _Connection(self, reader, writer)
+        try:
+            await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+            ftype, _, _, payload = await conn.recv_frame()
+            if ftype != _T_HELLO:
+                raise P2PDaemonError("expected HELLO")
+            conn.remote_id = PeerID(payload)
+        except Exception:
+            writer.close()
+            return
+        self._register_connection(conn)
+        conn.reader_task = asyncio.create_task(self._connection_loop(conn))
+
+    def _register_connection(self, conn: _Connection):
+        existing = self._connections.get(conn.remote_id)
+        if existing is None or existing.closed.is_set():
+            self._connections[conn.remote_id] = conn
+
+    async def _connect(self, peer: Union[PeerInfo, PeerID], endpoint: Optional[str] = None) -> _Connection:
+        if isinstance(peer, PeerInfo):
+            peer_id, endpoints = peer.peer_id, list(peer.endpoints)
+        else:
+            peer_id, endpoints = peer, []
+        conn = self._connections.get(peer_id)
+        if conn is not None and not conn.closed.is_set():
+            return conn
+        if endpoint is not None:
+            endpoints.insert(0, endpoint)
+        if peer_id in self._endpoint_book:
+            endpoints.append(self._endpoint_book[peer_id])
+        last_exc: Optional[Exception] = None
+        async with self._conn_lock:
+            conn = self._connections.get(peer_id)
+            if conn is not None and not conn.closed.is_set():
+                return conn
+            for ep in endpoints:
+                try:
+                    host, port = split_endpoint(ep)
+                    reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port), timeout=10)
+                    conn = _Connection(self, reader, writer)
+                    await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+                    ftype, _, _, payload = await conn.recv_frame()
+                    if ftype != _T_HELLO:
+                        raise P2PDaemonError("expected HELLO")
+                    conn.remote_id = PeerID(payload)
+                    if conn.remote_id != peer_id:
+                        logger.warning(f"peer at {ep} identifies as {conn.remote_id}, expected {peer_id}")
+                    self._endpoint_book[conn.remote_id] = ep
+                    self._register_connection(conn)
+                    conn.reader_task = asyncio.create_task(self._connection_loop(conn))
+                    return conn
+                except Exception as e:
+                    last_exc = e
+                    continue
+        raise P2PDaemonError(f"could not connect to {peer_id} via {endpoints}: {last_exc}")
+
+    async def _connection_loop(self, conn: _Connection):
+        try:
+            while True:
+                ftype, call_id, handler, payload = await conn.recv_frame()
+                if ftype == _T_REQUEST:
+                    task = asyncio.create_task(self._serve_unary(conn, call_id, handler, payload))
+                    conn.serving_tasks[call_id] = task
+                    task.add_done_callback(lambda _t, c=call_id: conn.serving_tasks.pop(c, None))
+                elif ftype == _T_RESPONSE:
+                    fut = conn.pending_unary.pop(call_id, None)
+                    if fut is not None and not fut.done():
+                        fut.set_result(payload)
+                elif ftype == _T_STREAM_OPEN:
+                    input_queue: asyncio.Queue = asyncio.Queue()
+                    conn.serving_streams[call_id] = input_queue
+                    task = asyncio.create_task(self._serve_stream(conn, call_id, handler, input_queue))
+                    conn.serving_tasks[call_id] = task
+                    task.add_done_callback(lambda _t, c=call_id: conn.serving_tasks.pop(c, None))
+                elif ftype == _T_STREAM_ITEM:
+                    if call_id in conn.serving_streams:
+                        conn.serving_streams[call_id].put_nowait(payload)
+                    elif call_id in conn.inbound_streams:
+                        conn.inbound_streams[call_id].put_nowait(payload)
+                elif ftype == _T_STREAM_HALF_CLOSE:
+                    if call_id in conn.serving_streams:
+                        conn.serving_streams[call_id].put_nowait(StopAsyncIteration)
+                elif ftype == _T_STREAM_END:
+                    if call_id in conn.inbound_streams:
+                        conn.inbound_streams[call_id].put_nowait(StopAsyncIteration)
+                elif ftype == _T_ERROR:
+                    err = P2PHandlerError(payload.decode("utf-8", "replace"))
+                    fut = conn.pending_unary.pop(call_id, None)
+                    if fut is not None and not fut.done():
+                        fut.set_exception(err)
+                    if call_id in conn.inbound_streams:
+                        conn.inbound_streams[call_id].put_nowait(err)
+                    if call_id in conn.serving_streams:
+                        conn.serving_streams[call_id].put_nowait(err)
+                elif ftype == _T_CANCEL:
+                    task = conn.serving_tasks.pop(call_id, None)
+                    if task is not None:
+                        task.cancel()
+                    conn.serving_streams.pop(call_id, None)
+                elif ftype == _T_PING:
+                    pass
+        except (asyncio.IncompleteReadError, ConnectionError, OSError):
+            pass
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:
+            logger.debug(f"connection loop error: {e!r}")
+        finally:
+            if self._connections.get(conn.remote_id) is conn:
+                self._connections.pop(conn.remote_id, None)
+            conn.fail_all(P2PDaemonError(f"connection to {conn.remote_id} lost"))
+
+    # -------------------------------------------------------------- serving
+
+    async def _serve_unary(self, conn: _Connection, call_id: bytes, handler_name: str, payload: bytes):
+        try:
+            handler, stream_in, stream_out = self._resolve_handler(handler_name)
+            assert not (stream_in or stream_out), f"{handler_name} is a streaming handler"
+            ctx = RpcContext(conn.remote_id, self.peer_id)
+            result = await handler(payload, ctx)
+            await conn.send_frame(_T_RESPONSE, call_id, "", result)
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:
+            logger.debug(f"handler {handler_name} failed: {e!r}")
+            try:
+                await conn.send_frame(_T_ERROR, call_id, "", f"{type(e).__name__}: {e}".encode())
+            except Exception:
+                pass
+
+    async def _serve_stream(self, conn: _Connection, call_id: bytes, handler_name: str, input_queue: asyncio.Queue):
+        async def _input_aiter() -> AsyncIterator[bytes]:
+            while True:
+                item = await input_queue.get()
+                if item is StopAsyncIteration:
+                    break
+                if isinstance(item, Exception):
+                    raise item
+                yield item
+
+        try:
+            handler, stream_in, stream_out = self._resolve_handler(handler_name)
+            ctx = RpcContext(conn.remote_id, self.peer_id)
+            if stream_in and stream_out:
+                aiter_out = handler(_input_aiter(), ctx)
+            elif stream_in and not stream_out:
+                async def _single_out():
+                    yield await handler(_input_aiter(), ctx)
+                aiter_out = _single_out()
+            else:  # unary in, stream out
+                first = await _input_aiter().__anext__()
+                aiter_out = handler(first, ctx)
+            async for item in aiter_out:
+                await conn.send_frame(_T_STREAM_ITEM, call_id, "", item)
+            await conn.send_frame(_T_STREAM_END, call_id, "", b"")
+        except asyncio.CancelledError:
+            raise
+        except Exception as e:
+            logger.debug(f"stream handler {handler_name} failed: {e!r}")
+            try:
+                await conn.send_frame(_T_ERROR, call_id, "", f"{type(e).__name__}: {e}".encode())
+            except Exception:
+                pass
+        finally:
+            conn.serving_streams.pop(call_id, None)
+
+    # --------------------------------------------------------------- calling
+
+    async def call_unary(
+        self,
+        peer: Union[PeerInfo, PeerID],
+        handler_name: str,
+        payload: bytes,
+        timeout: Optional[float] = None,
+    ) -> bytes:
+        conn = await self._connect(peer)
+        call_id = os.urandom(8)
+        fut: asyncio.Future = asyncio.get_event_loop().create_future()
+        conn.pending_unary[call_id] = fut
+        try:
+            await conn.send_frame(_T_REQUEST, call_id, handler_name, payload)
+            return await asyncio.wait_for(fut, timeout)
+        except (asyncio.TimeoutError, asyncio.CancelledError):
+            conn.pending_unary.pop(call_id, None)
+            try:
+                await conn.send_frame(_T_CANCEL, call_id, "", b"")
+            except Exception:
+                pass
+            raise
+
+    async def call_stream(
+        self,
+        peer: Union[PeerInfo, PeerID],
+        handler_name: str,
+        input_aiter: AsyncIterator[bytes],
+    ) -> AsyncIterator[bytes]:
+        """Bidirectional stream: feed `input_aiter` to the remote handler, yield its output."""
+        conn = await self._connect(peer)
+        call_id = os.urandom(8)
+        out_queue: asyncio.Queue = asyncio.Queue()
+        conn.inbound_streams[call_id] = out_queue
+        await conn.send_frame(_T_STREAM_OPEN, call_id, handler_name, b"")
+
+        async def _feeder():
+            try:
+                async for item in input_aiter:
+                    await conn.send_frame(_T_STREAM_ITEM, call_id, "", item)
+                await conn.send_frame(_T_STREAM_HALF_CLOSE, call_id, "", b"")
+            except asyncio.CancelledError:
+                raise
+            except Exception as e:
+                logger.debug(f"stream feeder failed: {e!r}")
+
+        feeder = asyncio.create_task(_feeder())
+        try:
+            while True:
+                item = await out_queue.get()
+                if item is StopAsyncIteration:
+                    break
+                if isinstance(item, Exception):
+                    raise item
+                yield item
+        except (GeneratorExit, asyncio.CancelledError):
+            try:
+                await conn.send_frame(_T_CANCEL, call_id, "", b"")
+            except Exception:
+                pass
+            raise
+        finally:
+            feeder.cancel()
+            conn.inbound_streams.pop(call_id, None)
+
+    # ------------------------------------------------------------- lifecycle
+
+    def learn_endpoint(self, peer_id: PeerID, endpoint: str):
+        """Remember where a peer can be reached (filled from DHT records)."""
+        self._endpoint_book[peer_id] = endpoint
+
+    async def ping(self, peer: Union[PeerInfo, PeerID]) -> bool:
+        try:
+            conn = await self._connect(peer)
+            await conn.send_frame(_T_PING, b"", "", b"")
+            return True
+        except Exception:
+            return False
+
+    async def list_peers(self) -> List[PeerInfo]:
+        return [
+            PeerInfo(pid, (self._endpoint_book.get(pid, ""),))
+            for pid, conn in self._connections.items()
+            if not conn.closed.is_set()
+        ]
+
+    async def disconnect(self, peer_id: PeerID):
+        conn = self._connections.pop(peer_id, None)
+        if conn is not None:
+            await conn.close()
+
+    async def shutdown(self):
+        self._alive = False
+        if self._server is not None:
+            self._server.close()
+            try:
+                await self._server.wait_closed()
+            except Exception:
+                pass
+        for conn in list(self._connections.values()):
+            await conn.close()
+        self._connections.clear()
+
+    def __repr__(self):
+        return f"P2P({self.peer_id}, {self.endpoint if self._listen else 'client-only'})"

@@ -1,0 +1,205 @@
+"""Peer identity & record signing: pure-Python Ed25519.
+
+The reference uses RSA-2048 PSS via the ``cryptography`` package
+(``hivemind/utils/crypto.py:35-101``). That package is unavailable here, and a
+native dependency would be overkill for signing kilobyte-scale DHT records, so
+this framework uses Ed25519 (stronger, smaller keys) implemented from the
+RFC 8032 equations. Signing a record takes ~2 ms in pure Python -- fine for the
+control plane (records are signed once per publish, every few seconds).
+
+API parity: ``PrivateKey.sign/get_public_key``, ``PublicKey.verify/to_bytes``,
+process-wide singleton key (reference crypto.py:35, process-wide RSA key).
+"""
+
+from __future__ import annotations
+
+import hashlib
+import os
+import threading
+from typing import Optional
+
+# ---------------------------------------------------------------------------
+# Ed25519 primitives (RFC 8032). Affine-free extended coordinates for speed.
+# ---------------------------------------------------------------------------
+
+_P = 2**255 - 19
+_L = 2**252 + 27742317777372353535851937790883648493
+_D = (-121665 * pow(121666, _P - 2, _P)) % _P
+_I = pow(2, (_P - 1) // 4, _P)
+
+
+def _sha512(data: bytes) -> bytes:
+    return hashlib.sha512(data).digest()
+
+
+def _inv(x: int) -> int:
+    return pow(x, _P - 2, _P)
+
+
+def _xrecover(y: int) -> int:
+    xx = (y * y - 1) * _inv(_D * y * y + 1)
+    x = pow(xx, (_P + 3) // 8, _P)
+    if (x * x - xx) % _P != 0:
+        x = (x * _I) % _P
+    if x % 2 != 0:
+        x = _P - x
+    return x
+
+
+_BY = 4 * _inv(5) % _P
+_BX = _xrecover(_BY)
+_B = (_BX % _P, _BY % _P, 1, (_BX * _BY) % _P)  # extended coords (X, Y, Z, T)
+_IDENT = (0, 1, 1, 0)
+
+
+def _edwards_add(p, q):
+    (x1, y1, z1, t1), (x2, y2, z2, t2) = p, q
+    a = (y1 - x1) * (y2 - x2) % _P
+    b = (y1 + x1) * (y2 + x2) % _P
+    c = t1 * 2 * _D * t2 % _P
+    dd = z1 * 2 * z2 % _P
+    e, f, g, h = b - a, dd - c, dd + c, b + a
+    return (e * f % _P, g * h % _P, f * g % _P, e * h % _P)
+
+
+def _edwards_double(p):
+    x1, y1, z1, _ = p
+    a = x1 * x1 % _P
+    b = y1 * y1 % _P
+    c = 2 * z1 * z1 % _P
+    e = ((x1 + y1) * (x1 + y1) - a - b) % _P
+    g = (-a + b) % _P  # a_edwards = -1
+    f = (g - c) % _P
+    h = (-a - b) % _P
+    return (e * f % _P, g * h % _P, f * g % _P, e * h % _P)
+
+
+def _scalarmult(p, e: int):
+    q = _IDENT
+    while e > 0:
+        if e & 1:
+            q = _edwards_add(q, p)
+        p = _edwards_double(p)
+        e >>= 1
+    return q
+
+
+def _point_compress(p) -> bytes:
+    x, y, z, _ = p
+    zi = _inv(z)
+    x, y = (x * zi) % _P, (y * zi) % _P
+    return int.to_bytes(y | ((x & 1) << 255), 32, "little")
+
+
+def _point_decompress(s: bytes):
+    y = int.from_bytes(s, "little")
+    sign = y >> 255
+    y &= (1 << 255) - 1
+    if y >= _P:
+        return None
+    x = _xrecover(y)
+    if x & 1 != sign:
+        x = _P - x
+    if (-x * x + y * y - 1 - _D * x * x * y * y) % _P != 0:
+        return None
+    return (x % _P, y % _P, 1, (x * y) % _P)
+
+
+def _point_equal(p, q) -> bool:
+    return (p[0] * q[2] - q[0] * p[2]) % _P == 0 and (p[1] * q[2] - q[1] * p[2]) % _P == 0
+
+
+def _secret_expand(secret: bytes):
+    h = _sha512(secret)
+    a = int.from_bytes(h[:32], "little")
+    a &= (1 << 254) - 8
+    a |= 1 << 254
+    return a, h[32:]
+
+
+def ed25519_public_key(secret: bytes) -> bytes:
+    a, _ = _secret_expand(secret)
+    return _point_compress(_scalarmult(_B, a))
+
+
+def ed25519_sign(secret: bytes, msg: bytes) -> bytes:
+    a, prefix = _secret_expand(secret)
+    pub = _point_compress(_scalarmult(_B, a))
+    r = int.from_bytes(_sha512(prefix + msg), "little") % _L
+    r_point = _point_compress(_scalarmult(_B, r))
+    h = int.from_bytes(_sha512(r_point + pub + msg), "little") % _L
+    s = (r + h * a) % _L
+    return r_point + int.to_bytes(s, 32, "little")
+
+
+def ed25519_verify(pub: bytes, msg: bytes, sig: bytes) -> bool:
+    if len(sig) != 64 or len(pub) != 32:
+        return False
+    a_point = _point_decompress(pub)
+    if a_point is None:
+        return False
+    r_point = _point_decompress(sig[:32])
+    if r_point is None:
+        return False
+    s = int.from_bytes(sig[32:], "little")
+    if s >= _L:
+        return False
+    h = int.from_bytes(_sha512(sig[:32] + pub + msg), "little") % _L
+    return _point_equal(_scalarmult(_B, s), _edwards_add(r_point, _scalarmult(a_point, h)))
+
+
+# ---------------------------------------------------------------------------
+# Key objects (API surface mirroring the reference's RSAPrivateKey/RSAPublicKey)
+# ---------------------------------------------------------------------------
+
+
+class PublicKey:
+    def __init__(self, key_bytes: bytes):
+        assert len(key_bytes) == 32, "Ed25519 public key must be 32 bytes"
+        self._bytes = key_bytes
+
+    def verify(self, data: bytes, signature: bytes) -> bool:
+        return ed25519_verify(self._bytes, data, signature)
+
+    def to_bytes(self) -> bytes:
+        return self._bytes
+
+    @classmethod
+    def from_bytes(cls, data: bytes) -> "PublicKey":
+        return cls(data)
+
+    def __eq__(self, other):
+        return isinstance(other, PublicKey) and other._bytes == self._bytes
+
+    def __hash__(self):
+        return hash(self._bytes)
+
+
+class PrivateKey:
+    _process_wide: Optional["PrivateKey"] = None
+    _lock = threading.Lock()
+
+    def __init__(self, secret: Optional[bytes] = None):
+        self._secret = secret if secret is not None else os.urandom(32)
+        self._public = PublicKey(ed25519_public_key(self._secret))
+
+    def sign(self, data: bytes) -> bytes:
+        return ed25519_sign(self._secret, data)
+
+    def get_public_key(self) -> PublicKey:
+        return self._public
+
+    def to_bytes(self) -> bytes:
+        return self._secret
+
+    @classmethod
+    def from_bytes(cls, data: bytes) -> "PrivateKey":
+        return cls(data)
+
+    @classmethod
+    def process_wide(cls) -> "PrivateKey":
+        """One key per process, generated lazily (reference crypto.py:45-56)."""
+        with cls._lock:
+            if cls._process_wide is None:
+                cls._process_wide = cls()
+            return cls._process_wide

@@ -1,0 +1,147 @@
+"""Transport-layer tests (reference shape: tests/test_p2p_daemon*.py)."""
+
+import asyncio
+from dataclasses import dataclass
+from typing import AsyncIterator
+
+import pytest
+
+from hivemind_amd.p2p import P2P, P2PHandlerError, PeerID, RpcMessage, ServicerBase
+from hivemind_amd.utils.crypto import PrivateKey
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_peer_id_roundtrip():
+    key = PrivateKey()
+    pid = PeerID.from_identity(key)
+    assert PeerID.from_base58(pid.to_base58()) == pid
+    assert pid == PeerID(pid.to_bytes())
+    assert len({pid, PeerID.from_identity(key)}) == 1
+
+
+def test_ed25519_sign_verify():
+    key = PrivateKey()
+    pub = key.get_public_key()
+    sig = key.sign(b"hello world")
+    assert pub.verify(b"hello world", sig)
+    assert not pub.verify(b"hello worlds", sig)
+    assert not pub.verify(b"hello world", sig[:-1] + bytes([sig[-1] ^ 1]))
+    # round-trip through bytes
+    from hivemind_amd.utils.crypto import PublicKey
+
+    assert PublicKey.from_bytes(pub.to_bytes()).verify(b"hello world", sig)
+
+
+def test_unary_call():
+    async def main():
+        server = await P2P.create()
+        client = await P2P.create()
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return b"echo:" + payload
+
+        server.add_unary_handler("echo", echo)
+        result = await client.call_unary(server.peer_info, "echo", b"hi", timeout=5)
+        assert result == b"echo:hi"
+        # error propagation
+        async def boom(payload: bytes, ctx) -> bytes:
+            raise ValueError("nope")
+
+        server.add_unary_handler("boom", boom)
+        with pytest.raises(P2PHandlerError):
+            await client.call_unary(server.peer_info, "boom", b"", timeout=5)
+        await client.shutdown()
+        await server.shutdown()
+
+    run(main())
+
+
+def test_stream_call():
+    async def main():
+        server = await P2P.create()
+        client = await P2P.create()
+
+        async def doubler(input_aiter, ctx):
+            async for item in input_aiter:
+                yield item * 2
+
+        server.add_stream_handler("doubler", doubler)
+
+        async def inputs():
+            for i in range(5):
+                yield bytes([i])
+
+        received = []
+        async for item in client.call_stream(server.peer_info, "doubler", inputs()):
+            received.append(item)
+        assert received == [bytes([i, i]) for i in range(5)]
+        await client.shutdown()
+        await server.shutdown()
+
+    run(main())
+
+
+def test_bidirectional_reuse():
+    """Both sides can call each other over one connection."""
+
+    async def main():
+        a = await P2P.create()
+        b = await P2P.create()
+
+        async def name_handler(payload, ctx):
+            return b"a"
+
+        async def name_handler_b(payload, ctx):
+            return b"b"
+
+        a.add_unary_handler("whoami", name_handler)
+        b.add_unary_handler("whoami", name_handler_b)
+        assert await a.call_unary(b.peer_info, "whoami", b"", timeout=5) == b"b"
+        # b now calls back over the same TCP connection (a's PeerID only)
+        assert await b.call_unary(a.peer_id, "whoami", b"", timeout=5) == b"a"
+        await a.shutdown()
+        await b.shutdown()
+
+    run(main())
+
+
+@dataclass
+class Ping(RpcMessage):
+    text: str = ""
+    number: int = 0
+
+
+@dataclass
+class Pong(RpcMessage):
+    reply: str = ""
+
+
+class PingServicer(ServicerBase):
+    async def rpc_ping(self, request: Ping, context) -> Pong:
+        return Pong(reply=f"{request.text}/{request.number}")
+
+    async def rpc_count(self, request: Ping, context) -> AsyncIterator[Pong]:
+        for i in range(request.number):
+            yield Pong(reply=str(i))
+
+
+def test_servicer_stub():
+    async def main():
+        server = await P2P.create()
+        client = await P2P.create()
+        servicer = PingServicer()
+        await servicer.add_p2p_handlers(server)
+        stub = PingServicer.get_stub(client, server.peer_info)
+        raw = await stub.rpc_ping(Ping(text="x", number=7), timeout=5)
+        assert Pong.loads(raw).reply == "x/7"
+        replies = []
+        async for payload in stub.rpc_count(Ping(number=3)):
+            replies.append(Pong.loads(payload).reply)
+        assert replies == ["0", "1", "2"]
+        await client.shutdown()
+        await server.shutdown()
+
+    run(main())
