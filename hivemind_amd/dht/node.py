@@ -1,0 +1,492 @@
+"""DHTNode: high-level Kademlia node with bulk get/store, caching and blacklist.
+
+Parity target: reference ``hivemind/dht/node.py:98-931`` -- ``DHTNode.create``
+(bootstrap from initial peers), ``store_many`` (traverse to ``num_replicas``
+nearest then bulk call_store), ``get_many`` (concurrent beam search with
+early-exit once a sufficiently fresh value is found, caching policies, reuse
+of concurrent gets), and an exponential-backoff ``Blacklist`` of dead peers.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import random
+from collections import defaultdict
+from dataclasses import dataclass, field as dataclass_field
+from typing import Any, Collection, Dict, List, Optional, Sequence, Tuple, Union
+
+from ..p2p import P2P, PeerID, PeerInfo
+from ..utils.logging import get_logger
+from ..utils.serializer import MSGPackSerializer
+from ..utils.timed_storage import DHTExpiration, TimedStorage, ValueWithExpiration, get_dht_time
+from .protocol import DHTProtocol
+from .routing import BinaryDHTValue, DHTID, DHTKey, Subkey
+from .storage import DictionaryDHTValue
+from .traverse import traverse_dht
+from .validation import RecordValidatorBase
+
+logger = get_logger(__name__)
+
+DEFAULT_NUM_WORKERS = int(os.environ.get("HIVEMIND_DHT_NUM_WORKERS", 4))
+
+
+class Blacklist:
+    """Exponential-backoff ban list for unresponsive peers (reference node.py:897-931)."""
+
+    def __init__(self, base_time: float = 5.0, backoff_rate: float = 2.0):
+        self.base_time, self.backoff = base_time, backoff_rate
+        self.banned_peers = TimedStorage()
+        self.ban_counter: Dict[PeerID, int] = defaultdict(int)
+
+    def register_failure(self, peer: PeerID):
+        if peer not in self.banned_peers and self.base_time > 0:
+            ban_duration = self.base_time * self.backoff ** self.ban_counter[peer]
+            self.banned_peers.store(peer, self.ban_counter[peer], expiration_time=get_dht_time() + ban_duration)
+            self.ban_counter[peer] += 1
+
+    def register_success(self, peer: PeerID):
+        del self.banned_peers[peer]
+        self.ban_counter.pop(peer, None)
+
+    def __contains__(self, peer: PeerID) -> bool:
+        return peer in self.banned_peers
+
+
+@dataclass
+class _SearchState:
+    """Progress of finding one key (reference node.py:814-895)."""
+
+    key_id: DHTID
+    sufficient_expiration_time: DHTExpiration
+    binary_value: Optional[Union[BinaryDHTValue, DictionaryDHTValue]] = None
+    expiration_time: Optional[DHTExpiration] = None
+    source_node_id: Optional[DHTID] = None
+    future: asyncio.Future = dataclass_field(default_factory=asyncio.Future)
+    nearest_nodes: List[DHTID] = dataclass_field(default_factory=list)
+
+    def add_candidate(self, candidate: Optional[ValueWithExpiration], source_node_id: Optional[DHTID]):
+        if self.finished or candidate is None:
+            return
+        if self.expiration_time is None or candidate.expiration_time > self.expiration_time:
+            self.binary_value, self.expiration_time = candidate.value, candidate.expiration_time
+            self.source_node_id = source_node_id
+        elif isinstance(self.binary_value, DictionaryDHTValue) and isinstance(candidate.value, DictionaryDHTValue):
+            for subkey, (sub_value, sub_expiration) in candidate.value.items():
+                self.binary_value.store(subkey, sub_value, sub_expiration)
+        if self.expiration_time is not None and self.expiration_time >= self.sufficient_expiration_time:
+            self.finish_search()
+
+    @property
+    def found_something(self) -> bool:
+        return self.expiration_time is not None
+
+    @property
+    def finished(self) -> bool:
+        return self.future.done()
+
+    def finish_search(self):
+        if not self.future.done():
+            result = (
+                ValueWithExpiration(self.binary_value, self.expiration_time) if self.found_something else None
+            )
+            self.future.set_result(result)
+
+
+class DHTNode:
+    """An asyncio Kademlia node. All methods run on the caller's event loop."""
+
+    def __init__(self):
+        self.node_id: Optional[DHTID] = None
+        self.protocol: Optional[DHTProtocol] = None
+        self.p2p: Optional[P2P] = None
+        self.num_replicas = 5
+        self.num_workers = DEFAULT_NUM_WORKERS
+        self.beam_size = 20
+        self.chunk_size = 16
+        self.cache_locally = True
+        self.cache_on_store = True
+        self.cache_nearest = 1
+        self.reuse_get_requests = True
+        self.blacklist = Blacklist()
+        self.is_alive = True
+        self._pending_get_requests: Dict[DHTID, List[_SearchState]] = defaultdict(list)
+        self._should_shutdown_p2p = False
+
+    @classmethod
+    async def create(
+        cls,
+        p2p: Optional[P2P] = None,
+        node_id: Optional[DHTID] = None,
+        initial_peers: Sequence[Union[str, PeerInfo]] = (),
+        bucket_size: int = 20,
+        num_replicas: int = 5,
+        depth_modulo: int = 5,
+        wait_timeout: float = 3.0,
+        bootstrap_timeout: Optional[float] = None,
+        num_workers: int = DEFAULT_NUM_WORKERS,
+        beam_size: Optional[int] = None,
+        cache_locally: bool = True,
+        cache_on_store: bool = True,
+        cache_nearest: int = 1,
+        cache_size: int = 10000,
+        reuse_get_requests: bool = True,
+        listen_host: str = "127.0.0.1",
+        port: int = 0,
+        client_mode: bool = False,
+        record_validator: Optional[RecordValidatorBase] = None,
+        blacklist_time: float = 5.0,
+        backoff_rate: float = 2.0,
+    ) -> "DHTNode":
+        self = cls()
+        self.node_id = node_id if node_id is not None else DHTID.generate()
+        self.num_replicas, self.num_workers = num_replicas, num_workers
+        self.beam_size = beam_size if beam_size is not None else bucket_size
+        self.cache_locally, self.cache_on_store, self.cache_nearest = cache_locally, cache_on_store, cache_nearest
+        self.reuse_get_requests = reuse_get_requests
+        self.blacklist = Blacklist(blacklist_time, backoff_rate)
+        if p2p is None:
+            p2p = await P2P.create(listen_host=listen_host, port=port, listen=not client_mode)
+            self._should_shutdown_p2p = True
+        self.p2p = p2p
+        self.protocol = await DHTProtocol.create(
+            p2p,
+            self.node_id,
+            bucket_size=bucket_size,
+            depth_modulo=depth_modulo,
+            num_replicas=num_replicas,
+            wait_timeout=wait_timeout,
+            cache_size=cache_size,
+            client_mode=client_mode,
+            record_validator=record_validator,
+        )
+        if initial_peers:
+            await self._bootstrap(initial_peers, bootstrap_timeout or wait_timeout * 8)
+        return self
+
+    async def _bootstrap(self, initial_peers: Sequence[Union[str, PeerInfo]], timeout: float):
+        """Ping initial peers, then look up our own id to fill the routing table
+        (reference node.py:98-180 bootstrap logic)."""
+        peer_infos = []
+        for peer in initial_peers:
+            if isinstance(peer, PeerInfo):
+                peer_infos.append(peer)
+            else:
+                # bare endpoint string: identity will be learned from the HELLO
+                peer_infos.append(peer)
+
+        async def ping_peer(peer) -> bool:
+            if isinstance(peer, str):
+                try:
+                    peer_obj = await self.p2p.connect_endpoint(peer)
+                except Exception:
+                    return False
+            else:
+                peer_obj = peer
+            return (await self.protocol.call_ping(peer_obj)) is not None
+
+        results = await asyncio.gather(*(ping_peer(p) for p in peer_infos), return_exceptions=True)
+        if not any(r is True for r in results):
+            logger.warning("DHTNode bootstrap: no initial peers responded")
+            return
+        try:
+            await asyncio.wait_for(self.find_nearest_nodes([self.node_id]), timeout)
+        except asyncio.TimeoutError:
+            logger.warning("DHTNode bootstrap: self-lookup timed out")
+
+    # ------------------------------------------------------------- traversal
+
+    def _peer_tuple(self, node_id: DHTID, node_to_peer: Optional[Dict[DHTID, PeerInfo]] = None) -> Optional[PeerInfo]:
+        if node_to_peer is not None and node_id in node_to_peer:
+            return node_to_peer[node_id]
+        entry = self.protocol.routing_table.get(node_id=node_id)
+        if entry is None:
+            return None
+        peer_id, endpoint = entry
+        return PeerInfo(peer_id, (endpoint,))
+
+    async def _call_find_with_blacklist(
+        self, node_id: DHTID, keys: Collection[DHTID], node_to_peer: Optional[Dict[DHTID, PeerInfo]] = None
+    ):
+        peer = self._peer_tuple(node_id, node_to_peer)
+        if peer is None or peer.peer_id in self.blacklist:
+            return None
+        result = await self.protocol.call_find(peer, keys)
+        if result is None:
+            self.blacklist.register_failure(peer.peer_id)
+        else:
+            self.blacklist.register_success(peer.peer_id)
+        return result
+
+    async def find_nearest_nodes(
+        self,
+        queries: Collection[DHTID],
+        k_nearest: Optional[int] = None,
+        beam_size: Optional[int] = None,
+        num_workers: Optional[int] = None,
+        node_to_peer: Optional[Dict[DHTID, PeerInfo]] = None,
+        exclude_self: bool = False,
+    ) -> Dict[DHTID, Dict[DHTID, PeerInfo]]:
+        """Beam-search the k nearest nodes for each query (reference node.py:278-350)."""
+        queries = list(queries)
+        k_nearest = k_nearest if k_nearest is not None else self.protocol.bucket_size
+        beam_size = beam_size if beam_size is not None else max(self.beam_size, k_nearest)
+        num_workers = num_workers if num_workers is not None else self.num_workers
+        node_to_peer = dict(node_to_peer or {})
+
+        initial: List[DHTID] = []
+        for q in queries:
+            for nid, (peer_id, endpoint) in self.protocol.routing_table.get_nearest_neighbors(
+                q, beam_size, exclude=self.node_id
+            ):
+                node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                if nid not in initial:
+                    initial.append(nid)
+
+        async def get_neighbors(node: DHTID, lookup_queries: Collection[DHTID]):
+            response = await self._call_find_with_blacklist(node, lookup_queries, node_to_peer)
+            if response is None:
+                return {q: ([], False) for q in lookup_queries}
+            output = {}
+            for q, (_, nearest) in response.items():
+                for nid, (peer_id, endpoint) in nearest.items():
+                    node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                output[q] = (list(nearest.keys()), False)
+            return output
+
+        nearest_nodes, _visited = await traverse_dht(
+            queries, initial, beam_size, num_workers, queries_per_call=min(len(queries), self.chunk_size),
+            get_neighbors=get_neighbors,
+        )
+        output: Dict[DHTID, Dict[DHTID, PeerInfo]] = {}
+        for q in queries:
+            found = nearest_nodes.get(q, [])
+            if not exclude_self:
+                found = sorted(set(found) | {self.node_id}, key=q.xor_distance)
+                node_to_peer[self.node_id] = self.p2p.peer_info
+            output[q] = {nid: node_to_peer[nid] for nid in found[:k_nearest] if nid in node_to_peer or nid == self.node_id}
+        return output
+
+    # ----------------------------------------------------------------- store
+
+    async def store(
+        self, key: DHTKey, value: Any, expiration_time: DHTExpiration, subkey: Optional[Subkey] = None, **kwargs
+    ) -> bool:
+        result = await self.store_many([key], [value], [expiration_time], subkeys=[subkey], **kwargs)
+        return result[(key, subkey) if subkey is not None else key]
+
+    async def store_many(
+        self,
+        keys: List[DHTKey],
+        values: List[Any],
+        expiration_time: Union[DHTExpiration, List[DHTExpiration]],
+        subkeys: Optional[List[Optional[Subkey]]] = None,
+        exclude_self: bool = False,
+        await_all_replicas: bool = True,
+    ) -> Dict[Any, bool]:
+        """Traverse to find num_replicas nearest nodes per key, then bulk-store
+        (reference node.py:351-503)."""
+        if isinstance(expiration_time, (int, float)):
+            expiration_time = [expiration_time] * len(keys)
+        if subkeys is None:
+            subkeys = [None] * len(keys)
+        assert len(keys) == len(values) == len(expiration_time) == len(subkeys)
+
+        key_ids = [DHTID.generate(source=key) for key in keys]
+        id_to_original: Dict[Tuple[DHTID, Any], Any] = {}
+        binary_values: Dict[Tuple[DHTID, Any], bytes] = {}
+        expirations: Dict[Tuple[DHTID, Any], DHTExpiration] = {}
+        for key, key_id, subkey, value, expiration in zip(keys, key_ids, subkeys, values, expiration_time):
+            composite = (key_id, subkey)
+            id_to_original[composite] = (key, subkey) if subkey is not None else key
+            binary_values[composite] = value if isinstance(value, (bytes, DictionaryDHTValue)) else MSGPackSerializer.dumps(value)
+            expirations[composite] = expiration
+
+        unfinished = set(binary_values.keys())
+        store_ok: Dict[Any, bool] = {id_to_original[c]: False for c in unfinished}
+        nearest_per_query = await self.find_nearest_nodes(
+            list(dict.fromkeys(key_ids)), k_nearest=self.num_replicas, exclude_self=exclude_self
+        )
+
+        async def store_on_peers(key_id: DHTID, nearest: Dict[DHTID, PeerInfo]):
+            composites = [c for c in binary_values if c[0] == key_id]
+            in_my_range = not exclude_self and self.node_id in nearest
+            tasks = []
+            for nid, peer in nearest.items():
+                if nid == self.node_id:
+                    # store locally
+                    for c in composites:
+                        _, subkey = c
+                        if subkey is None:
+                            ok = self.protocol.storage.store(key_id, binary_values[c], expirations[c])
+                        else:
+                            ok = self.protocol.storage.store_subkey(key_id, subkey, binary_values[c], expirations[c])
+                        store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or ok
+                    continue
+                ks = [key_id] * len(composites)
+                vs = [binary_values[c] for c in composites]
+                exps = [expirations[c] for c in composites]
+                subs = [c[1] for c in composites]
+                tasks.append(self.protocol.call_store(peer, ks, vs, exps, subkeys=subs))
+            responses = await asyncio.gather(*tasks, return_exceptions=True)
+            for response in responses:
+                if isinstance(response, Exception) or response is None:
+                    continue
+                for c, ok in zip(composites, response):
+                    store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or bool(ok)
+
+        await asyncio.gather(*(store_on_peers(key_id, nearest) for key_id, nearest in nearest_per_query.items()))
+
+        if self.cache_on_store:
+            for c, value in binary_values.items():
+                key_id, subkey = c
+                if subkey is None and isinstance(value, bytes):
+                    self.protocol.cache.store(key_id, value, expirations[c])
+        return store_ok
+
+    # ------------------------------------------------------------------- get
+
+    async def get(self, key: DHTKey, latest: bool = False, **kwargs) -> Optional[ValueWithExpiration]:
+        if latest:
+            kwargs["sufficient_expiration_time"] = float("inf")
+        result = await self.get_many([key], **kwargs)
+        return result[key]
+
+    async def get_many(
+        self,
+        keys: Collection[DHTKey],
+        sufficient_expiration_time: Optional[DHTExpiration] = None,
+        **kwargs,
+    ) -> Dict[DHTKey, Optional[ValueWithExpiration]]:
+        keys = tuple(keys)
+        key_ids = [DHTID.generate(source=key) for key in keys]
+        id_to_original_key = dict(zip(key_ids, keys))
+        results_by_id = await self.get_many_by_id(key_ids, sufficient_expiration_time, **kwargs)
+        return {id_to_original_key[key_id]: result for key_id, result in results_by_id.items()}
+
+    async def get_many_by_id(
+        self,
+        key_ids: Collection[DHTID],
+        sufficient_expiration_time: Optional[DHTExpiration] = None,
+        num_workers: Optional[int] = None,
+        beam_size: Optional[int] = None,
+        return_futures: bool = False,
+    ) -> Dict[DHTID, Union[Optional[ValueWithExpiration], asyncio.Future]]:
+        """Traverse the DHT for each key, returning the freshest value found
+        (reference node.py:569-812)."""
+        key_ids = list(key_ids)
+        sufficient_expiration_time = sufficient_expiration_time or get_dht_time()
+        beam_size = beam_size if beam_size is not None else self.beam_size
+        num_workers = num_workers if num_workers is not None else self.num_workers
+        search_results: Dict[DHTID, _SearchState] = {
+            key_id: _SearchState(key_id, sufficient_expiration_time) for key_id in key_ids
+        }
+
+        # reuse concurrent requests for the same key (reference node.py:586-605)
+        if self.reuse_get_requests:
+            for key_id, search in search_results.items():
+                self._pending_get_requests[key_id].append(search)
+                search.future.add_done_callback(
+                    lambda _fut, key_id=key_id, search=search: self._pending_get_requests[key_id].remove(search)
+                    if search in self._pending_get_requests[key_id]
+                    else None
+                )
+
+        # check local storage and cache first
+        for key_id, search in search_results.items():
+            search.add_candidate(self.protocol.storage.get(key_id), source_node_id=self.node_id)
+            search.add_candidate(self.protocol.cache.get(key_id), source_node_id=self.node_id)
+
+        unfinished_ids = [key_id for key_id, s in search_results.items() if not s.finished]
+        node_to_peer: Dict[DHTID, PeerInfo] = {}
+        initial: List[DHTID] = []
+        for q in unfinished_ids:
+            for nid, (peer_id, endpoint) in self.protocol.routing_table.get_nearest_neighbors(
+                q, beam_size, exclude=self.node_id
+            ):
+                node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                if nid not in initial:
+                    initial.append(nid)
+
+        async def get_neighbors(node: DHTID, lookup_queries: Collection[DHTID]):
+            queries_alive = [q for q in lookup_queries if not search_results[q].finished]
+            if not queries_alive:
+                return {q: ([], True) for q in lookup_queries}
+            response = await self._call_find_with_blacklist(node, queries_alive, node_to_peer)
+            if response is None:
+                return {q: ([], False) for q in lookup_queries}
+            output = {}
+            for q, (maybe_value, nearest) in response.items():
+                search_results[q].add_candidate(maybe_value, source_node_id=node)
+                for nid, (peer_id, endpoint) in nearest.items():
+                    node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                output[q] = (list(nearest.keys()), search_results[q].finished)
+            return output
+
+        async def found_callback(query: DHTID, nearest_nodes: List[DHTID], _visited):
+            search = search_results[query]
+            search.nearest_nodes = nearest_nodes
+            search.finish_search()
+            self._cache_new_result(search, nearest_nodes, node_to_peer)
+
+        if unfinished_ids:
+            traverse_task = asyncio.create_task(
+                traverse_dht(
+                    unfinished_ids,
+                    initial,
+                    beam_size,
+                    num_workers,
+                    queries_per_call=min(len(unfinished_ids), self.chunk_size),
+                    get_neighbors=get_neighbors,
+                    found_callback=found_callback,
+                )
+            )
+        else:
+            traverse_task = None
+            for search in search_results.values():
+                search.finish_search()
+
+        if return_futures:
+            return {key_id: search.future for key_id, search in search_results.items()}
+        try:
+            if traverse_task is not None:
+                await traverse_task
+            for search in search_results.values():
+                search.finish_search()
+            return {key_id: await search.future for key_id, search in search_results.items()}
+        except asyncio.CancelledError:
+            if traverse_task is not None:
+                traverse_task.cancel()
+            raise
+
+    def _cache_new_result(self, search: _SearchState, nearest_nodes: List[DHTID], node_to_peer: Dict[DHTID, PeerInfo]):
+        """After a search: cache locally and/or on the nearest nodes (reference node.py:763-794)."""
+        if not search.found_something:
+            return
+        cached_value = search.binary_value
+        if isinstance(cached_value, DictionaryDHTValue):
+            return  # dictionaries are merged, not cached wholesale
+        if self.cache_locally and search.source_node_id != self.node_id:
+            self.protocol.cache.store(search.key_id, cached_value, search.expiration_time)
+        if self.cache_nearest:
+            num_cached = 0
+            for node_id in nearest_nodes:
+                if node_id == search.source_node_id or node_id == self.node_id:
+                    continue
+                peer = node_to_peer.get(node_id)
+                if peer is None:
+                    continue
+                asyncio.get_event_loop().create_task(
+                    self.protocol.call_store(
+                        peer, [search.key_id], [cached_value], [search.expiration_time], in_cache=True
+                    )
+                )
+                num_cached += 1
+                if num_cached >= self.cache_nearest:
+                    break
+
+    async def shutdown(self):
+        self.is_alive = False
+        await self.protocol.shutdown()
+        if self._should_shutdown_p2p:
+            await self.p2p.shutdown()

@@ -229,6 +229,25 @@ class P2P:
         self._register_connection(conn)
         conn.reader_task = asyncio.create_task(self._connection_loop(conn))
 
+    async def connect_endpoint(self, endpoint: str) -> PeerInfo:
+        """Dial a bare endpoint and learn the peer's identity (bootstrap helper)."""
+        host, port = split_endpoint(endpoint)
+        reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port), timeout=10)
+        conn = _Connection(self, reader, writer)
+        await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+        ftype, _, _, payload = await conn.recv_frame()
+        if ftype != _T_HELLO:
+            raise P2PDaemonError("expected HELLO")
+        conn.remote_id = PeerID(payload)
+        self._endpoint_book[conn.remote_id] = endpoint
+        existing = self._connections.get(conn.remote_id)
+        if existing is not None and not existing.closed.is_set():
+            await conn.close()
+            return PeerInfo(existing.remote_id, (endpoint,))
+        self._register_connection(conn)
+        conn.reader_task = asyncio.create_task(self._connection_loop(conn))
+        return PeerInfo(conn.remote_id, (endpoint,))
+
     def _register_connection(self, conn: _Connection):
         existing = self._connections.get(conn.remote_id)
         if existing is None or existing.closed.is_set():

@@ -1,0 +1,147 @@
+"""DHT tests (reference shape: tests/test_dht_node.py, test_dht.py, test_routing.py)."""
+
+import asyncio
+import random
+import time
+
+import pytest
+
+from hivemind_amd.dht import DHT, DHTID, DHTNode, RoutingTable
+from hivemind_amd.dht.storage import DHTLocalStorage, DictionaryDHTValue
+from hivemind_amd.utils import MSGPackSerializer, get_dht_time
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+def test_dht_id():
+    a = DHTID.generate(source=b"hello")
+    b = DHTID.generate(source=b"hello")
+    c = DHTID.generate(source=b"world")
+    assert a == b != c
+    assert a.xor_distance(a) == 0
+    assert a.xor_distance(c) == c.xor_distance(a) > 0
+    assert DHTID.from_bytes(a.to_bytes()) == a
+    d = DHTID.generate()
+    assert DHTID.MIN <= d <= DHTID.MAX
+
+
+def test_routing_table_basic():
+    node_id = DHTID.generate()
+    table = RoutingTable(node_id, bucket_size=4, depth_modulo=5)
+    from hivemind_amd.p2p import PeerID
+
+    added = []
+    for i in range(100):
+        uid = DHTID.generate()
+        pid = PeerID(bytes([i]) * 4)
+        table.add_or_update_node(uid, pid, f"127.0.0.1:{1000+i}")
+        added.append(uid)
+    present = [uid for uid in added if uid in table]
+    assert len(present) > 4  # buckets split around our own id
+    # nearest queries are sorted by xor distance
+    query = DHTID.generate()
+    nearest = table.get_nearest_neighbors(query, k=8)
+    dists = [query.xor_distance(uid) for uid, _ in nearest]
+    assert dists == sorted(dists)
+
+
+def test_timed_storage_and_dictionary():
+    storage = DHTLocalStorage()
+    key = DHTID.generate()
+    now = get_dht_time()
+    assert storage.store(key, b"v1", now + 10)
+    assert storage.get(key).value == b"v1"
+    assert not storage.store(key, b"v0", now + 5)  # older expiration loses
+    assert storage.store(key, b"v2", now + 20)
+    assert storage.get(key).value == b"v2"
+    # subkeys
+    key2 = DHTID.generate()
+    assert storage.store_subkey(key2, "alpha", b"a", now + 10)
+    assert storage.store_subkey(key2, "beta", b"b", now + 15)
+    value = storage.get(key2).value
+    assert isinstance(value, DictionaryDHTValue)
+    assert value.get("alpha").value == b"a"
+    assert storage.get(key2).expiration_time == now + 15
+    # dictionary round-trips through msgpack ext
+    packed = MSGPackSerializer.dumps(value)
+    unpacked = MSGPackSerializer.loads(packed)
+    assert isinstance(unpacked, DictionaryDHTValue)
+    assert unpacked.get("beta").value == b"b"
+
+
+def test_dht_node_store_get():
+    async def main():
+        alice = await DHTNode.create()
+        bob = await DHTNode.create(initial_peers=[alice.p2p.endpoint])
+        carol = await DHTNode.create(initial_peers=[alice.p2p.endpoint])
+
+        now = get_dht_time()
+        assert await bob.store("key1", MSGPackSerializer.dumps([1, 2, 3]), now + 30)
+        result = await carol.get("key1")
+        assert result is not None and MSGPackSerializer.loads(result.value) == [1, 2, 3]
+        # missing key
+        assert await carol.get("no-such-key") is None
+        # subkey stores merge
+        assert await alice.store("dictkey", MSGPackSerializer.dumps("a"), now + 30, subkey="s1")
+        assert await bob.store("dictkey", MSGPackSerializer.dumps("b"), now + 31, subkey="s2")
+        result = await carol.get("dictkey", latest=True)
+        assert isinstance(result.value, DictionaryDHTValue)
+        subkeys = {k for k, _ in result.value.items()}
+        assert subkeys == {"s1", "s2"}
+        for node in (alice, bob, carol):
+            await node.shutdown()
+
+    run(main())
+
+
+def test_dht_node_swarm_20():
+    """Store/get across a 20-node swarm (reference test_dht_node.py:24-164 shape)."""
+
+    async def main():
+        nodes = [await DHTNode.create()]
+        for _ in range(19):
+            peers = random.sample([n.p2p.endpoint for n in nodes], min(3, len(nodes)))
+            nodes.append(await DHTNode.create(initial_peers=peers))
+        now = get_dht_time()
+        for i in range(10):
+            writer, reader = random.sample(nodes, 2)
+            assert await writer.store(f"key{i}", MSGPackSerializer.dumps(i), now + 60)
+            result = await reader.get(f"key{i}")
+            assert result is not None and MSGPackSerializer.loads(result.value) == i
+        await asyncio.gather(*(n.shutdown() for n in nodes))
+
+    run(main())
+
+
+def test_dht_facade():
+    dht1 = DHT(start=True)
+    dht2 = DHT(initial_peers=[dht1.endpoint], start=True)
+    now = get_dht_time()
+    assert dht1.store("facade_key", {"x": 1}, now + 30)
+    result = dht2.get("facade_key")
+    assert result is not None and result.value == {"x": 1}
+    assert dht2.get("missing_key") is None
+    # store with subkey
+    assert dht1.store("fdict", 11, now + 30, subkey="a")
+    assert dht2.store("fdict", 22, now + 30, subkey="b")
+    result = dht1.get("fdict", latest=True)
+    assert result.value["a"].value == 11 and result.value["b"].value == 22
+    # run_coroutine
+    async def get_node_id(dht, node):
+        return node.node_id
+
+    assert dht1.run_coroutine(get_node_id) == dht1.node_id
+    dht2.shutdown()
+    dht1.shutdown()
+
+
+def test_dht_expiration():
+    dht = DHT(start=True)
+    now = get_dht_time()
+    assert dht.store("ephemeral", 1, now + 0.5)
+    assert dht.get("ephemeral").value == 1
+    time.sleep(0.7)
+    assert dht.get("ephemeral") is None
+    dht.shutdown()
