@@ -1,0 +1,175 @@
+"""GradientAverager: accumulates local gradients, averages them with the swarm.
+
+Parity target: reference ``hivemind/optim/grad_averager.py:18-239``. Three
+buffer sets, as in the reference:
+
+* ``param.grad``            -- written by autograd;
+* local accumulators        -- running sum across micro-batches (equal to
+  ``param.grad`` itself when ``reuse_grad_buffers=True``);
+* averaged tensors          -- the averager's buffers, filled from the
+  accumulators right before each round.
+
+Unlike the reference (which keeps averaged tensors in shared CPU memory for
+its fork-based averager), the averaged buffers here live on the *parameter
+device* -- on MI355X that keeps the whole averaging round in HBM and lets the
+RCCL data plane run without host round-trips.
+"""
+
+from __future__ import annotations
+
+import contextlib
+from typing import Iterable, Iterator, Optional
+
+import torch
+
+from ..averaging import DecentralizedAverager
+from ..averaging.control import StepControl
+from ..dht import DHT
+from ..utils.logging import get_logger
+from ..utils.timed_storage import DHTExpiration
+
+logger = get_logger(__name__)
+
+
+class GradientAverager(DecentralizedAverager):
+    def __init__(
+        self,
+        parameters: Iterable[torch.nn.Parameter],
+        dht: DHT,
+        *,
+        prefix: str,
+        reuse_grad_buffers: bool = False,
+        accumulate_grads_on: Optional[torch.device] = None,
+        client_mode: Optional[bool] = None,
+        warn: bool = True,
+        **kwargs,
+    ):
+        if reuse_grad_buffers and accumulate_grads_on is not None:
+            logger.warning("accumulate_grads_on is ignored with reuse_grad_buffers=True")
+        client_mode = client_mode if client_mode is not None else False
+        self.parameters = tuple(parameters)
+        self.reuse_grad_buffers = reuse_grad_buffers
+        self.warn = warn
+        self.local_samples_accumulated = 0
+        self.local_times_accumulated = 0
+        self._anchor_batch_size: Optional[int] = None
+        self._accumulators_used_in_step = False
+        self._new_averaged_grads = False
+
+        with torch.no_grad():
+            if reuse_grad_buffers:
+                self._local_accumulators = None
+            else:
+                self._local_accumulators = tuple(
+                    torch.zeros_like(param, device=accumulate_grads_on or param.device)
+                    for param in self.parameters
+                )
+            averaged_grads = tuple(torch.zeros_like(param) for param in self.parameters)
+        super().__init__(averaged_tensors=averaged_grads, dht=dht, prefix=prefix, client_mode=client_mode, **kwargs)
+
+    def _grads_from_parameters(self) -> Iterator[torch.Tensor]:
+        for param in self.parameters:
+            if param.grad is None:
+                param.grad = torch.zeros_like(param)
+            yield param.grad
+
+    @torch.no_grad()
+    def _grad_accumulators(self) -> Iterator[torch.Tensor]:
+        assert (self._local_accumulators is None) == self.reuse_grad_buffers
+        yield from self._grads_from_parameters() if self.reuse_grad_buffers else self._local_accumulators
+
+    @torch.no_grad()
+    def accumulate_grads_(self, batch_size: int):
+        """Add current autograd gradients into the accumulators (reference grad_averager.py:130-148).
+
+        With ``reuse_grad_buffers=True`` autograd already accumulates into
+        ``param.grad`` across micro-batches; only the sample counters advance.
+        """
+        if self._accumulators_used_in_step and self.warn:
+            logger.warning(
+                "gradient accumulators were not reset since the last averaging round; "
+                "call reset_accumulated_grads_() or use step(reset_accumulators=True)"
+            )
+            self._accumulators_used_in_step = False
+        if self._anchor_batch_size is None:
+            self._anchor_batch_size = batch_size
+        self.local_samples_accumulated += batch_size
+        self.local_times_accumulated += 1
+        if not self.reuse_grad_buffers:
+            # re-scale so batches of different sizes contribute proportionally
+            alpha = float(batch_size) / self._anchor_batch_size
+            for grad_buf, grad_acc in zip(self._grads_from_parameters(), self._grad_accumulators()):
+                grad_acc.add_(grad_buf.to(grad_acc.device, grad_acc.dtype), alpha=alpha)
+
+    def schedule_step(self, scheduled_time: Optional[DHTExpiration] = None, **kwargs) -> StepControl:
+        """Begin matchmaking early; the all-reduce starts on trigger
+        (reference grad_averager.py:155-162)."""
+        assert kwargs.get("weight") is None, "setting weight during schedule is not supported"
+        return super().step(scheduled_time=scheduled_time, wait=False, require_trigger=True, **kwargs)
+
+    def step(
+        self,
+        weight: Optional[float] = None,
+        reset_accumulators: bool = True,
+        control: Optional[StepControl] = None,
+        timeout: Optional[float] = None,
+        wait: bool = True,
+        **kwargs,
+    ):
+        """Average accumulated gradients with the group (reference grad_averager.py:163-201)."""
+        if control is None:
+            control = self.schedule_step(timeout=timeout, **kwargs)
+        elif len(kwargs) > 0:
+            raise RuntimeError(f"kwargs {kwargs} have no effect with a pre-scheduled control")
+        assert not control.triggered, "this step control was already triggered"
+        if self._new_averaged_grads and self.warn:
+            logger.warning("starting a new round but the previous averaged gradients were never used")
+
+        self.load_accumulators_into_averager_()
+        self._accumulators_used_in_step = True
+        self._new_averaged_grads = True
+
+        control.weight = float(self.local_samples_accumulated) if weight is None else weight
+        if reset_accumulators:
+            self.reset_accumulated_grads_()
+        control.allow_allreduce()
+        return control.result(timeout) if wait else control
+
+    @torch.no_grad()
+    def load_accumulators_into_averager_(self):
+        """averaged_grad <- accumulator / times_accumulated (reference grad_averager.py:204-212)."""
+        grad_scale = (1.0 / self.local_times_accumulated) if self.local_times_accumulated else 0.0
+        with self.get_tensors() as averaged_grads:
+            for grad_acc, averaged_grad in zip(self._grad_accumulators(), averaged_grads):
+                averaged_grad.copy_(grad_acc.to(averaged_grad.device, averaged_grad.dtype), non_blocking=True)
+                averaged_grad.mul_(grad_scale)
+
+    @torch.no_grad()
+    def reset_accumulated_grads_(self):
+        """Zero the accumulators for the next round (reference grad_averager.py:214-221)."""
+        self._accumulators_used_in_step = False
+        self.local_samples_accumulated = self.local_times_accumulated = 0
+        self._anchor_batch_size = None
+        for grad_buf in self._grad_accumulators():
+            grad_buf.zero_()
+
+    @contextlib.contextmanager
+    @torch.no_grad()
+    def use_averaged_gradients(self):
+        """Temporarily substitute param.grad with the averaged gradients
+        (reference grad_averager.py:223-239)."""
+        self._new_averaged_grads = False
+        with self.get_tensors() as averaged_grads:
+            assert len(averaged_grads) == len(self.parameters)
+            old_grads = [param.grad for param in self.parameters]
+            try:
+                for param, new_grad in zip(self.parameters, averaged_grads):
+                    param.grad = new_grad.to(param.device, param.dtype)
+                yield averaged_grads
+            finally:
+                for param, old_grad in zip(self.parameters, old_grads):
+                    param.grad = old_grad
+
+    def notify_used_averaged_gradients(self):
+        """The previous round's results were consumed (clears the warning flag)."""
+        self._new_averaged_grads = False

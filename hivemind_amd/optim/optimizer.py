@@ -1,0 +1,509 @@
+"""hivemind_amd.Optimizer -- the drop-in decentralized optimizer.
+
+Parity target: reference ``hivemind/optim/optimizer.py:32-790``. Semantics
+preserved:
+
+* peers accumulate gradients locally until the swarm jointly reaches
+  ``target_batch_size`` (tracked via ProgressTracker records in the DHT), then
+  run one "global step": average gradients with the group, run the inner
+  optimizer, optionally average parameters/statistics, advance the epoch;
+* training is epoch-synchronized, not step-synchronized: hyperparameters stay
+  invariant to swarm size and peers may join/leave freely;
+* ``delay_optimizer_step``/``delay_grad_averaging`` (DPU) overlap averaging
+  and the optimizer step with subsequent forward/backward passes;
+* ``use_local_updates`` runs the inner optimizer every step and only averages
+  parameters periodically (local-SGD mode);
+* peers that fall behind the swarm epoch download the latest state from a
+  donor (``load_state_from_peers``).
+
+On one MI355X node, each of the 8 GPUs is one peer; gradient averaging rides
+the RCCL-over-xGMI data plane automatically (averaging/rccl.py).
+"""
+
+from __future__ import annotations
+
+import logging
+import os
+import time
+from functools import partial
+from typing import Any, Callable, Optional, Sequence, Union
+
+import torch
+
+from ..averaging.control import AveragingStage, StepControl
+from ..compression import CompressionBase, NoCompression
+from ..dht import DHT
+from ..utils.logging import get_logger
+from ..utils.timed_storage import DHTExpiration, get_dht_time
+from .grad_averager import GradientAverager
+from .grad_scaler import GradScaler
+from .progress_tracker import ProgressTracker
+from .state_averager import TrainingStateAverager
+
+logger = get_logger(__name__)
+
+
+class Optimizer(torch.optim.Optimizer):
+    """A torch.optim.Optimizer-compatible decentralized optimizer."""
+
+    def __init__(
+        self,
+        *,
+        dht: DHT,
+        run_id: str,
+        target_batch_size: int,
+        batch_size_per_step: Optional[int] = None,
+        optimizer: Union[torch.optim.Optimizer, Callable[[Any], torch.optim.Optimizer]],
+        params: Optional[Any] = None,
+        scheduler: Optional[Any] = None,
+        matchmaking_time: float = 5.0,
+        averaging_timeout: float = 60.0,
+        allreduce_timeout: Optional[float] = None,
+        next_chunk_timeout: Optional[float] = None,
+        load_state_timeout: float = 600.0,
+        reuse_grad_buffers: bool = False,
+        offload_optimizer: Optional[bool] = None,
+        delay_optimizer_step: Optional[bool] = None,
+        delay_grad_averaging: bool = False,
+        delay_state_averaging: bool = True,
+        average_state_every: int = 1,
+        use_local_updates: bool = False,
+        client_mode: bool = False,
+        auxiliary: bool = False,
+        grad_compression: CompressionBase = NoCompression(),
+        grad_averager_factory: Optional[Callable[..., GradientAverager]] = None,
+        state_averaging_compression: CompressionBase = NoCompression(),
+        load_state_compression: CompressionBase = NoCompression(),
+        average_opt_statistics: Sequence[str] = (),
+        extra_tensors: Sequence[torch.Tensor] = (),
+        averager_opts: Optional[dict] = None,
+        tracker_opts: Optional[dict] = None,
+        performance_ema_alpha: float = 0.1,
+        shutdown_timeout: float = 5.0,
+        verbose: bool = False,
+    ):
+        self._parent_pid = os.getpid()
+        client_mode = client_mode if client_mode is not None else False
+        delay_optimizer_step = delay_optimizer_step if delay_optimizer_step is not None else delay_grad_averaging
+        offload_optimizer = offload_optimizer if offload_optimizer is not None else (params is not None and not use_local_updates)
+        assert not delay_grad_averaging or delay_optimizer_step, "delay_grad_averaging requires delay_optimizer_step"
+        assert not (client_mode and auxiliary), "auxiliary peers must be able to accept connections"
+        if auxiliary:
+            assert batch_size_per_step is None, "auxiliary peers should not accumulate batches"
+
+        self.dht, self.run_id = dht, run_id
+        self.batch_size_per_step, self.target_batch_size = batch_size_per_step, target_batch_size
+        self.matchmaking_time, self.offload_optimizer = matchmaking_time, offload_optimizer
+        self.delay_state_averaging, self.average_state_every = delay_state_averaging, average_state_every
+        self.delay_grad_averaging, self.delay_optimizer_step = delay_grad_averaging, delay_optimizer_step
+        self.averaging_timeout = averaging_timeout
+        self.allreduce_timeout = allreduce_timeout if allreduce_timeout is not None else averaging_timeout
+        self.load_state_timeout = load_state_timeout
+        self.shutdown_timeout = shutdown_timeout
+        self.next_chunk_timeout = next_chunk_timeout
+
+        self.status_loglevel = logging.INFO if verbose else logging.DEBUG
+        self.scheduled_grads: Optional[StepControl] = None
+        self.scheduled_state: Optional[StepControl] = None
+        self.auxiliary, self.client_mode, self.use_local_updates = auxiliary, client_mode, use_local_updates
+
+        self.tracker = self._make_progress_tracker(
+            target_batch_size, performance_ema_alpha=performance_ema_alpha, **(tracker_opts or {})
+        )
+        self.state_averager = self._make_state_averager(
+            optimizer=optimizer,
+            params=params,
+            scheduler=scheduler,
+            offload_optimizer=offload_optimizer,
+            custom_gradients=reuse_grad_buffers,
+            performance_ema_alpha=performance_ema_alpha,
+            compression=state_averaging_compression,
+            state_compression=load_state_compression,
+            average_opt_statistics=average_opt_statistics,
+            extra_tensors=extra_tensors,
+            **(averager_opts or {}),
+        )
+        if not use_local_updates:
+            self.grad_averager: Optional[GradientAverager] = self._make_gradient_averager(
+                grad_averager_factory, reuse_grad_buffers=reuse_grad_buffers, compression=grad_compression,
+                **(averager_opts or {}),
+            )
+        else:
+            self.grad_averager = None
+
+        self._should_check_synchronization_on_update = True
+        self._schema_hash = self._compute_schema_hash()
+        self.delay_before_state_averaging = _SimpleEMA(alpha=performance_ema_alpha)
+        # averaging-related statistics for pre-scheduling
+        self._step_supports_amp_scaling = reuse_grad_buffers  # for torch.amp compat
+
+    def _make_state_averager(self, **kwargs) -> TrainingStateAverager:
+        return TrainingStateAverager(
+            dht=self.dht,
+            prefix=f"{self.run_id}_state_averager",
+            min_matchmaking_time=self.matchmaking_time,
+            allreduce_timeout=self.allreduce_timeout,
+            shutdown_timeout=self.shutdown_timeout,
+            client_mode=self.client_mode,
+            auxiliary=self.auxiliary,
+            start=True,
+            **kwargs,
+        )
+
+    def _make_gradient_averager(self, factory: Optional[Callable], **kwargs) -> GradientAverager:
+        factory = factory if factory is not None else GradientAverager
+        grad_averager = factory(
+            dht=self.dht,
+            prefix=f"{self.run_id}_grad_averager",
+            parameters=self.state_averager.main_parameters,
+            min_matchmaking_time=self.matchmaking_time,
+            allreduce_timeout=self.allreduce_timeout,
+            shutdown_timeout=self.shutdown_timeout,
+            client_mode=self.client_mode,
+            auxiliary=self.auxiliary,
+            start=True,
+            **kwargs,
+        )
+        if self.offload_optimizer:
+            optimized_param_groups = self.state_averager.optimizer.param_groups
+            optimized_parameters = [p for group in optimized_param_groups for p in group["params"]]
+            with grad_averager.get_tensors() as averaged_gradients:
+                assert len(averaged_gradients) == len(optimized_parameters)
+                for opt_param, averaged_grad in zip(optimized_parameters, averaged_gradients):
+                    opt_param.grad = averaged_grad
+        return grad_averager
+
+    def _make_progress_tracker(self, target_batch_size: int, **kwargs) -> ProgressTracker:
+        return ProgressTracker(
+            dht=self.dht,
+            prefix=self.run_id,
+            target_batch_size=target_batch_size,
+            client_mode=self.client_mode,
+            status_loglevel=self.status_loglevel,
+            start=True,
+            **kwargs,
+        )
+
+    def _compute_schema_hash(self) -> int:
+        import hashlib
+
+        parameters = self.state_averager.main_parameters
+        param_shapes = tuple(tuple(p.shape) for p in parameters)
+        grad_ids = None if self.use_local_updates else tuple(id(p) for p in parameters)
+        return hash((grad_ids is None, param_shapes))
+
+    # ----------------------------------------------------------- properties
+
+    @property
+    def local_epoch(self) -> int:
+        """This peer's current epoch (different from tracker.global_epoch while catching up)."""
+        return self.state_averager.local_epoch
+
+    @property
+    def local_progress(self):
+        return self.tracker.local_progress
+
+    @property
+    def use_gradient_averaging(self) -> bool:
+        return self.grad_averager is not None
+
+    @property
+    def param_groups(self):
+        return self.state_averager.optimizer.param_groups
+
+    @param_groups.setter
+    def param_groups(self, value):
+        pass  # managed by the inner optimizer
+
+    @property
+    def state(self):
+        return self.state_averager.optimizer.state
+
+    @property
+    def defaults(self):
+        return getattr(self.state_averager.optimizer, "defaults", {})
+
+    @property
+    def opt(self) -> torch.optim.Optimizer:
+        return self.state_averager.optimizer
+
+    # ----------------------------------------------------------------- step
+
+    def step(
+        self,
+        closure: Optional[Callable[[], torch.Tensor]] = None,
+        batch_size: Optional[int] = None,
+        grad_scaler: Optional[GradScaler] = None,
+    ):
+        """Report progress; average and run the inner optimizer when the swarm
+        collectively reaches target_batch_size (reference optimizer.py:369-436)."""
+        if grad_scaler is not None and not isinstance(grad_scaler, GradScaler):
+            raise ValueError("hivemind_amd.Optimizer requires hivemind_amd.GradScaler")
+        if self.batch_size_per_step is None and batch_size is None and not self.auxiliary:
+            raise ValueError("supply either batch_size_per_step in __init__ or batch_size in step()")
+        if self.auxiliary and (closure is not None or batch_size is not None or grad_scaler is not None):
+            raise ValueError("auxiliary peers should not have batch size, closure or grad_scaler")
+        batch_size = batch_size if batch_size is not None else self.batch_size_per_step
+
+        loss = None
+        if closure is not None:
+            with torch.enable_grad():
+                loss = closure()
+
+        if not self.auxiliary and self._should_load_state_from_peers():
+            logger.log(self.status_loglevel, "peer is out of sync; downloading the latest state")
+            self.load_state_from_peers()
+            return loss
+        if grad_scaler is not None and not grad_scaler.are_grads_finite(self, use_cached=True):
+            logger.log(self.status_loglevel, "gradients are inf/nan: resetting accumulators")
+            if self.grad_averager is not None:
+                self.grad_averager.reset_accumulated_grads_()
+            self.tracker.report_local_progress(self.local_epoch, samples_accumulated=0)
+            self._schedule_if_needed()
+            return loss
+
+        if not self.auxiliary:
+            if self.use_gradient_averaging:
+                self.grad_averager.accumulate_grads_(batch_size)
+            new_samples = self.tracker.local_progress.samples_accumulated + batch_size
+            self.tracker.report_local_progress(self.local_epoch, new_samples)
+            self._schedule_if_needed()
+
+        if self.use_local_updates and not self.auxiliary:
+            # local-SGD mode: run the inner optimizer every step, average params on epoch
+            self.state_averager.step(optimizer_step=True, zero_grad=False, grad_scaler=grad_scaler)
+
+        if self.tracker.ready_to_update_epoch or self.auxiliary:
+            if self.auxiliary:
+                # aux peers just assist in averaging
+                self._update_global_epoch(grad_scaler=None)
+            else:
+                self._update_global_epoch(grad_scaler=grad_scaler)
+        return loss
+
+    def _schedule_if_needed(self):
+        """Pre-schedule averaging rounds near the end of an epoch
+        (reference optimizer.py:559-592)."""
+        if self.use_gradient_averaging and self.scheduled_grads is None:
+            eta_seconds = self.tracker.estimated_next_update_time - get_dht_time()
+            if eta_seconds <= self.matchmaking_time:
+                self.scheduled_grads = self.grad_averager.schedule_step(
+                    timeout=self.averaging_timeout,
+                    scheduled_time=max(get_dht_time() + max(0.0, eta_seconds), get_dht_time() + 0.5),
+                )
+        if self.scheduled_state is None and self._should_average_state():
+            eta_seconds = self.tracker.estimated_next_update_time - get_dht_time()
+            if eta_seconds <= self.matchmaking_time:
+                self.scheduled_state = self.state_averager.schedule_step(
+                    timeout=self.averaging_timeout,
+                    scheduled_time=max(get_dht_time() + max(0.0, eta_seconds), get_dht_time() + 0.5),
+                )
+
+    def _should_average_state(self) -> bool:
+        next_epoch = max(self.local_epoch + 1, self.tracker.global_epoch)
+        return next_epoch % self.average_state_every == 0
+
+    def _update_global_epoch(self, grad_scaler: Optional[GradScaler]):
+        """The global step: average grads, run the optimizer, advance the epoch
+        (reference optimizer.py:438-509)."""
+        assert self._parent_pid == os.getpid()
+        _epoch_start = time.perf_counter()
+        began_averaging_gradients = False
+        if self.use_gradient_averaging and not self.auxiliary:
+            began_averaging_gradients = self._begin_averaging_gradients()
+            if not began_averaging_gradients:
+                pass  # failed to average; apply local gradients anyway
+
+        next_epoch = max(self.local_epoch + 1, self.tracker.global_epoch)
+        swarm_not_empty = self.tracker.global_progress.num_peers > 1
+        should_perform_optimizer_step = not self.auxiliary and not self.use_local_updates
+        should_average_state = (
+            swarm_not_empty
+            and next_epoch % self.average_state_every == 0
+        )
+
+        if should_average_state and self.scheduled_state is not None:
+            if self.scheduled_state.triggered or self.scheduled_state.done():
+                self.scheduled_state = None
+        averaging_control = self.scheduled_state if should_average_state else None
+        if self.scheduled_state is not None and not should_average_state:
+            self.scheduled_state.cancel()
+            self.scheduled_state = None
+
+        if self.auxiliary:
+            # assist gradient averaging rounds of others, then return
+            if self.grad_averager is not None:
+                try:
+                    self.grad_averager.step(timeout=self.averaging_timeout, wait=True)
+                except Exception as e:
+                    logger.debug(f"aux averaging assist failed: {e!r}")
+            self.state_averager.local_epoch = self.tracker.global_epoch
+            self.tracker.update_epoch(self.tracker.global_epoch)
+            return
+
+        if self.use_gradient_averaging:
+            if not self.delay_grad_averaging:
+                self._average_gradients_and_load_into_optimizer(self.scheduled_grads)
+
+        self.state_averager.step(
+            increment_epoch=True,
+            wait_for_trigger=(partial(self._average_gradients_and_load_into_optimizer, self.scheduled_grads)
+                              if self.use_gradient_averaging and self.delay_grad_averaging else None),
+            optimizer_step=should_perform_optimizer_step,
+            delay_optimizer_step=self.delay_optimizer_step and should_perform_optimizer_step,
+            grad_scaler=grad_scaler,
+            averaging_round=should_average_state,
+            delay_averaging=self.delay_state_averaging and not self.auxiliary,
+            averaging_control=averaging_control,
+            averaging_opts=dict(timeout=self.averaging_timeout) if should_average_state else None,
+            zero_grad=True,
+        )
+        self.scheduled_state = None
+
+        self.tracker.update_epoch(new_epoch=self.state_averager.local_epoch)
+        self._should_check_synchronization_on_update = True
+
+        if not self.client_mode:
+            self.state_averager.allow_state_sharing = True
+        logger.log(self.status_loglevel, f"transitioning to epoch {self.local_epoch} "
+                   f"({time.perf_counter() - _epoch_start:.2f}s)")
+
+    def _begin_averaging_gradients(self) -> bool:
+        """Start (or join) this epoch's gradient averaging round
+        (reference optimizer.py:511-544)."""
+        began = False
+        if self.scheduled_grads is not None and (self.scheduled_grads.triggered or self.scheduled_grads.done()):
+            self.scheduled_grads = None
+        try:
+            self.scheduled_grads = self.grad_averager.step(
+                control=self.scheduled_grads, reset_accumulators=True, wait=False,
+                timeout=None if self.scheduled_grads is not None else self.averaging_timeout,
+            )
+            began = True
+        except Exception as e:
+            logger.log(self.status_loglevel, f"failed to begin gradient averaging: {e!r}")
+            self.scheduled_grads = None
+        return began
+
+    def _average_gradients_and_load_into_optimizer(self, maybe_step_control: Optional[StepControl]):
+        """Wait for the averaging round and put results into the optimizer
+        (reference optimizer.py:593-624)."""
+        assert self.use_gradient_averaging
+        averaged = False
+        if maybe_step_control is not None:
+            try:
+                maybe_step_control.result(self.averaging_timeout)
+                averaged = True
+            except Exception as e:
+                logger.log(self.status_loglevel, f"gradient averaging failed: {e!r}; using local gradients")
+        self.scheduled_grads = None
+        self._load_averaged_gradients_into_optimizer_()
+
+    def _load_averaged_gradients_into_optimizer_(self):
+        optimized_parameters = [p for group in self.state_averager.optimizer.param_groups for p in group["params"]]
+        if self.offload_optimizer:
+            pass  # grads are bound to the averager's buffers already (see _make_gradient_averager)
+        else:
+            with self.grad_averager.get_tensors() as averaged_gradients:
+                for opt_param, averaged_grad in zip(optimized_parameters, averaged_gradients):
+                    opt_param.grad = averaged_grad.to(opt_param.device, opt_param.dtype)
+        self.grad_averager.notify_used_averaged_gradients()
+
+    # --------------------------------------------------------------- re-sync
+
+    def _should_load_state_from_peers(self) -> bool:
+        """True if the swarm advanced 2+ epochs past us (reference optimizer.py:655-673)."""
+        if self._should_check_synchronization_on_update and self.tracker.fetched_global_progress_this_epoch.is_set():
+            self._should_check_synchronization_on_update = False
+            return self.local_epoch < self.tracker.global_epoch - 1
+        return self.local_epoch < self.tracker.global_epoch - 1
+
+    def is_synchronized_with_peers(self) -> bool:
+        return self.local_epoch >= self.tracker.global_epoch - 1
+
+    def load_state_from_peers(self, **kwargs):
+        """Download the latest state from a donor peer (reference optimizer.py:679-717)."""
+        self._finish_scheduled_averaging()
+        with self.state_averager.lock_averaged_tensors if False else _nullcontext():
+            loaded = self.state_averager.load_state_from_peers(timeout=self.load_state_timeout, **kwargs)
+        if loaded is None:
+            # nobody to download from: fast-forward to the global epoch anyway
+            self.state_averager.local_epoch = self.tracker.global_epoch
+        if self.grad_averager is not None:
+            self.grad_averager.reset_accumulated_grads_()
+        self.tracker.report_local_progress(local_epoch=self.local_epoch, samples_accumulated=0)
+
+    def _finish_scheduled_averaging(self):
+        for scheduled in (self.scheduled_grads, self.scheduled_state):
+            if scheduled is not None and not scheduled.done():
+                scheduled.cancel()
+        self.scheduled_grads = self.scheduled_state = None
+
+    # ------------------------------------------------- torch optimizer compat
+
+    def zero_grad(self, set_to_none: bool = True):
+        if self.use_gradient_averaging and self.grad_averager.reuse_grad_buffers:
+            raise ValueError(
+                "with reuse_grad_buffers=True, gradients are reset automatically at the global step; "
+                "do not call zero_grad manually"
+            )
+        for group in self.param_groups:
+            for param in group["params"]:
+                if param.grad is None:
+                    continue
+                if set_to_none:
+                    param.grad = None
+                else:
+                    param.grad.zero_()
+
+    def state_dict(self) -> dict:
+        state_dict = self.state_averager.optimizer.state_dict()
+        state_dict["state"]["local_epoch"] = self.local_epoch
+        return state_dict
+
+    def load_state_dict(self, state_dict: dict):
+        if "local_epoch" in state_dict.get("state", {}):
+            self.state_averager.local_epoch = state_dict["state"].pop("local_epoch")
+        return self.state_averager.optimizer.load_state_dict(state_dict)
+
+    def add_param_group(self, param_group: dict):
+        raise NotImplementedError("add_param_group is not supported: parameters are fixed by the run's schema")
+
+    def __repr__(self):
+        return f"{self.__class__.__name__}(run_id={self.run_id}, epoch={self.local_epoch})"
+
+    def shutdown(self):
+        logger.log(self.status_loglevel, "shutting down optimizer")
+        self._finish_scheduled_averaging()
+        self.tracker.shutdown(self.shutdown_timeout)
+        self.state_averager.step(wait_for_delayed_updates=True)
+        for averager in [self.grad_averager, self.state_averager]:
+            if averager is not None:
+                try:
+                    averager.shutdown()
+                except Exception:
+                    pass
+
+    def __del__(self):
+        if getattr(self, "_parent_pid", None) == os.getpid():
+            try:
+                self.shutdown()
+            except Exception:
+                pass
+
+
+class _SimpleEMA:
+    def __init__(self, alpha: float):
+        self.alpha, self.value = alpha, 0.0
+
+    def update(self, x: float) -> float:
+        self.value = self.alpha * x + (1 - self.alpha) * self.value
+        return self.value
+
+
+class _nullcontext:
+    def __enter__(self):
+        return None
+
+    def __exit__(self, *args):
+        return False
