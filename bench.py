@@ -1,0 +1,189 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: ALBERT-base MLM training with hivemind_amd.Optimizer.
+
+Measures the BASELINE.json headline metric -- samples/sec over the whole swarm
+for collaborative ALBERT training (reference: examples/albert, ~20.9
+samples/s/peer on 1080 Ti-class GPUs, target_batch_size 4096) -- on N MI355X
+GPUs of one node, one process per GPU over RCCL/xGMI.
+
+Usage:  python bench.py [--gpus N] [--steps K] [--warmup W]
+For N>1 the driver launches this via torch.distributed.run with one rank per
+GPU; ranks read RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the environment.
+
+Synthetic data (random token ids of the benchmark shape), random-init weights.
+"""
+
+import argparse
+import json
+import os
+import sys
+import time
+
+import torch
+
+BASELINE_SAMPLES_PER_SEC_PER_PEER = 20.9  # ALBERT collaborative trainer
+
+
+def log(msg):
+    print(f"[bench rank{os.environ.get('RANK', '0')}] {msg}", file=sys.stderr, flush=True)
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--gpus", type=int, default=1)
+    parser.add_argument("--steps", type=int, default=24)
+    parser.add_argument("--warmup", type=int, default=12)
+    parser.add_argument("--batch", type=int, default=32, help="per-GPU samples per step")
+    parser.add_argument("--seq-len", type=int, default=512)
+    parser.add_argument("--target-batch-size", type=int, default=4096)
+    parser.add_argument("--model", type=str, default="albert-base", choices=["albert-base", "albert-large", "tiny"])
+    args = parser.parse_args()
+
+    world_size = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    n_gpus = world_size if world_size > 1 else args.gpus
+    use_gpu = torch.cuda.is_available()
+
+    import torch.distributed as dist
+
+    if world_size > 1:
+        backend = "nccl" if use_gpu else "gloo"
+        dist.init_process_group(backend, rank=rank, world_size=world_size)
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+        device = torch.device("cuda", local_rank)
+    else:
+        device = torch.device("cpu")
+
+    from hivemind_amd import DHT, Optimizer
+    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM
+    from hivemind_amd.ops import FusedAdamW, hip_available
+
+    if use_gpu and not hip_available():
+        raise RuntimeError("HIP ops extension missing on a GPU node -- refusing to run a fallback bench")
+
+    if args.model == "albert-base":
+        config = AlbertConfig.base()
+    elif args.model == "albert-large":
+        config = AlbertConfig.large()
+    else:
+        config = AlbertConfig.tiny()
+    if not use_gpu:
+        config.dtype = torch.float32  # CPU bf16 matmuls are pathologically slow
+
+    torch.manual_seed(1234 + rank)
+    model = AlbertForMaskedLM(config).to(device)
+    log(f"model: {args.model}, {model.num_parameters()/1e6:.1f}M params, device={device}")
+
+    # ---------------------------------------------------------------- swarm
+    if world_size > 1:
+        if rank == 0:
+            dht = DHT(start=True)
+            endpoint = [dht.endpoint]
+        else:
+            dht, endpoint = None, [None]
+        dist.broadcast_object_list(endpoint, src=0)
+        if rank != 0:
+            dht = DHT(initial_peers=[endpoint[0]], start=True)
+        dist.barrier()
+    else:
+        dht = DHT(start=True)
+
+    opt = Optimizer(
+        dht=dht,
+        run_id="albert_bench",
+        target_batch_size=args.target_batch_size,
+        batch_size_per_step=args.batch,
+        optimizer=lambda param_groups: FusedAdamW(param_groups, lr=2e-3, weight_decay=0.01),
+        params=[{"params": list(model.parameters())}],
+        offload_optimizer=True,
+        matchmaking_time=1.0 if world_size > 1 else 0.5,
+        averaging_timeout=120.0,
+        reuse_grad_buffers=False,
+        average_state_every=4,
+        averager_opts=dict(
+            request_timeout=0.5,
+            min_group_size=min(2, n_gpus),
+            target_group_size=n_gpus if n_gpus > 1 else None,
+        ),
+        tracker_opts=dict(min_refresh_period=0.2, default_refresh_period=0.5, max_refresh_period=2.0),
+        verbose=rank == 0,
+    )
+
+    def make_batch():
+        input_ids = torch.randint(0, config.vocab_size, (args.batch, args.seq_len), device=device)
+        labels = input_ids.clone()
+        # mask 15% of positions for MLM, rest ignored in the loss
+        mask = torch.rand(labels.shape, device=device) > 0.15
+        labels[mask] = -100
+        return input_ids, labels
+
+    def one_step():
+        input_ids, labels = make_batch()
+        loss, _ = model(input_ids, labels=labels)
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+        return loss
+
+    def sync():
+        if use_gpu:
+            torch.cuda.synchronize(device)
+        if world_size > 1:
+            dist.barrier()
+
+    log(f"warmup: {args.warmup} steps")
+    for i in range(args.warmup):
+        one_step()
+    sync()
+
+    log(f"timing: {args.steps} steps")
+    t0 = time.perf_counter()
+    for i in range(args.steps):
+        one_step()
+    sync()
+    elapsed = time.perf_counter() - t0
+
+    # take the max elapsed across ranks (slowest peer defines the swarm rate)
+    if world_size > 1:
+        t = torch.tensor([elapsed], dtype=torch.float64, device=device if use_gpu else "cpu")
+        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    total_samples = args.steps * args.batch * n_gpus
+    samples_per_sec = total_samples / elapsed
+    ms_per_step = elapsed / args.steps * 1000.0
+
+    if rank == 0:
+        result = {
+            "metric": "samples/sec (whole swarm) ALBERT-base hivemind.Optimizer",
+            "value": round(samples_per_sec, 2),
+            "unit": "samples/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_per_step, 2),
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": round(samples_per_sec / (BASELINE_SAMPLES_PER_SEC_PER_PEER * n_gpus), 2),
+            "dtype": "bf16" if use_gpu else "fp32",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.target_batch_size,
+                "per_gpu_batch": args.batch,
+                "seq_len": args.seq_len,
+                "parallelism": f"dp{n_gpus}",
+            },
+        }
+        print(json.dumps(result), flush=True)
+
+    opt.shutdown()
+    dht.shutdown()
+    if world_size > 1:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,238 @@
+"""hivemind_amd.ops: hand-written CDNA4 HIP kernels + torch reference impls.
+
+GPU tensors dispatch to the in-tree ``hivemind_amd._hip_ops`` extension
+(built by setup.py with PYTORCH_ROCM_ARCH=gfx950). If a GPU tensor arrives
+and the extension is missing, ops raise immediately -- there is no silent
+eager fallback on the GPU path. CPU tensors use the torch reference
+implementations (which are also the numerics oracle for the GPU tests).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional, Tuple
+
+import torch
+
+from ..utils.logging import get_logger
+
+logger = get_logger(__name__)
+
+_hip_ops = None
+_hip_import_error: Optional[BaseException] = None
+try:
+    from hivemind_amd import _hip_ops  # type: ignore[no-redef]
+except BaseException as e:  # pragma: no cover
+    _hip_import_error = e
+
+
+def hip_ops():
+    """The extension module; raises loudly if unavailable (never silently falls back)."""
+    if _hip_ops is None:
+        raise RuntimeError(
+            "hivemind_amd._hip_ops extension is not built -- run "
+            "`PYTORCH_ROCM_ARCH=gfx950 python setup.py build_ext --inplace` "
+            f"(import error: {_hip_import_error!r})"
+        )
+    return _hip_ops
+
+
+def hip_available() -> bool:
+    return _hip_ops is not None
+
+
+# ---------------------------------------------------------------------------
+# fused layernorm (optionally fused residual add)
+# ---------------------------------------------------------------------------
+
+
+class _FusedLayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, residual, gamma, beta, eps):
+        outs = hip_ops().layernorm_fwd(x.contiguous(), residual.contiguous() if residual is not None else None,
+                                       gamma.contiguous(), beta.contiguous(), eps)
+        if residual is not None:
+            y, mean, rstd, h = outs
+        else:
+            y, mean, rstd = outs
+            h = x
+        ctx.save_for_backward(h, gamma, mean, rstd)
+        ctx.has_residual = residual is not None
+        return (y, h) if residual is not None else (y, x)
+
+    @staticmethod
+    def backward(ctx, dy, dh_extra):
+        h, gamma, mean, rstd = ctx.saved_tensors
+        dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy.contiguous(), h.contiguous(), gamma, mean, rstd)
+        if dh_extra is not None and dh_extra.abs().sum() != 0:  # pragma: no cover - rare path
+            dx = dx + dh_extra
+        # d(x) and d(residual) are identical for the fused residual add
+        dres = dx if ctx.has_residual else None
+        return dx, dres, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
+
+
+def fused_layernorm(
+    x: torch.Tensor,
+    gamma: torch.Tensor,
+    beta: torch.Tensor,
+    residual: Optional[torch.Tensor] = None,
+    eps: float = 1e-12,
+) -> Tuple[torch.Tensor, torch.Tensor]:
+    """y = LayerNorm(x + residual); returns (y, x + residual).
+
+    GPU: one fused HIP kernel (wave-per-row, bf16). CPU: torch reference.
+    """
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _FusedLayerNorm.apply(x, residual, gamma, beta, eps)
+    h = x + residual if residual is not None else x
+    y = torch.nn.functional.layer_norm(h.float(), (h.shape[-1],), gamma.float(), beta.float(), eps).to(x.dtype)
+    return y, h
+
+
+class _FusedBiasGelu(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, bias):
+        out, pre_act = hip_ops().bias_gelu_fwd(x.contiguous(), bias.contiguous(), True)
+        ctx.save_for_backward(pre_act)
+        ctx.bias_dtype = bias.dtype
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        (pre_act,) = ctx.saved_tensors
+        dx, dbias = hip_ops().bias_gelu_bwd(dy.contiguous(), pre_act)
+        return dx, dbias.to(ctx.bias_dtype)
+
+
+def fused_bias_gelu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
+    """gelu_tanh(x + bias) -- the reference's gelu_fast (layers/common.py:10-16)."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _FusedBiasGelu.apply(x, bias)
+    return torch.nn.functional.gelu((x.float() + bias.float()), approximate="tanh").to(x.dtype)
+
+
+# ---------------------------------------------------------------------------
+# averaging primitives
+# ---------------------------------------------------------------------------
+
+
+@torch.no_grad()
+def apply_delta_(tensor: torch.Tensor, delta: torch.Tensor, alpha: float = 1.0):
+    """tensor += alpha * delta."""
+    if tensor.is_cuda and tensor.dtype == delta.dtype and tensor.dtype in (torch.bfloat16, torch.float32):
+        hip_ops().apply_delta_(tensor, delta.contiguous(), float(alpha))
+    else:
+        tensor.add_(delta.to(tensor.device, tensor.dtype), alpha=alpha)
+    return tensor
+
+
+@torch.no_grad()
+def weighted_accumulate_(acc: torch.Tensor, x: torch.Tensor, w: float = 1.0):
+    """acc += w * x."""
+    if acc.is_cuda and acc.dtype == torch.float32 and x.dtype in (torch.float32, torch.bfloat16):
+        hip_ops().weighted_accumulate_(acc, x.contiguous(), float(w))
+    else:
+        acc.add_(x.to(acc.device, acc.dtype), alpha=w)
+    return acc
+
+
+# ---------------------------------------------------------------------------
+# codecs (GPU fast path used by the compression layer)
+# ---------------------------------------------------------------------------
+
+
+@torch.no_grad()
+def compress_fp16(t: torch.Tensor) -> torch.Tensor:
+    if t.is_cuda:
+        return hip_ops().compress_fp16(t.contiguous())
+    return t.float().clamp_(-65504.0, 65504.0).half()
+
+
+@torch.no_grad()
+def decompress_fp16(t: torch.Tensor) -> torch.Tensor:
+    if t.is_cuda:
+        return hip_ops().decompress_fp16(t.contiguous())
+    return t.float()
+
+
+@torch.no_grad()
+def quantize_blockwise(t: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
+    """(int8 codes, per-4096-block absmax) -- SURVEY K7."""
+    if t.is_cuda:
+        q, absmax = hip_ops().quantize_blockwise(t.contiguous())
+        return q, absmax
+    flat = t.detach().float().flatten()
+    n = flat.numel()
+    blocks = (n + 4095) // 4096
+    padded = torch.zeros(blocks * 4096)
+    padded[:n] = flat
+    view = padded.view(blocks, 4096)
+    absmax = view.abs().amax(1)
+    scale = torch.clamp_min(absmax / 127.0, torch.finfo(torch.float32).eps)
+    q = torch.round(view / scale.unsqueeze(1)).clamp_(-127, 127).to(torch.int8).flatten()[:n]
+    return q, absmax
+
+
+@torch.no_grad()
+def dequantize_blockwise(q: torch.Tensor, absmax: torch.Tensor) -> torch.Tensor:
+    if q.is_cuda:
+        return hip_ops().dequantize_blockwise(q.contiguous(), absmax.contiguous())
+    n = q.numel()
+    idx = torch.arange(n) // 4096
+    return q.float() * (absmax[idx] / 127.0)
+
+
+# ---------------------------------------------------------------------------
+# fused AdamW
+# ---------------------------------------------------------------------------
+
+
+class FusedAdamW(torch.optim.Optimizer):
+    """AdamW whose step is one HIP kernel per parameter on the GPU.
+
+    Master params must be fp32; an optional bf16 mirror (the live model param)
+    is refreshed inside the same kernel -- this is the MI355X expression of the
+    reference's "offloaded optimizer + parameter copy-back"
+    (state_averager.py:515-536) with zero extra memory traffic.
+    """
+
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-8, weight_decay=0.01, mirrors=None):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self._mirrors = mirrors or {}
+
+    def set_mirror(self, param: torch.Tensor, mirror: torch.Tensor):
+        self._mirrors[param] = mirror
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                state["step"] += 1
+                mirror = self._mirrors.get(p)
+                if p.is_cuda and p.dtype == torch.float32:
+                    hip_ops().fused_adamw_(
+                        p.data, p.grad.contiguous(), state["exp_avg"], state["exp_avg_sq"],
+                        mirror, group["lr"], beta1, beta2, group["eps"], group["weight_decay"], state["step"],
+                    )
+                else:
+                    g = p.grad.float()
+                    state["exp_avg"].mul_(beta1).add_(g, alpha=1 - beta1)
+                    state["exp_avg_sq"].mul_(beta2).addcmul_(g, g, value=1 - beta2)
+                    bias_corr1 = 1 - beta1 ** state["step"]
+                    bias_corr2 = 1 - beta2 ** state["step"]
+                    denom = (state["exp_avg_sq"] / bias_corr2).sqrt_().add_(group["eps"])
+                    update = state["exp_avg"] / bias_corr1 / denom + group["weight_decay"] * p.float()
+                    p.data.add_(-group["lr"] * update.to(p.dtype))
+                    if mirror is not None:
+                        mirror.copy_(p.data.to(mirror.dtype))
+        return loss

@@ -1,0 +1,234 @@
+// PyTorch bindings for the hand-written CDNA4 kernels. Compiled by hipcc via
+// torch.utils.cpp_extension with PYTORCH_ROCM_ARCH=gfx950 (see setup.py).
+// All launches go onto the current PyTorch HIP stream so they compose with
+// autograd and the RCCL side-stream logic in averaging/rccl.py.
+#include <torch/extension.h>
+#include <ATen/hip/HIPContext.h>
+#include <hip/hip_runtime.h>
+
+#include "elementwise.hip"
+
+#define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
+#define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
+
+namespace {
+
+inline hipStream_t current_stream() {
+  return at::hip::getCurrentHIPStream().stream();
+}
+
+inline int grid_1d(long long n, int block = 256, int cap = 2048) {
+  long long blocks = (n + block - 1) / block;
+  if (blocks > cap) blocks = cap;   // grid-stride handles the rest (guide G11)
+  if (blocks < 1) blocks = 1;
+  return (int)blocks;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------- averaging
+
+void apply_delta_(torch::Tensor tensor, torch::Tensor delta, double alpha) {
+  CHECK_GPU(tensor); CHECK_CONTIG(tensor); CHECK_CONTIG(delta);
+  TORCH_CHECK(tensor.numel() == delta.numel(), "size mismatch");
+  long long n = tensor.numel();
+  if (tensor.scalar_type() == torch::kBFloat16 && delta.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(apply_delta_bf16, dim3(grid_1d(n / 2 + 1)), dim3(256), 0, current_stream(),
+                       (unsigned short*)tensor.data_ptr(), (const unsigned short*)delta.data_ptr(),
+                       (float)alpha, n);
+  } else if (tensor.scalar_type() == torch::kFloat32 && delta.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(apply_delta_f32, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                       tensor.data_ptr<float>(), delta.data_ptr<float>(), (float)alpha, n);
+  } else {
+    TORCH_CHECK(false, "apply_delta_: unsupported dtype combination");
+  }
+}
+
+void weighted_accumulate_(torch::Tensor acc, torch::Tensor x, double w) {
+  CHECK_GPU(acc); CHECK_CONTIG(acc); CHECK_CONTIG(x);
+  TORCH_CHECK(acc.scalar_type() == torch::kFloat32, "accumulator must be fp32");
+  long long n = acc.numel();
+  if (x.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(weighted_accumulate_f32, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                       acc.data_ptr<float>(), x.data_ptr<float>(), (float)w, n);
+  } else if (x.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(weighted_accumulate_bf16_f32, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                       acc.data_ptr<float>(), (const unsigned short*)x.data_ptr(), (float)w, n);
+  } else {
+    TORCH_CHECK(false, "weighted_accumulate_: unsupported input dtype");
+  }
+}
+
+// ------------------------------------------------------------------- codecs
+
+torch::Tensor compress_fp16_gpu(torch::Tensor input) {
+  CHECK_GPU(input); CHECK_CONTIG(input);
+  auto in32 = input.scalar_type() == torch::kFloat32 ? input : input.to(torch::kFloat32);
+  auto out = torch::empty_like(in32, in32.options().dtype(torch::kFloat16));
+  long long n = in32.numel();
+  hipLaunchKernelGGL(compress_fp16, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                     in32.data_ptr<float>(), (__half*)out.data_ptr(), n);
+  return out;
+}
+
+torch::Tensor decompress_fp16_gpu(torch::Tensor input) {
+  CHECK_GPU(input); CHECK_CONTIG(input);
+  auto out = torch::empty_like(input, input.options().dtype(torch::kFloat32));
+  long long n = input.numel();
+  hipLaunchKernelGGL(decompress_fp16, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                     (const __half*)input.data_ptr(), out.data_ptr<float>(), n);
+  return out;
+}
+
+std::vector<torch::Tensor> quantize_blockwise_gpu(torch::Tensor input) {
+  CHECK_GPU(input); CHECK_CONTIG(input);
+  auto in32 = input.scalar_type() == torch::kFloat32 ? input : input.to(torch::kFloat32);
+  long long n = in32.numel();
+  long long num_blocks = (n + 4095) / 4096;
+  auto q = torch::empty({n}, in32.options().dtype(torch::kInt8));
+  auto absmax = torch::empty({num_blocks}, in32.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(quantize_blockwise_int8, dim3((int)num_blocks), dim3(256), 0, current_stream(),
+                     in32.data_ptr<float>(), q.data_ptr<int8_t>(), absmax.data_ptr<float>(), n);
+  return {q, absmax};
+}
+
+torch::Tensor dequantize_blockwise_gpu(torch::Tensor q, torch::Tensor absmax) {
+  CHECK_GPU(q); CHECK_CONTIG(q); CHECK_CONTIG(absmax);
+  long long n = q.numel();
+  auto out = torch::empty({n}, q.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(dequantize_blockwise_int8, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                     q.data_ptr<int8_t>(), absmax.data_ptr<float>(), out.data_ptr<float>(), n);
+  return out;
+}
+
+// ---------------------------------------------------------------- optimizer
+
+void fused_adamw_(torch::Tensor param, torch::Tensor grad, torch::Tensor exp_avg,
+                  torch::Tensor exp_avg_sq, c10::optional<torch::Tensor> bf16_mirror,
+                  double lr, double beta1, double beta2, double eps,
+                  double weight_decay, long step) {
+  CHECK_GPU(param); CHECK_CONTIG(param); CHECK_CONTIG(grad);
+  CHECK_CONTIG(exp_avg); CHECK_CONTIG(exp_avg_sq);
+  TORCH_CHECK(param.scalar_type() == torch::kFloat32, "master params must be fp32");
+  long long n = param.numel();
+  float bias_corr1 = 1.0f - powf((float)beta1, (float)step);
+  float bias_corr2 = 1.0f - powf((float)beta2, (float)step);
+  unsigned short* mirror_ptr = nullptr;
+  if (bf16_mirror.has_value()) {
+    CHECK_CONTIG(bf16_mirror.value());
+    TORCH_CHECK(bf16_mirror->scalar_type() == torch::kBFloat16, "mirror must be bf16");
+    mirror_ptr = (unsigned short*)bf16_mirror->data_ptr();
+  }
+  if (grad.scalar_type() == torch::kFloat32) {
+    hipLaunchKernelGGL(fused_adamw_f32, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                       param.data_ptr<float>(), grad.data_ptr<float>(), exp_avg.data_ptr<float>(),
+                       exp_avg_sq.data_ptr<float>(), mirror_ptr,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                       bias_corr1, bias_corr2, n);
+  } else if (grad.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(fused_adamw_bf16grad, dim3(grid_1d(n)), dim3(256), 0, current_stream(),
+                       param.data_ptr<float>(), (const unsigned short*)grad.data_ptr(),
+                       exp_avg.data_ptr<float>(), exp_avg_sq.data_ptr<float>(), mirror_ptr,
+                       (float)lr, (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                       bias_corr1, bias_corr2, n);
+  } else {
+    TORCH_CHECK(false, "fused_adamw_: unsupported grad dtype");
+  }
+}
+
+// --------------------------------------------------------------- activations
+
+std::vector<torch::Tensor> bias_gelu_fwd(torch::Tensor x, torch::Tensor bias, bool save_pre_act) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(bias);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && bias.scalar_type() == torch::kBFloat16,
+              "bias_gelu expects bf16");
+  long long cols = x.size(-1);
+  long long rows = x.numel() / cols;
+  auto out = torch::empty_like(x);
+  torch::Tensor pre_act;
+  unsigned short* pre_ptr = nullptr;
+  if (save_pre_act) {
+    pre_act = torch::empty_like(x);
+    pre_ptr = (unsigned short*)pre_act.data_ptr();
+  }
+  hipLaunchKernelGGL(bias_gelu_fwd_bf16, dim3(grid_1d(rows * cols)), dim3(256), 0, current_stream(),
+                     (const unsigned short*)x.data_ptr(), (const unsigned short*)bias.data_ptr(),
+                     (unsigned short*)out.data_ptr(), pre_ptr, rows, cols);
+  if (save_pre_act) return {out, pre_act};
+  return {out};
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(pre_act);
+  long long cols = dy.size(-1);
+  long long rows = dy.numel() / cols;
+  auto dx = torch::empty_like(dy);
+  auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(grid_1d(rows * cols)), dim3(256), 0, current_stream(),
+                     (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
+                     (unsigned short*)dx.data_ptr(), dbias.data_ptr<float>(), rows, cols);
+  return {dx, dbias};
+}
+
+// ----------------------------------------------------------------- layernorm
+
+std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, c10::optional<torch::Tensor> residual,
+                                         torch::Tensor gamma, torch::Tensor beta, double eps) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(gamma); CHECK_CONTIG(beta);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "layernorm expects bf16 input");
+  TORCH_CHECK(gamma.scalar_type() == torch::kFloat32, "gamma/beta must be fp32");
+  int cols = (int)x.size(-1);
+  long long rows = x.numel() / cols;
+  auto y = torch::empty_like(x);
+  auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  const unsigned short* res_ptr = nullptr;
+  torch::Tensor h;
+  unsigned short* h_ptr = nullptr;
+  if (residual.has_value()) {
+    CHECK_CONTIG(residual.value());
+    res_ptr = (const unsigned short*)residual->data_ptr();
+    h = torch::empty_like(x);
+    h_ptr = (unsigned short*)h.data_ptr();
+  }
+  int blocks = (int)((rows + 3) / 4);
+  hipLaunchKernelGGL(layernorm_fwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
+                     (const unsigned short*)x.data_ptr(), res_ptr,
+                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                     (unsigned short*)y.data_ptr(), h_ptr,
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     (float)eps, rows, cols);
+  if (residual.has_value()) return {y, mean, rstd, h};
+  return {y, mean, rstd};
+}
+
+std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torch::Tensor gamma,
+                                         torch::Tensor mean, torch::Tensor rstd) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(h);
+  int cols = (int)dy.size(-1);
+  long long rows = dy.numel() / cols;
+  auto dx = torch::empty_like(dy);
+  auto dgamma = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  auto dbeta = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  int blocks = (int)((rows + 3) / 4);
+  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
+                     (const unsigned short*)dy.data_ptr(), (const unsigned short*)h.data_ptr(),
+                     gamma.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                     (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                     rows, cols);
+  return {dx, dgamma, dbeta};
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("apply_delta_", &apply_delta_, "tensor += alpha * delta (in place)");
+  m.def("weighted_accumulate_", &weighted_accumulate_, "acc += w * x (in place)");
+  m.def("compress_fp16", &compress_fp16_gpu, "clamp + cast to fp16");
+  m.def("decompress_fp16", &decompress_fp16_gpu, "fp16 -> fp32");
+  m.def("quantize_blockwise", &quantize_blockwise_gpu, "blockwise int8 quantize -> (q, absmax)");
+  m.def("dequantize_blockwise", &dequantize_blockwise_gpu, "blockwise int8 dequantize");
+  m.def("fused_adamw_", &fused_adamw_, "fused AdamW step on fp32 master params");
+  m.def("bias_gelu_fwd", &bias_gelu_fwd, "out = gelu(x + bias)");
+  m.def("bias_gelu_bwd", &bias_gelu_bwd, "backward of bias+gelu");
+  m.def("layernorm_fwd", &layernorm_fwd, "fused (residual+)layernorm forward");
+  m.def("layernorm_bwd", &layernorm_bwd, "layernorm backward");
+}

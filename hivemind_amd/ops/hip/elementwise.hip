@@ -1,0 +1,356 @@
+// Memory-bound fused kernels for the training hot path, written directly for
+// CDNA4 / gfx950 (MI355X): 64-wide wavefronts, vectorized bf16 loads
+// (ushort2/short4 reinterpret -- hipcc does not auto-vectorize bf16, guide
+// G13), grid-stride loops capped near CU count * waves (guide G11).
+//
+// Kernels (reference hot spots):
+//   K2  apply_delta            tensor += alpha * delta
+//   K1  weighted_accumulate    acc += w * x
+//   K3  fp16 codec             clamp + cast both directions
+//   K7  blockwise int8 codec   absmax per 4096 block, linear int8
+//   K13 fused Adam             fp32 master step + bf16 param mirror
+//   K9  bias+GELU fwd/bwd      (tanh approximation, as the reference's gelu_fast)
+//   --  LayerNorm fwd/bwd      fused residual-add option
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <hip/hip_fp16.h>
+#include <cstdint>
+
+#define DEVINL __device__ __forceinline__
+
+typedef unsigned short ushort_t;
+
+DEVINL float bf16_to_f32(ushort_t u) {
+  unsigned int w = ((unsigned int)u) << 16;
+  return __uint_as_float(w);
+}
+DEVINL ushort_t f32_to_bf16(float f) {
+  unsigned int w = __float_as_uint(f);
+  // round-to-nearest-even
+  unsigned int rounding_bias = 0x7FFF + ((w >> 16) & 1);
+  return (ushort_t)((w + rounding_bias) >> 16);
+}
+
+// ---------------------------------------------------------------------------
+// K2: tensor += alpha * delta   (bf16 tensor, fp32/bf16 delta)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void apply_delta_bf16(ushort_t* __restrict__ tensor,
+                                            const ushort_t* __restrict__ delta,
+                                            float alpha, long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  // vectorize: 2 bf16 per lane per iteration (4-byte coalesced)
+  long long n2 = n / 2;
+  const uint32_t* d2 = reinterpret_cast<const uint32_t*>(delta);
+  uint32_t* t2 = reinterpret_cast<uint32_t*>(tensor);
+  for (long long i = i0; i < n2; i += stride) {
+    uint32_t tv = t2[i], dv = d2[i];
+    float a0 = bf16_to_f32((ushort_t)(tv & 0xffff)) + alpha * bf16_to_f32((ushort_t)(dv & 0xffff));
+    float a1 = bf16_to_f32((ushort_t)(tv >> 16)) + alpha * bf16_to_f32((ushort_t)(dv >> 16));
+    t2[i] = (uint32_t)f32_to_bf16(a0) | ((uint32_t)f32_to_bf16(a1) << 16);
+  }
+  if (i0 == 0 && (n & 1)) {
+    long long last = n - 1;
+    tensor[last] = f32_to_bf16(bf16_to_f32(tensor[last]) + alpha * bf16_to_f32(delta[last]));
+  }
+}
+
+extern "C" __global__ void apply_delta_f32(float* __restrict__ tensor,
+                                           const float* __restrict__ delta,
+                                           float alpha, long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) tensor[i] += alpha * delta[i];
+}
+
+// ---------------------------------------------------------------------------
+// K1: acc += w * x (fp32 accumulator, bf16 or fp32 input)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void weighted_accumulate_f32(float* __restrict__ acc,
+                                                   const float* __restrict__ x,
+                                                   float w, long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) acc[i] = fmaf(w, x[i], acc[i]);
+}
+
+extern "C" __global__ void weighted_accumulate_bf16_f32(float* __restrict__ acc,
+                                                        const ushort_t* __restrict__ x,
+                                                        float w, long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) acc[i] = fmaf(w, bf16_to_f32(x[i]), acc[i]);
+}
+
+// ---------------------------------------------------------------------------
+// K3: fp16 codec with clamp (reference floating.py:14-41 semantics)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void compress_fp16(const float* __restrict__ in,
+                                         __half* __restrict__ out, long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float v = in[i];
+    v = fminf(fmaxf(v, -65504.f), 65504.f);
+    out[i] = __float2half(v);
+  }
+}
+
+extern "C" __global__ void decompress_fp16(const __half* __restrict__ in,
+                                           float* __restrict__ out, long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) out[i] = __half2float(in[i]);
+}
+
+// ---------------------------------------------------------------------------
+// K7: blockwise int8: absmax per BLK elements, q = round(x * 127 / absmax)
+// one workgroup (256 threads) handles one 4096-block: two-stage reduction in LDS
+// ---------------------------------------------------------------------------
+#define QBLK 4096
+extern "C" __global__ void quantize_blockwise_int8(const float* __restrict__ in,
+                                                   int8_t* __restrict__ q,
+                                                   float* __restrict__ absmax,
+                                                   long long n) {
+  __shared__ float red[256 / 64];  // one partial max per wave
+  long long block = blockIdx.x;
+  long long base = block * QBLK;
+  int tid = threadIdx.x;
+  int lane = tid & 63, wave = tid >> 6;
+  float local_max = 0.f;
+  for (int i = tid; i < QBLK; i += blockDim.x) {
+    long long idx = base + i;
+    float v = (idx < n) ? in[idx] : 0.f;
+    local_max = fmaxf(local_max, fabsf(v));
+  }
+  // wave reduce (64 lanes)
+  for (int off = 32; off > 0; off >>= 1)
+    local_max = fmaxf(local_max, __shfl_down(local_max, off));
+  if (lane == 0) red[wave] = local_max;
+  __syncthreads();
+  if (tid == 0) {
+    float m = red[0];
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) m = fmaxf(m, red[w]);
+    red[0] = m;
+    absmax[block] = m;
+  }
+  __syncthreads();
+  float scale = red[0] > 0.f ? 127.f / red[0] : 0.f;
+  for (int i = tid; i < QBLK; i += blockDim.x) {
+    long long idx = base + i;
+    if (idx < n) {
+      float v = in[idx] * scale;
+      v = fminf(fmaxf(v, -127.f), 127.f);
+      q[idx] = (int8_t)__float2int_rn(v);
+    }
+  }
+}
+
+extern "C" __global__ void dequantize_blockwise_int8(const int8_t* __restrict__ q,
+                                                     const float* __restrict__ absmax,
+                                                     float* __restrict__ out,
+                                                     long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float scale = absmax[i / QBLK] / 127.f;
+    out[i] = (float)q[i] * scale;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K13: fused Adam step on fp32 master params with optional bf16 mirror
+// p -= lr * ( m_hat / (sqrt(v_hat) + eps) + wd * p )   (AdamW-style decoupled wd)
+// ---------------------------------------------------------------------------
+extern "C" __global__ void fused_adamw_f32(float* __restrict__ param,
+                                           const float* __restrict__ grad,
+                                           float* __restrict__ exp_avg,
+                                           float* __restrict__ exp_avg_sq,
+                                           ushort_t* __restrict__ param_bf16_mirror,  // may be null
+                                           float lr, float beta1, float beta2,
+                                           float eps, float weight_decay,
+                                           float bias_corr1, float bias_corr2,
+                                           long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float g = grad[i];
+    float m = exp_avg[i] = fmaf(beta1, exp_avg[i], (1.f - beta1) * g);
+    float v = exp_avg_sq[i] = fmaf(beta2, exp_avg_sq[i], (1.f - beta2) * g * g);
+    float m_hat = m / bias_corr1;
+    float v_hat = v / bias_corr2;
+    float p = param[i];
+    p -= lr * (m_hat / (sqrtf(v_hat) + eps) + weight_decay * p);
+    param[i] = p;
+    if (param_bf16_mirror != nullptr) param_bf16_mirror[i] = f32_to_bf16(p);
+  }
+}
+
+// grads arriving in bf16 (model grads) -- same step, bf16 grad load
+extern "C" __global__ void fused_adamw_bf16grad(float* __restrict__ param,
+                                                const ushort_t* __restrict__ grad,
+                                                float* __restrict__ exp_avg,
+                                                float* __restrict__ exp_avg_sq,
+                                                ushort_t* __restrict__ param_bf16_mirror,
+                                                float lr, float beta1, float beta2,
+                                                float eps, float weight_decay,
+                                                float bias_corr1, float bias_corr2,
+                                                long long n) {
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float g = bf16_to_f32(grad[i]);
+    float m = exp_avg[i] = fmaf(beta1, exp_avg[i], (1.f - beta1) * g);
+    float v = exp_avg_sq[i] = fmaf(beta2, exp_avg_sq[i], (1.f - beta2) * g * g);
+    float m_hat = m / bias_corr1;
+    float v_hat = v / bias_corr2;
+    float p = param[i];
+    p -= lr * (m_hat / (sqrtf(v_hat) + eps) + weight_decay * p);
+    param[i] = p;
+    if (param_bf16_mirror != nullptr) param_bf16_mirror[i] = f32_to_bf16(p);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// K9a: bias + GELU (tanh approx) forward, bf16 activations
+//   out = gelu(x + bias);  x: [rows, cols], bias: [cols]
+// ---------------------------------------------------------------------------
+DEVINL float gelu_tanh(float x) {
+  const float c = 0.7978845608028654f;  // sqrt(2/pi)
+  float inner = c * (x + 0.044715f * x * x * x);
+  return 0.5f * x * (1.f + tanhf(inner));
+}
+DEVINL float gelu_tanh_grad(float x) {
+  const float c = 0.7978845608028654f;
+  float x2 = x * x;
+  float inner = c * (x + 0.044715f * x * x2);
+  float t = tanhf(inner);
+  float sech2 = 1.f - t * t;
+  return 0.5f * (1.f + t) + 0.5f * x * sech2 * c * (1.f + 3.f * 0.044715f * x2);
+}
+
+extern "C" __global__ void bias_gelu_fwd_bf16(const ushort_t* __restrict__ x,
+                                              const ushort_t* __restrict__ bias,
+                                              ushort_t* __restrict__ out,
+                                              ushort_t* __restrict__ pre_act,  // saved for bwd (x+bias); may be null
+                                              long long rows, long long cols) {
+  long long n = rows * cols;
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float v = bf16_to_f32(x[i]) + bf16_to_f32(bias[i % cols]);
+    if (pre_act != nullptr) pre_act[i] = f32_to_bf16(v);
+    out[i] = f32_to_bf16(gelu_tanh(v));
+  }
+}
+
+// K9b: backward: dx = dy * gelu'(pre_act); also accumulates dbias via atomics
+extern "C" __global__ void bias_gelu_bwd_bf16(const ushort_t* __restrict__ dy,
+                                              const ushort_t* __restrict__ pre_act,
+                                              ushort_t* __restrict__ dx,
+                                              float* __restrict__ dbias,  // fp32 accumulators [cols]
+                                              long long rows, long long cols) {
+  long long n = rows * cols;
+  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
+  long long stride = (long long)gridDim.x * blockDim.x;
+  for (long long i = i0; i < n; i += stride) {
+    float g = bf16_to_f32(dy[i]) * gelu_tanh_grad(bf16_to_f32(pre_act[i]));
+    dx[i] = f32_to_bf16(g);
+    atomicAdd(&dbias[i % cols], g);
+  }
+}
+
+// ---------------------------------------------------------------------------
+// LayerNorm forward: one wave per row (cols <= 8192), bf16 in/out, fp32 stats.
+// Optionally fuses a residual add: h = x + residual; y = ln(h).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void layernorm_fwd_bf16(const ushort_t* __restrict__ x,
+                                              const ushort_t* __restrict__ residual,  // may be null
+                                              const float* __restrict__ gamma,
+                                              const float* __restrict__ beta,
+                                              ushort_t* __restrict__ y,
+                                              ushort_t* __restrict__ h_out,  // saved normalized input source (x+res); may be null
+                                              float* __restrict__ mean_out,
+                                              float* __restrict__ rstd_out,
+                                              float eps, long long rows, int cols) {
+  // blockDim.x = 256 -> 4 waves; each wave owns one row
+  int wave = threadIdx.x >> 6;
+  int lane = threadIdx.x & 63;
+  long long row = (long long)blockIdx.x * 4 + wave;
+  if (row >= rows) return;
+  const ushort_t* xr = x + row * cols;
+  const ushort_t* rr = residual ? residual + row * cols : nullptr;
+  ushort_t* hr = h_out ? h_out + row * cols : nullptr;
+  ushort_t* yr = y + row * cols;
+
+  float sum = 0.f, sumsq = 0.f;
+  for (int i = lane; i < cols; i += 64) {
+    float v = bf16_to_f32(xr[i]);
+    if (rr) v += bf16_to_f32(rr[i]);
+    if (hr) hr[i] = f32_to_bf16(v);
+    sum += v;
+    sumsq += v * v;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    sum += __shfl_down(sum, off);
+    sumsq += __shfl_down(sumsq, off);
+  }
+  sum = __shfl(sum, 0);
+  sumsq = __shfl(sumsq, 0);
+  float mean = sum / cols;
+  float var = sumsq / cols - mean * mean;
+  float rstd = rsqrtf(var + eps);
+  if (lane == 0) {
+    mean_out[row] = mean;
+    rstd_out[row] = rstd;
+  }
+  for (int i = lane; i < cols; i += 64) {
+    float v = bf16_to_f32(xr[i]);
+    if (rr) v += bf16_to_f32(rr[i]);
+    float norm = (v - mean) * rstd;
+    yr[i] = f32_to_bf16(fmaf(norm, gamma[i], beta[i]));
+  }
+}
+
+// LayerNorm backward: dx, plus fp32 atomically-accumulated dgamma/dbeta.
+extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
+                                              const ushort_t* __restrict__ h,  // ln input (post-residual)
+                                              const float* __restrict__ gamma,
+                                              const float* __restrict__ mean,
+                                              const float* __restrict__ rstd,
+                                              ushort_t* __restrict__ dx,
+                                              float* __restrict__ dgamma,
+                                              float* __restrict__ dbeta,
+                                              long long rows, int cols) {
+  int wave = threadIdx.x >> 6;
+  int lane = threadIdx.x & 63;
+  long long row = (long long)blockIdx.x * 4 + wave;
+  if (row >= rows) return;
+  const ushort_t* dyr = dy + row * cols;
+  const ushort_t* hr = h + row * cols;
+  ushort_t* dxr = dx + row * cols;
+  float mu = mean[row], rs = rstd[row];
+
+  // first pass: c1 = sum(dy*gamma*xhat), c2 = sum(dy*gamma)
+  float c1 = 0.f, c2 = 0.f;
+  for (int i = lane; i < cols; i += 64) {
+    float dyv = bf16_to_f32(dyr[i]);
+    float xhat = (bf16_to_f32(hr[i]) - mu) * rs;
+    float dg = dyv * gamma[i];
+    c1 += dg * xhat;
+    c2 += dg;
+    atomicAdd(&dgamma[i], dyv * xhat);
+    atomicAdd(&dbeta[i], dyv);
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    c1 += __shfl_down(c1, off);
+    c2 += __shfl_down(c2, off);
+  }
+  c1 = __shfl(c1, 0) / cols;
+  c2 = __shfl(c2, 0) / cols;
+  for (int i = lane; i < cols; i += 64) {
+    float xhat = (bf16_to_f32(hr[i]) - mu) * rs;
+    float dg = bf16_to_f32(dyr[i]) * gamma[i];
+    dxr[i] = f32_to_bf16((dg - c2 - xhat * c1) * rs);
+  }
+}

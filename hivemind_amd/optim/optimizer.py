@@ -308,14 +308,17 @@ class Optimizer(torch.optim.Optimizer):
         (reference optimizer.py:438-509)."""
         assert self._parent_pid == os.getpid()
         _epoch_start = time.perf_counter()
-        began_averaging_gradients = False
-        if self.use_gradient_averaging and not self.auxiliary:
-            began_averaging_gradients = self._begin_averaging_gradients()
-            if not began_averaging_gradients:
-                pass  # failed to average; apply local gradients anyway
-
         next_epoch = max(self.local_epoch + 1, self.tracker.global_epoch)
         swarm_not_empty = self.tracker.global_progress.num_peers > 1
+        began_averaging_gradients = False
+        if self.use_gradient_averaging and not self.auxiliary:
+            if swarm_not_empty:
+                began_averaging_gradients = self._begin_averaging_gradients()
+            else:
+                # single-peer swarm: skip matchmaking, use local accumulators as-is
+                self.grad_averager.load_accumulators_into_averager_()
+                self.grad_averager.reset_accumulated_grads_()
+
         should_perform_optimizer_step = not self.auxiliary and not self.use_local_updates
         should_average_state = (
             swarm_not_empty
@@ -473,6 +476,9 @@ class Optimizer(torch.optim.Optimizer):
         return f"{self.__class__.__name__}(run_id={self.run_id}, epoch={self.local_epoch})"
 
     def shutdown(self):
+        if getattr(self, "_shutdown_complete", False):
+            return
+        self._shutdown_complete = True
         logger.log(self.status_loglevel, "shutting down optimizer")
         self._finish_scheduled_averaging()
         self.tracker.shutdown(self.shutdown_timeout)

@@ -313,6 +313,8 @@ class ProgressTracker(threading.Thread):
         self.should_report_progress.set()
         self.global_state_updated.set()
         self.shutdown_complete.wait(timeout)
+        if not self.dht.is_alive:
+            return
         try:
             self.dht.store(
                 key=self.training_progress_key,
