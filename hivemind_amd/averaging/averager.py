@@ -278,6 +278,7 @@ class DecentralizedAverager(ServicerBase):
         )
         if not require_trigger:
             control.allow_allreduce()
+        logger.debug(f"{self.prefix}@{self.peer_id}: step() created control {id(control):#x} trig={control.triggered}")
         asyncio.run_coroutine_threadsafe(self._step(control), self._loop)
         return control.result(timeout) if wait else control
 
@@ -287,15 +288,26 @@ class DecentralizedAverager(ServicerBase):
                 try:
                     self._pending_groups_registered = asyncio.Event()
                     step.stage = AveragingStage.LOOKING_FOR_GROUP
+                    logger.debug(f"{self.prefix}@{self.peer_id}: looking for group (deadline in {step.deadline - get_dht_time():.1f}s)")
                     group_info = await self._matchmaking.look_for_group(step)
+                    logger.debug(f"{self.prefix}@{self.peer_id}: matchmaking returned {group_info}")
                     if group_info is None:
                         if step.allow_retries and get_dht_time() < step.deadline:
                             continue
                         raise AveragingError("Averaging step failed: could not find a group")
 
+                    if step.done():
+                        break  # cancelled while matchmaking
                     if not step.triggered:
                         step.stage = AveragingStage.AWAITING_TRIGGER
-                        await step.wait_for_trigger()
+                        logger.debug(f"{self.prefix}@{self.peer_id}: awaiting trigger on {id(step):#x}")
+                        try:
+                            await step.wait_for_trigger()
+                        except (asyncio.CancelledError, concurrent.futures.CancelledError):
+                            break  # step was cancelled: do not run the round
+                    if step.done() or step.cancelled():
+                        break
+                    logger.debug(f"{self.prefix}@{self.peer_id}: trigger received")
                     step.stage = AveragingStage.RUNNING_ALLREDUCE
                     gathered = await asyncio.wait_for(
                         self._aggregate_with_group(group_info, step.weight), timeout=self.allreduce_timeout
@@ -321,6 +333,7 @@ class DecentralizedAverager(ServicerBase):
 
     async def _aggregate_with_group(self, group_info: GroupInfo, weight: float) -> Dict[PeerID, Any]:
         """Run the data plane for one assembled group (reference averager.py:514-562)."""
+        logger.debug(f"{self.prefix}@{self.peer_id}: entering aggregation, group={group_info.group_id.hex()[:8]}")
         try:
             num_peers = len(group_info.peer_ids)
             # unpack per-peer metadata: [bandwidth, mode, user_gather, dist_info]
@@ -506,6 +519,9 @@ class DecentralizedAverager(ServicerBase):
             if len(local) == len(tensors):
                 for mine, new in zip(local, tensors):
                     mine.detach().copy_(new.to(mine.device, mine.dtype))
+
+    def _debug_state(self) -> str:
+        return f"{self.prefix}@{self.peer_id}"
 
     async def rpc_download_state(self, request: DownloadRequest, context: RpcContext) -> AsyncIterator[DownloadData]:
         """Stream (metadata, tensors) to a joining peer (reference averager.py:628-666)."""

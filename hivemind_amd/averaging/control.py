@@ -101,7 +101,7 @@ class StepControl:
 
     @property
     def triggered(self) -> bool:
-        return self._trigger.done()
+        return self._trigger.done() and not self._trigger.cancelled()
 
     async def wait_for_trigger(self):
         import asyncio

@@ -133,6 +133,7 @@ class GradientAverager(DecentralizedAverager):
         if reset_accumulators:
             self.reset_accumulated_grads_()
         control.allow_allreduce()
+        logger.debug(f"grad_averager: triggered control {id(control):#x}")
         return control.result(timeout) if wait else control
 
     @torch.no_grad()

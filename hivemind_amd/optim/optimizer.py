@@ -318,6 +318,10 @@ class Optimizer(torch.optim.Optimizer):
                 # single-peer swarm: skip matchmaking, use local accumulators as-is
                 self.grad_averager.load_accumulators_into_averager_()
                 self.grad_averager.reset_accumulated_grads_()
+            if not began_averaging_gradients and self.scheduled_grads is not None:
+                # never wait on a control that was not (and will not be) triggered
+                self.scheduled_grads.cancel()
+                self.scheduled_grads = None
 
         should_perform_optimizer_step = not self.auxiliary and not self.use_local_updates
         should_average_state = (

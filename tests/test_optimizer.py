@@ -103,7 +103,7 @@ def _run_training_peer(dht, X, y, target_batch_size, batch_per_step, results, id
         matchmaking_time=1.0,
         averaging_timeout=30.0,
         reuse_grad_buffers=reuse_grad_buffers,
-        averager_opts=dict(request_timeout=0.5, min_group_size=1),
+        averager_opts=dict(request_timeout=0.5, min_group_size=2),
         tracker_opts=dict(min_refresh_period=0.2, default_refresh_period=0.5),
         verbose=False,
         **opt_kwargs,
