@@ -186,4 +186,12 @@ def main():
 
 
 if __name__ == "__main__":
-    main()
+    try:
+        main()
+    except BaseException:
+        import traceback
+
+        traceback.print_exc()
+        # daemon threads (DHT loop, tracker) must not keep a failed bench alive
+        os._exit(1)
+    os._exit(0)

@@ -41,6 +41,19 @@ def hip_available() -> bool:
     return _hip_ops is not None
 
 
+def bind_grad(param: torch.Tensor, grad: torch.Tensor) -> None:
+    """Assign a gradient buffer whose dtype may differ from the parameter's
+    (e.g. bf16 swarm-averaged grads bound to fp32 master params; the fused
+    AdamW kernel consumes bf16 grads directly). torch >= 2.10 enforces
+    grad-dtype matching unless grad_dtype is relaxed first."""
+    if grad is not None and grad.dtype != param.dtype:
+        try:
+            param.grad_dtype = None
+        except (AttributeError, RuntimeError):
+            grad = grad.to(param.dtype)
+    param.grad = grad
+
+
 # ---------------------------------------------------------------------------
 # fused layernorm (optionally fused residual add)
 # ---------------------------------------------------------------------------

@@ -165,12 +165,14 @@ class Optimizer(torch.optim.Optimizer):
             **kwargs,
         )
         if self.offload_optimizer:
+            from ..ops import bind_grad
+
             optimized_param_groups = self.state_averager.optimizer.param_groups
             optimized_parameters = [p for group in optimized_param_groups for p in group["params"]]
             with grad_averager.get_tensors() as averaged_gradients:
                 assert len(averaged_gradients) == len(optimized_parameters)
                 for opt_param, averaged_grad in zip(optimized_parameters, averaged_gradients):
-                    opt_param.grad = averaged_grad
+                    bind_grad(opt_param, averaged_grad)
         return grad_averager
 
     def _make_progress_tracker(self, target_batch_size: int, **kwargs) -> ProgressTracker:
@@ -411,9 +413,11 @@ class Optimizer(torch.optim.Optimizer):
         if self.offload_optimizer:
             pass  # grads are bound to the averager's buffers already (see _make_gradient_averager)
         else:
+            from ..ops import bind_grad
+
             with self.grad_averager.get_tensors() as averaged_gradients:
                 for opt_param, averaged_grad in zip(optimized_parameters, averaged_gradients):
-                    opt_param.grad = averaged_grad.to(opt_param.device, opt_param.dtype)
+                    bind_grad(opt_param, averaged_grad)
         self.grad_averager.notify_used_averaged_gradients()
 
     # --------------------------------------------------------------- re-sync

@@ -99,11 +99,13 @@ def test_fused_adamw_bf16_grads_and_mirror():
     from hivemind_amd.ops import FusedAdamW
 
     torch.manual_seed(0)
+    from hivemind_amd.ops import bind_grad
+
     p = torch.randn(4096, device="cuda", dtype=torch.float32).requires_grad_(True)
     mirror = p.detach().bfloat16().clone()
     opt = FusedAdamW([p], lr=1e-2)
     opt.set_mirror(p, mirror)
-    p.grad = torch.randn(4096, device="cuda", dtype=torch.bfloat16)
+    bind_grad(p, torch.randn(4096, device="cuda", dtype=torch.bfloat16))
     opt.step()
     assert torch.allclose(mirror.float(), p.detach().float(), atol=1e-2)
 
