@@ -68,8 +68,7 @@ class FusedLayerNorm(nn.Module):
         self.eps = eps
 
     def forward(self, x: torch.Tensor, residual: Optional[torch.Tensor] = None) -> torch.Tensor:
-        y, _h = fused_layernorm(x, self.weight, self.bias, residual=residual, eps=self.eps)
-        return y
+        return fused_layernorm(x, self.weight, self.bias, residual=residual, eps=self.eps)
 
 
 class AlbertLayer(nn.Module):

@@ -68,17 +68,15 @@ class _FusedLayerNorm(torch.autograd.Function):
             y, mean, rstd, h = outs
         else:
             y, mean, rstd = outs
-            h = x
+            h = x.detach()
         ctx.save_for_backward(h, gamma, mean, rstd)
         ctx.has_residual = residual is not None
-        return (y, h) if residual is not None else (y, x)
+        return y
 
     @staticmethod
-    def backward(ctx, dy, dh_extra):
+    def backward(ctx, dy):
         h, gamma, mean, rstd = ctx.saved_tensors
         dx, dgamma, dbeta = hip_ops().layernorm_bwd(dy.contiguous(), h.contiguous(), gamma, mean, rstd)
-        if dh_extra is not None and dh_extra.abs().sum() != 0:  # pragma: no cover - rare path
-            dx = dx + dh_extra
         # d(x) and d(residual) are identical for the fused residual add
         dres = dx if ctx.has_residual else None
         return dx, dres, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
@@ -90,16 +88,13 @@ def fused_layernorm(
     beta: torch.Tensor,
     residual: Optional[torch.Tensor] = None,
     eps: float = 1e-12,
-) -> Tuple[torch.Tensor, torch.Tensor]:
-    """y = LayerNorm(x + residual); returns (y, x + residual).
-
-    GPU: one fused HIP kernel (wave-per-row, bf16). CPU: torch reference.
-    """
+) -> torch.Tensor:
+    """y = LayerNorm(x + residual), one fused HIP kernel (wave-per-row, bf16)
+    on GPU; torch reference on CPU."""
     if x.is_cuda and x.dtype == torch.bfloat16:
         return _FusedLayerNorm.apply(x, residual, gamma, beta, eps)
     h = x + residual if residual is not None else x
-    y = torch.nn.functional.layer_norm(h.float(), (h.shape[-1],), gamma.float(), beta.float(), eps).to(x.dtype)
-    return y, h
+    return torch.nn.functional.layer_norm(h.float(), (h.shape[-1],), gamma.float(), beta.float(), eps).to(x.dtype)
 
 
 class _FusedBiasGelu(torch.autograd.Function):
