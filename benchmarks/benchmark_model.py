@@ -6,8 +6,11 @@ Separates model compute from optimizer/averaging overhead in bench.py.
 
 import argparse
 import json
+import os
 import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 

@@ -228,10 +228,13 @@ class FusedAdamW(torch.optim.Optimizer):
                 state["step"] += 1
                 mirror = self._mirrors.get(p)
                 if p.is_cuda and p.dtype == torch.float32:
+                    kernel_mirror = mirror if (mirror is not None and mirror.dtype == torch.bfloat16) else None
                     hip_ops().fused_adamw_(
                         p.data, p.grad.contiguous(), state["exp_avg"], state["exp_avg_sq"],
-                        mirror, group["lr"], beta1, beta2, group["eps"], group["weight_decay"], state["step"],
+                        kernel_mirror, group["lr"], beta1, beta2, group["eps"], group["weight_decay"], state["step"],
                     )
+                    if mirror is not None and kernel_mirror is None:
+                        mirror.copy_(p.data.to(mirror.dtype))
                 else:
                     g = p.grad.float()
                     state["exp_avg"].mul_(beta1).add_(g, alpha=1 - beta1)
