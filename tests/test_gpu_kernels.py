@@ -118,7 +118,7 @@ def test_layernorm_forward_backward():
     B, H = 64, 768
     x = torch.randn(B, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
     res = torch.randn(B, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
-    gamma = torch.rand(H, device="cuda", dtype=torch.float32, requires_grad=True) + 0.5
+    gamma = (torch.rand(H, device="cuda", dtype=torch.float32) + 0.5).requires_grad_(True)
     beta = torch.randn(H, device="cuda", dtype=torch.float32, requires_grad=True)
 
     y = fused_layernorm(x, gamma, beta, residual=res, eps=1e-12)
