@@ -145,16 +145,18 @@ class AlbertForMaskedLM(nn.Module):
         self.albert = AlbertModel(config)
         self.mlm_dense = nn.Linear(config.hidden_size, config.embedding_size, dtype=dt)
         self.mlm_norm = FusedLayerNorm(config.embedding_size, config.layer_norm_eps)
-        self.mlm_bias = nn.Parameter(torch.zeros(config.vocab_size, dtype=torch.float32))
+        self.mlm_bias = nn.Parameter(torch.zeros(config.vocab_size, dtype=dt))
 
     def forward(self, input_ids: torch.Tensor, labels: Optional[torch.Tensor] = None):
         hidden = self.albert(input_ids)
         x = self.mlm_dense(hidden)
         x = F.gelu(x, approximate="tanh")
         x = self.mlm_norm(x)
-        logits = F.linear(x, self.albert.word_embeddings.weight) + self.mlm_bias.to(x.dtype)
+        # logits stay in bf16: cross_entropy's log-softmax accumulates in fp32
+        # internally; materializing [B*S, vocab] fp32 would be ~8 GB at batch 128
+        logits = F.linear(x, self.albert.word_embeddings.weight, self.mlm_bias)
         if labels is not None:
-            loss = F.cross_entropy(logits.float().view(-1, self.config.vocab_size), labels.view(-1), ignore_index=-100)
+            loss = F.cross_entropy(logits.view(-1, self.config.vocab_size), labels.view(-1), ignore_index=-100)
             return loss, logits
         return logits
 

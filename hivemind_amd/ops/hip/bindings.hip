@@ -161,10 +161,13 @@ std::vector<torch::Tensor> bias_gelu_fwd(torch::Tensor x, torch::Tensor bias, bo
 std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act) {
   CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(pre_act);
   long long cols = dy.size(-1);
+  TORCH_CHECK(cols % 2 == 0, "bias_gelu_bwd requires an even column count");
   long long rows = dy.numel() / cols;
   auto dx = torch::empty_like(dy);
   auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
-  hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(grid_1d(rows * cols)), dim3(256), 0, current_stream(),
+  int blocks = (int)std::min<long long>(rows, 1024);
+  size_t lds_bytes = (size_t)cols * sizeof(float);
+  hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(blocks), dim3(256), lds_bytes, current_stream(),
                      (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
                      (unsigned short*)dx.data_ptr(), dbias.data_ptr<float>(), rows, cols);
   return {dx, dbias};
@@ -178,6 +181,7 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, c10::optional<torch::T
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16, "layernorm expects bf16 input");
   TORCH_CHECK(gamma.scalar_type() == torch::kFloat32, "gamma/beta must be fp32");
   int cols = (int)x.size(-1);
+  TORCH_CHECK(cols % 2 == 0, "layernorm requires an even column count");
   long long rows = x.numel() / cols;
   auto y = torch::empty_like(x);
   auto mean = torch::empty({rows}, x.options().dtype(torch::kFloat32));
@@ -210,8 +214,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torc
   auto dx = torch::empty_like(dy);
   auto dgamma = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   auto dbeta = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
-  int blocks = (int)((rows + 3) / 4);
-  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
+  int blocks = (int)std::min<long long>((rows + 3) / 4, 1024);
+  size_t lds_bytes = 2 * (size_t)cols * sizeof(float);
+  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(blocks), dim3(256), lds_bytes, current_stream(),
                      (const unsigned short*)dy.data_ptr(), (const unsigned short*)h.data_ptr(),
                      gamma.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
