@@ -143,6 +143,7 @@ std::vector<torch::Tensor> bias_gelu_fwd(torch::Tensor x, torch::Tensor bias, bo
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && bias.scalar_type() == torch::kBFloat16,
               "bias_gelu expects bf16");
   long long cols = x.size(-1);
+  TORCH_CHECK(cols % 2 == 0 && cols <= 2 * 8 * 256, "bias_gelu: cols must be even and <= 4096");
   long long rows = x.numel() / cols;
   auto out = torch::empty_like(x);
   torch::Tensor pre_act;
@@ -151,7 +152,7 @@ std::vector<torch::Tensor> bias_gelu_fwd(torch::Tensor x, torch::Tensor bias, bo
     pre_act = torch::empty_like(x);
     pre_ptr = (unsigned short*)pre_act.data_ptr();
   }
-  hipLaunchKernelGGL(bias_gelu_fwd_bf16, dim3(grid_1d(rows * cols)), dim3(256), 0, current_stream(),
+  hipLaunchKernelGGL(bias_gelu_fwd_bf16, dim3((int)std::min<long long>(rows, 2048)), dim3(256), 0, current_stream(),
                      (const unsigned short*)x.data_ptr(), (const unsigned short*)bias.data_ptr(),
                      (unsigned short*)out.data_ptr(), pre_ptr, rows, cols);
   if (save_pre_act) return {out, pre_act};
@@ -165,9 +166,9 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act
   long long rows = dy.numel() / cols;
   auto dx = torch::empty_like(dy);
   auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
-  int blocks = (int)std::min<long long>(rows, 1024);
-  size_t lds_bytes = (size_t)cols * sizeof(float);
-  hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(blocks), dim3(256), lds_bytes, current_stream(),
+  TORCH_CHECK(cols <= 2 * 8 * 256, "bias_gelu_bwd: cols must be <= 4096");
+  int blocks = (int)std::min<long long>(rows, 2048);
+  hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
                      (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
                      (unsigned short*)dx.data_ptr(), dbias.data_ptr<float>(), rows, cols);
   return {dx, dbias};
@@ -214,9 +215,9 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torc
   auto dx = torch::empty_like(dy);
   auto dgamma = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   auto dbeta = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
-  int blocks = (int)std::min<long long>((rows + 3) / 4, 1024);
-  size_t lds_bytes = 2 * (size_t)cols * sizeof(float);
-  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(blocks), dim3(256), lds_bytes, current_stream(),
+  TORCH_CHECK(cols % 2 == 0 && cols <= 2 * 64 * 16, "layernorm_bwd: cols must be even and <= 2048");
+  int blocks = (int)std::min<long long>((rows + 3) / 4, 2048);
+  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
                      (const unsigned short*)dy.data_ptr(), (const unsigned short*)h.data_ptr(),
                      gamma.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
                      (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),

@@ -234,13 +234,21 @@ extern "C" __global__ void bias_gelu_fwd_bf16(const ushort_t* __restrict__ x,
                                               ushort_t* __restrict__ out,
                                               ushort_t* __restrict__ pre_act,  // saved for bwd (x+bias); may be null
                                               long long rows, long long cols) {
-  long long n = rows * cols;
-  long long i0 = (long long)(blockIdx.x) * blockDim.x + threadIdx.x;
-  long long stride = (long long)gridDim.x * blockDim.x;
-  for (long long i = i0; i < n; i += stride) {
-    float v = bf16_to_f32(x[i]) + bf16_to_f32(bias[i % cols]);
-    if (pre_act != nullptr) pre_act[i] = f32_to_bf16(v);
-    out[i] = f32_to_bf16(gelu_tanh(v));
+  const uint32_t* x2 = reinterpret_cast<const uint32_t*>(x);
+  const uint32_t* b2 = reinterpret_cast<const uint32_t*>(bias);
+  uint32_t* out2 = reinterpret_cast<uint32_t*>(out);
+  uint32_t* pre2 = reinterpret_cast<uint32_t*>(pre_act);
+  long long cols2 = cols >> 1;
+  for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
+    long long base = row * cols2;
+    for (long long c = threadIdx.x; c < cols2; c += blockDim.x) {
+      uint32_t xv = x2[base + c], bv = b2[c];
+      float v0 = bf16_to_f32((ushort_t)(xv & 0xffff)) + bf16_to_f32((ushort_t)(bv & 0xffff));
+      float v1 = bf16_to_f32((ushort_t)(xv >> 16)) + bf16_to_f32((ushort_t)(bv >> 16));
+      if (pre_act != nullptr)
+        pre2[base + c] = (uint32_t)f32_to_bf16(v0) | ((uint32_t)f32_to_bf16(v1) << 16);
+      out2[base + c] = (uint32_t)f32_to_bf16(gelu_tanh(v0)) | ((uint32_t)f32_to_bf16(gelu_tanh(v1)) << 16);
+    }
   }
 }
 
@@ -250,34 +258,41 @@ extern "C" __global__ void bias_gelu_fwd_bf16(const ushort_t* __restrict__ x,
 // (per-element global atomics to `cols` addresses serialize catastrophically
 // -- measured 1.6 ms/call on [65536, 3072] before this restructuring).
 // Requires dynamic LDS of cols * sizeof(float); cols must be even.
+// Each thread owns a fixed set of column-pairs (c = tid + k*blockDim) across
+// every row the block visits, so dbias accumulates in REGISTERS -- no atomics
+// in the row loop at all. One global atomicAdd per owned column at the end
+// (spread over `cols` addresses -> negligible). Supports cols <= 2 * 8 * 256.
+#define BG_MAX_PAIRS 8
 extern "C" __global__ void bias_gelu_bwd_bf16(const ushort_t* __restrict__ dy,
                                               const ushort_t* __restrict__ pre_act,
                                               ushort_t* __restrict__ dx,
                                               float* __restrict__ dbias,  // fp32 accumulators [cols]
                                               long long rows, long long cols) {
-  extern __shared__ float dbias_local[];
-  for (int c = threadIdx.x; c < cols; c += blockDim.x) dbias_local[c] = 0.f;
-  __syncthreads();
-
   const uint32_t* dy2 = reinterpret_cast<const uint32_t*>(dy);
   const uint32_t* pre2 = reinterpret_cast<const uint32_t*>(pre_act);
   uint32_t* dx2 = reinterpret_cast<uint32_t*>(dx);
   long long cols2 = cols >> 1;
 
+  float acc0[BG_MAX_PAIRS], acc1[BG_MAX_PAIRS];
+#pragma unroll
+  for (int p = 0; p < BG_MAX_PAIRS; ++p) acc0[p] = acc1[p] = 0.f;
+
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     long long base = row * cols2;
-    for (long long c = threadIdx.x; c < cols2; c += blockDim.x) {
+    int p = 0;
+    for (long long c = threadIdx.x; c < cols2; c += blockDim.x, ++p) {
       uint32_t dyv = dy2[base + c], pv = pre2[base + c];
       float g0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gelu_tanh_grad(bf16_to_f32((ushort_t)(pv & 0xffff)));
       float g1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gelu_tanh_grad(bf16_to_f32((ushort_t)(pv >> 16)));
       dx2[base + c] = (uint32_t)f32_to_bf16(g0) | ((uint32_t)f32_to_bf16(g1) << 16);
-      atomicAdd(&dbias_local[2 * c], g0);
-      atomicAdd(&dbias_local[2 * c + 1], g1);
+      acc0[p] += g0;
+      acc1[p] += g1;
     }
   }
-  __syncthreads();
-  for (int c = threadIdx.x; c < cols; c += blockDim.x) {
-    if (dbias_local[c] != 0.f) atomicAdd(&dbias[c], dbias_local[c]);
+  int p = 0;
+  for (long long c = threadIdx.x; c < cols2; c += blockDim.x, ++p) {
+    if (acc0[p] != 0.f) atomicAdd(&dbias[2 * c], acc0[p]);
+    if (acc1[p] != 0.f) atomicAdd(&dbias[2 * c + 1], acc1[p]);
   }
 }
 
@@ -357,6 +372,11 @@ extern "C" __global__ void layernorm_fwd_bf16(const ushort_t* __restrict__ x,
 // grid-strides over rows (4 waves x many rows) so the number of global
 // atomics is cols x gridDim, independent of row count. Dynamic LDS:
 // 2 * cols * sizeof(float).
+// Wave-per-row; each lane owns fixed column-pairs (c = lane + k*64, the same
+// in every row), so dgamma/dbeta accumulate in registers across the rows this
+// wave visits; one global atomicAdd per owned column at the end. Vectorized
+// paired bf16 loads throughout. Supports cols <= 2 * 64 * LN_MAX_PAIRS.
+#define LN_MAX_PAIRS 16
 extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
                                               const ushort_t* __restrict__ h,  // ln input (post-residual)
                                               const float* __restrict__ gamma,
@@ -366,33 +386,38 @@ extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
                                               float* __restrict__ dgamma,
                                               float* __restrict__ dbeta,
                                               long long rows, int cols) {
-  extern __shared__ float smem[];
-  float* dgamma_local = smem;          // [cols]
-  float* dbeta_local = smem + cols;    // [cols]
-  for (int c = threadIdx.x; c < 2 * cols; c += blockDim.x) smem[c] = 0.f;
-  __syncthreads();
-
   int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
   int waves_per_block = blockDim.x >> 6;
+  int cols2 = cols >> 1;
+
+  float accg0[LN_MAX_PAIRS], accg1[LN_MAX_PAIRS], accb0[LN_MAX_PAIRS], accb1[LN_MAX_PAIRS];
+#pragma unroll
+  for (int p = 0; p < LN_MAX_PAIRS; ++p) accg0[p] = accg1[p] = accb0[p] = accb1[p] = 0.f;
 
   for (long long row = (long long)blockIdx.x * waves_per_block + wave; row < rows;
        row += (long long)gridDim.x * waves_per_block) {
-    const ushort_t* dyr = dy + row * cols;
-    const ushort_t* hr = h + row * cols;
-    ushort_t* dxr = dx + row * cols;
+    const uint32_t* dyr2 = reinterpret_cast<const uint32_t*>(dy + row * cols);
+    const uint32_t* hr2 = reinterpret_cast<const uint32_t*>(h + row * cols);
+    uint32_t* dxr2 = reinterpret_cast<uint32_t*>(dx + row * cols);
     float mu = mean[row], rs = rstd[row];
 
-    // first pass: c1 = sum(dy*gamma*xhat), c2 = sum(dy*gamma); LDS-accumulate stats
+    // pass 1: c1 = sum(dy*gamma*xhat), c2 = sum(dy*gamma) + register stats
     float c1 = 0.f, c2 = 0.f;
-    for (int i = lane; i < cols; i += 64) {
-      float dyv = bf16_to_f32(dyr[i]);
-      float xhat = (bf16_to_f32(hr[i]) - mu) * rs;
-      float dg = dyv * gamma[i];
-      c1 += dg * xhat;
-      c2 += dg;
-      atomicAdd(&dgamma_local[i], dyv * xhat);
-      atomicAdd(&dbeta_local[i], dyv);
+    int p = 0;
+    for (int c = lane; c < cols2; c += 64, ++p) {
+      uint32_t dyv = dyr2[c], hv = hr2[c];
+      float dy0 = bf16_to_f32((ushort_t)(dyv & 0xffff));
+      float dy1 = bf16_to_f32((ushort_t)(dyv >> 16));
+      float xh0 = (bf16_to_f32((ushort_t)(hv & 0xffff)) - mu) * rs;
+      float xh1 = (bf16_to_f32((ushort_t)(hv >> 16)) - mu) * rs;
+      float dg0 = dy0 * gamma[2 * c], dg1 = dy1 * gamma[2 * c + 1];
+      c1 += dg0 * xh0 + dg1 * xh1;
+      c2 += dg0 + dg1;
+      accg0[p] += dy0 * xh0;
+      accg1[p] += dy1 * xh1;
+      accb0[p] += dy0;
+      accb1[p] += dy1;
     }
     for (int off = 32; off > 0; off >>= 1) {
       c1 += __shfl_down(c1, off);
@@ -400,15 +425,26 @@ extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
     }
     c1 = __shfl(c1, 0) / cols;
     c2 = __shfl(c2, 0) / cols;
-    for (int i = lane; i < cols; i += 64) {
-      float xhat = (bf16_to_f32(hr[i]) - mu) * rs;
-      float dg = bf16_to_f32(dyr[i]) * gamma[i];
-      dxr[i] = f32_to_bf16((dg - c2 - xhat * c1) * rs);
+    for (int c = lane; c < cols2; c += 64) {
+      uint32_t dyv = dyr2[c], hv = hr2[c];
+      float xh0 = (bf16_to_f32((ushort_t)(hv & 0xffff)) - mu) * rs;
+      float xh1 = (bf16_to_f32((ushort_t)(hv >> 16)) - mu) * rs;
+      float dg0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gamma[2 * c];
+      float dg1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gamma[2 * c + 1];
+      float o0 = (dg0 - c2 - xh0 * c1) * rs;
+      float o1 = (dg1 - c2 - xh1 * c1) * rs;
+      dxr2[c] = (uint32_t)f32_to_bf16(o0) | ((uint32_t)f32_to_bf16(o1) << 16);
     }
   }
-  __syncthreads();
-  for (int c = threadIdx.x; c < cols; c += blockDim.x) {
-    if (dgamma_local[c] != 0.f) atomicAdd(&dgamma[c], dgamma_local[c]);
-    if (dbeta_local[c] != 0.f) atomicAdd(&dbeta[c], dbeta_local[c]);
+  int p = 0;
+  for (int c = lane; c < cols2; c += 64, ++p) {
+    if (accg0[p] != 0.f || accb0[p] != 0.f) {
+      atomicAdd(&dgamma[2 * c], accg0[p]);
+      atomicAdd(&dbeta[2 * c], accb0[p]);
+    }
+    if (accg1[p] != 0.f || accb1[p] != 0.f) {
+      atomicAdd(&dgamma[2 * c + 1], accg1[p]);
+      atomicAdd(&dbeta[2 * c + 1], accb1[p]);
+    }
   }
 }
