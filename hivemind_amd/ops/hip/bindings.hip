@@ -168,9 +168,20 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act
   auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   TORCH_CHECK(cols <= 2 * 8 * 256, "bias_gelu_bwd: cols must be <= 4096");
   int blocks = (int)std::min<long long>(rows, 2048);
-  hipLaunchKernelGGL(bias_gelu_bwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
-                     (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
-                     (unsigned short*)dx.data_ptr(), dbias.data_ptr<float>(), rows, cols);
+  int pairs = (int)((cols / 2 + 255) / 256);
+  auto launch = [&](auto kernel) {
+    hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
+                       (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
+                       (unsigned short*)dx.data_ptr(), dbias.data_ptr<float>(), rows, cols);
+  };
+  switch (pairs) {
+    case 1: launch(bias_gelu_bwd_bf16_t<1>); break;
+    case 2: launch(bias_gelu_bwd_bf16_t<2>); break;
+    case 3: launch(bias_gelu_bwd_bf16_t<3>); break;
+    case 4: launch(bias_gelu_bwd_bf16_t<4>); break;
+    case 5: case 6: launch(bias_gelu_bwd_bf16_t<6>); break;
+    default: launch(bias_gelu_bwd_bf16_t<8>); break;
+  }
   return {dx, dbias};
 }
 
@@ -217,11 +228,23 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torc
   auto dbeta = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
   TORCH_CHECK(cols % 2 == 0 && cols <= 2 * 64 * 16, "layernorm_bwd: cols must be even and <= 2048");
   int blocks = (int)std::min<long long>((rows + 3) / 4, 2048);
-  hipLaunchKernelGGL(layernorm_bwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
-                     (const unsigned short*)dy.data_ptr(), (const unsigned short*)h.data_ptr(),
-                     gamma.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
-                     rows, cols);
+  int pairs = (cols / 2 + 63) / 64;
+  auto launch = [&](auto kernel) {
+    hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
+                       (const unsigned short*)dy.data_ptr(), (const unsigned short*)h.data_ptr(),
+                       gamma.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                       rows, cols);
+  };
+  switch (pairs) {
+    case 1: launch(layernorm_bwd_bf16_t<1>); break;
+    case 2: launch(layernorm_bwd_bf16_t<2>); break;
+    case 3: case 4: launch(layernorm_bwd_bf16_t<4>); break;
+    case 5: case 6: launch(layernorm_bwd_bf16_t<6>); break;
+    case 7: case 8: launch(layernorm_bwd_bf16_t<8>); break;
+    case 9: case 10: case 11: case 12: launch(layernorm_bwd_bf16_t<12>); break;
+    default: launch(layernorm_bwd_bf16_t<16>); break;
+  }
   return {dx, dgamma, dbeta};
 }
 

@@ -252,35 +252,34 @@ extern "C" __global__ void bias_gelu_fwd_bf16(const ushort_t* __restrict__ x,
   }
 }
 
-// K9b: backward: dx = dy * gelu'(pre_act) with dbias reduced through LDS.
-// One global atomicAdd per column per BLOCK (not per element): each block
-// strides over whole rows, accumulating its dbias contribution in LDS first
-// (per-element global atomics to `cols` addresses serialize catastrophically
-// -- measured 1.6 ms/call on [65536, 3072] before this restructuring).
-// Requires dynamic LDS of cols * sizeof(float); cols must be even.
-// Each thread owns a fixed set of column-pairs (c = tid + k*blockDim) across
-// every row the block visits, so dbias accumulates in REGISTERS -- no atomics
-// in the row loop at all. One global atomicAdd per owned column at the end
-// (spread over `cols` addresses -> negligible). Supports cols <= 2 * 8 * 256.
-#define BG_MAX_PAIRS 8
-extern "C" __global__ void bias_gelu_bwd_bf16(const ushort_t* __restrict__ dy,
-                                              const ushort_t* __restrict__ pre_act,
-                                              ushort_t* __restrict__ dx,
-                                              float* __restrict__ dbias,  // fp32 accumulators [cols]
-                                              long long rows, long long cols) {
+// K9b: backward: dx = dy * gelu'(pre_act). Each thread owns a fixed set of
+// column-pairs (c = tid + p*blockDim, identical in every row), so dbias
+// accumulates in REGISTERS -- zero atomics inside the row loop; one global
+// atomicAdd per owned column at the end (spread over `cols` addresses).
+// PAIRS is compile-time: runtime-indexed register arrays spill to scratch
+// (cdna_hip_programming.md rule #20). Dispatch in bindings covers
+// cols <= 2 * PAIRS_MAX * 256.
+template <int PAIRS>
+__global__ void bias_gelu_bwd_bf16_t(const ushort_t* __restrict__ dy,
+                                     const ushort_t* __restrict__ pre_act,
+                                     ushort_t* __restrict__ dx,
+                                     float* __restrict__ dbias,
+                                     long long rows, long long cols) {
   const uint32_t* dy2 = reinterpret_cast<const uint32_t*>(dy);
   const uint32_t* pre2 = reinterpret_cast<const uint32_t*>(pre_act);
   uint32_t* dx2 = reinterpret_cast<uint32_t*>(dx);
   long long cols2 = cols >> 1;
 
-  float acc0[BG_MAX_PAIRS], acc1[BG_MAX_PAIRS];
+  float acc0[PAIRS], acc1[PAIRS];
 #pragma unroll
-  for (int p = 0; p < BG_MAX_PAIRS; ++p) acc0[p] = acc1[p] = 0.f;
+  for (int p = 0; p < PAIRS; ++p) acc0[p] = acc1[p] = 0.f;
 
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     long long base = row * cols2;
-    int p = 0;
-    for (long long c = threadIdx.x; c < cols2; c += blockDim.x, ++p) {
+#pragma unroll
+    for (int p = 0; p < PAIRS; ++p) {
+      long long c = threadIdx.x + (long long)p * blockDim.x;
+      if (c >= cols2) break;
       uint32_t dyv = dy2[base + c], pv = pre2[base + c];
       float g0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gelu_tanh_grad(bf16_to_f32((ushort_t)(pv & 0xffff)));
       float g1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gelu_tanh_grad(bf16_to_f32((ushort_t)(pv >> 16)));
@@ -289,10 +288,12 @@ extern "C" __global__ void bias_gelu_bwd_bf16(const ushort_t* __restrict__ dy,
       acc1[p] += g1;
     }
   }
-  int p = 0;
-  for (long long c = threadIdx.x; c < cols2; c += blockDim.x, ++p) {
-    if (acc0[p] != 0.f) atomicAdd(&dbias[2 * c], acc0[p]);
-    if (acc1[p] != 0.f) atomicAdd(&dbias[2 * c + 1], acc1[p]);
+#pragma unroll
+  for (int p = 0; p < PAIRS; ++p) {
+    long long c = threadIdx.x + (long long)p * blockDim.x;
+    if (c >= cols2) break;
+    atomicAdd(&dbias[2 * c], acc0[p]);
+    atomicAdd(&dbias[2 * c + 1], acc1[p]);
   }
 }
 
@@ -372,28 +373,28 @@ extern "C" __global__ void layernorm_fwd_bf16(const ushort_t* __restrict__ x,
 // grid-strides over rows (4 waves x many rows) so the number of global
 // atomics is cols x gridDim, independent of row count. Dynamic LDS:
 // 2 * cols * sizeof(float).
-// Wave-per-row; each lane owns fixed column-pairs (c = lane + k*64, the same
-// in every row), so dgamma/dbeta accumulate in registers across the rows this
-// wave visits; one global atomicAdd per owned column at the end. Vectorized
-// paired bf16 loads throughout. Supports cols <= 2 * 64 * LN_MAX_PAIRS.
-#define LN_MAX_PAIRS 16
-extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
-                                              const ushort_t* __restrict__ h,  // ln input (post-residual)
-                                              const float* __restrict__ gamma,
-                                              const float* __restrict__ mean,
-                                              const float* __restrict__ rstd,
-                                              ushort_t* __restrict__ dx,
-                                              float* __restrict__ dgamma,
-                                              float* __restrict__ dbeta,
-                                              long long rows, int cols) {
+// LayerNorm backward: wave-per-row dx; each lane owns fixed column-pairs
+// (c = lane + p*64, identical in every row), so dgamma/dbeta accumulate in
+// registers across all rows the wave visits; one global atomicAdd per owned
+// column at the end. PAIRS compile-time (rule #20). cols <= 2 * PAIRS * 64.
+template <int PAIRS>
+__global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
+                                     const ushort_t* __restrict__ h,
+                                     const float* __restrict__ gamma,
+                                     const float* __restrict__ mean,
+                                     const float* __restrict__ rstd,
+                                     ushort_t* __restrict__ dx,
+                                     float* __restrict__ dgamma,
+                                     float* __restrict__ dbeta,
+                                     long long rows, int cols) {
   int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
   int waves_per_block = blockDim.x >> 6;
   int cols2 = cols >> 1;
 
-  float accg0[LN_MAX_PAIRS], accg1[LN_MAX_PAIRS], accb0[LN_MAX_PAIRS], accb1[LN_MAX_PAIRS];
+  float accg0[PAIRS], accg1[PAIRS], accb0[PAIRS], accb1[PAIRS];
 #pragma unroll
-  for (int p = 0; p < LN_MAX_PAIRS; ++p) accg0[p] = accg1[p] = accb0[p] = accb1[p] = 0.f;
+  for (int p = 0; p < PAIRS; ++p) accg0[p] = accg1[p] = accb0[p] = accb1[p] = 0.f;
 
   for (long long row = (long long)blockIdx.x * waves_per_block + wave; row < rows;
        row += (long long)gridDim.x * waves_per_block) {
@@ -402,10 +403,11 @@ extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
     uint32_t* dxr2 = reinterpret_cast<uint32_t*>(dx + row * cols);
     float mu = mean[row], rs = rstd[row];
 
-    // pass 1: c1 = sum(dy*gamma*xhat), c2 = sum(dy*gamma) + register stats
     float c1 = 0.f, c2 = 0.f;
-    int p = 0;
-    for (int c = lane; c < cols2; c += 64, ++p) {
+#pragma unroll
+    for (int p = 0; p < PAIRS; ++p) {
+      int c = lane + p * 64;
+      if (c >= cols2) break;
       uint32_t dyv = dyr2[c], hv = hr2[c];
       float dy0 = bf16_to_f32((ushort_t)(dyv & 0xffff));
       float dy1 = bf16_to_f32((ushort_t)(dyv >> 16));
@@ -425,7 +427,10 @@ extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
     }
     c1 = __shfl(c1, 0) / cols;
     c2 = __shfl(c2, 0) / cols;
-    for (int c = lane; c < cols2; c += 64) {
+#pragma unroll
+    for (int p = 0; p < PAIRS; ++p) {
+      int c = lane + p * 64;
+      if (c >= cols2) break;
       uint32_t dyv = dyr2[c], hv = hr2[c];
       float xh0 = (bf16_to_f32((ushort_t)(hv & 0xffff)) - mu) * rs;
       float xh1 = (bf16_to_f32((ushort_t)(hv >> 16)) - mu) * rs;
@@ -436,15 +441,13 @@ extern "C" __global__ void layernorm_bwd_bf16(const ushort_t* __restrict__ dy,
       dxr2[c] = (uint32_t)f32_to_bf16(o0) | ((uint32_t)f32_to_bf16(o1) << 16);
     }
   }
-  int p = 0;
-  for (int c = lane; c < cols2; c += 64, ++p) {
-    if (accg0[p] != 0.f || accb0[p] != 0.f) {
-      atomicAdd(&dgamma[2 * c], accg0[p]);
-      atomicAdd(&dbeta[2 * c], accb0[p]);
-    }
-    if (accg1[p] != 0.f || accb1[p] != 0.f) {
-      atomicAdd(&dgamma[2 * c + 1], accg1[p]);
-      atomicAdd(&dbeta[2 * c + 1], accb1[p]);
-    }
+#pragma unroll
+  for (int p = 0; p < PAIRS; ++p) {
+    int c = lane + p * 64;
+    if (c >= cols2) break;
+    atomicAdd(&dgamma[2 * c], accg0[p]);
+    atomicAdd(&dbeta[2 * c], accb0[p]);
+    atomicAdd(&dgamma[2 * c + 1], accg1[p]);
+    atomicAdd(&dbeta[2 * c + 1], accb1[p]);
   }
 }
