@@ -33,7 +33,7 @@ def main():
     parser.add_argument("--gpus", type=int, default=1)
     parser.add_argument("--steps", type=int, default=24)
     parser.add_argument("--warmup", type=int, default=12)
-    parser.add_argument("--batch", type=int, default=32, help="per-GPU samples per step")
+    parser.add_argument("--batch", type=int, default=128, help="per-GPU samples per step")
     parser.add_argument("--seq-len", type=int, default=512)
     parser.add_argument("--target-batch-size", type=int, default=4096)
     parser.add_argument("--model", type=str, default="albert-base", choices=["albert-base", "albert-large", "tiny"])
