@@ -23,6 +23,43 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 
+def _client_proc_main(proc_idx, dht_endpoint, uid_list, cfg, clients_per_proc, error_queue):
+    try:
+        from hivemind_amd import DHT
+        from hivemind_amd.moe import get_experts
+
+        client_dht = DHT(initial_peers=[dht_endpoint], start=True, client_mode=True)
+        client_experts = get_experts(client_dht, uid_list)
+        threads = []
+
+        def one_client(client_idx):
+            torch.manual_seed(client_idx)
+            for b in range(cfg["batches_per_client"]):
+                expert = client_experts[(client_idx + b) % len(client_experts)]
+                x = torch.randn(cfg["batch_size"], cfg["hidden_dim"])
+                if cfg["backprop"]:
+                    x.requires_grad_(True)
+                    out = expert(x)
+                    out.sum().backward()
+                else:
+                    with torch.no_grad():
+                        expert(x)
+
+        for i in range(clients_per_proc):
+            t = threading.Thread(target=one_client, args=(proc_idx * clients_per_proc + i,))
+            t.start()
+            threads.append(t)
+        for t in threads:
+            t.join()
+        client_dht.shutdown()
+    except Exception:
+        import traceback
+
+        error_queue.put(traceback.format_exc())
+    finally:
+        os._exit(0)
+
+
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--preset", choices=["default", "ffn_forward", "minimal"], default="default")
@@ -67,32 +104,30 @@ def main():
     assert all(e is not None for e in experts)
 
     total_samples = cfg["num_clients"] * cfg["batches_per_client"] * cfg["batch_size"]
-    errors = []
 
-    def client_fn(client_idx: int):
-        torch.manual_seed(client_idx)
-        try:
-            for b in range(cfg["batches_per_client"]):
-                expert = experts[(client_idx + b) % len(experts)]
-                x = torch.randn(cfg["batch_size"], cfg["hidden_dim"])
-                if cfg["backprop"]:
-                    x.requires_grad_(True)
-                    out = expert(x)
-                    out.sum().backward()
-                else:
-                    with torch.no_grad():
-                        expert(x)
-        except Exception as e:
-            errors.append(e)
+    # clients run in separate processes (their serialization must not share the
+    # server's GIL) -- reference benchmark_throughput.py uses client processes too
+    num_client_procs = max(1, min(8, cfg["num_clients"] // 4)) if cfg["num_clients"] >= 4 else 1
+    clients_per_proc = cfg["num_clients"] // num_client_procs
 
+    import torch.multiprocessing as mp
+
+    ctx = mp.get_context("spawn")
+    error_queue = ctx.SimpleQueue()
     t0 = time.perf_counter()
-    threads = [threading.Thread(target=client_fn, args=(i,)) for i in range(cfg["num_clients"])]
-    for t in threads:
-        t.start()
-    for t in threads:
-        t.join()
+    procs = [
+        ctx.Process(
+            target=_client_proc_main,
+            args=(i, dht.endpoint, uids, cfg, clients_per_proc, error_queue),
+        )
+        for i in range(num_client_procs)
+    ]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join()
     elapsed = time.perf_counter() - t0
-    assert not errors, f"client errors: {errors[:3]}"
+    assert error_queue.empty(), f"client errors: {error_queue.get()}"
 
     result = {
         "metric": "MoE server throughput (samples/s)",

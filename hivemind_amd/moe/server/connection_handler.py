@@ -73,15 +73,25 @@ class ConnectionHandler(ServicerBase):
         outputs = await asyncio.wrap_future(future)
         return list(outputs)
 
+    @staticmethod
+    async def _in_executor(func, *args):
+        """numpy (de)serialization releases the GIL -- run it off the event loop
+        so concurrent requests pipeline instead of serializing on the loop."""
+        return await asyncio.get_event_loop().run_in_executor(None, func, *args)
+
     async def rpc_forward(self, request: ExpertRequest, context: RpcContext) -> ExpertResponse:
-        inputs = [deserialize_torch_tensor(t) for t in request.tensors]
+        inputs = await self._in_executor(lambda: [deserialize_torch_tensor(t) for t in request.tensors])
         outputs = await self._process(request.uid, inputs, backward=False)
-        return ExpertResponse(tensors=[serialize_torch_tensor(t.cpu()) for t in outputs])
+        return ExpertResponse(
+            tensors=await self._in_executor(lambda: [serialize_torch_tensor(t.cpu()) for t in outputs])
+        )
 
     async def rpc_backward(self, request: ExpertRequest, context: RpcContext) -> ExpertResponse:
-        inputs = [deserialize_torch_tensor(t) for t in request.tensors]
+        inputs = await self._in_executor(lambda: [deserialize_torch_tensor(t) for t in request.tensors])
         outputs = await self._process(request.uid, inputs, backward=True)
-        return ExpertResponse(tensors=[serialize_torch_tensor(t.cpu()) for t in outputs])
+        return ExpertResponse(
+            tensors=await self._in_executor(lambda: [serialize_torch_tensor(t.cpu()) for t in outputs])
+        )
 
     async def rpc_forward_stream(
         self, requests: AsyncIterator[ExpertRequest], context: RpcContext
