@@ -71,6 +71,25 @@ class AveragingError(Exception):
 
 
 @dataclass
+class GroupMetadata:
+    """Per-peer metadata distributed with group assembly (gather side channel)."""
+
+    bandwidths: list
+    modes: list
+    user_gathered: dict
+    dist_infos: list
+    weights: list
+
+
+class _null_actx:
+    async def __aenter__(self):
+        return None
+
+    async def __aexit__(self, *args):
+        return False
+
+
+@dataclass
 class DownloadRequest(RpcMessage):
     pass
 
@@ -331,65 +350,91 @@ class DecentralizedAverager(ServicerBase):
         except BaseException as e:
             step.set_exception(e if isinstance(e, Exception) else AveragingError(repr(e)))
 
+    def _parse_group_metadata(self, group_info: GroupInfo, weight: float) -> "GroupMetadata":
+        """Unpack per-peer gathered metadata: [bandwidth, mode, user_gather, dist_info, weight]."""
+        bandwidths, modes, user_gathered_raw, dist_infos, weights = [], [], [], [], []
+        for raw in group_info.gathered:
+            meta = MSGPackSerializer.loads(raw) if raw else {}
+            bandwidths.append(meta.get("bandwidth"))
+            modes.append(AveragingMode(meta.get("mode", AveragingMode.NODE.value)))
+            user_gathered_raw.append(meta.get("gather"))
+            dist_infos.append(tuple(meta["dist"]) if meta.get("dist") else None)
+            weights.append(float(meta.get("weight", 1.0)))
+        my_index = group_info.peer_ids.index(self.peer_id)
+        weights[my_index] = weight  # trust our own latest weight
+        return GroupMetadata(
+            bandwidths=bandwidths,
+            modes=modes,
+            user_gathered=dict(zip(group_info.peer_ids, user_gathered_raw)),
+            dist_infos=dist_infos,
+            weights=weights,
+        )
+
+    def _can_use_rccl(self, meta: "GroupMetadata") -> bool:
+        return (
+            self.use_rccl_when_available
+            and group_matches_world(meta.dist_infos)
+            and all(m == AveragingMode.NODE for m in meta.modes)
+        )
+
+    async def _average_tensors_with_group(
+        self,
+        tensors: Sequence[torch.Tensor],
+        group_info: GroupInfo,
+        weight: float,
+        meta: "GroupMetadata",
+        group_id: Optional[bytes] = None,
+        take_lock: bool = True,
+    ):
+        """Average an arbitrary tensor list with the group, picking the data plane:
+        bucketed RCCL-over-xGMI when the group is exactly our torch.distributed
+        world, else the RPC butterfly. Subclasses (PowerSGD) call this several
+        times per round with distinct group_id suffixes."""
+        lock_ctx = enter_asynchronously(self.lock_averaged_tensors) if take_lock else _null_actx()
+        if self._can_use_rccl(meta):
+            self.last_data_plane = "rccl"
+            async with lock_ctx:
+                await asyncio.get_event_loop().run_in_executor(
+                    None,
+                    DistributedAllReduceRunner(
+                        tensors,
+                        weight,
+                        wire_dtype=self.allreduce_wire_dtype,
+                        averaging_alpha=self.averaging_alpha,
+                    ).run,
+                )
+            return
+        self.last_data_plane = "rpc"
+        download_bandwidths = [
+            (0.0 if mode == AveragingMode.CLIENT else bw) for mode, bw in zip(meta.modes, meta.bandwidths)
+        ]
+        total_size = sum(t.numel() for t in tensors)
+        peer_fractions = await asyncio.get_event_loop().run_in_executor(
+            None, load_balance_peers, total_size, download_bandwidths, self.part_size_bytes
+        )
+        async with lock_ctx:
+            await self._run_allreduce_inplace_(
+                tensors, group_info, group_id=group_id, peer_fractions=peer_fractions, weight=weight, modes=meta.modes
+            )
+
     async def _aggregate_with_group(self, group_info: GroupInfo, weight: float) -> Dict[PeerID, Any]:
         """Run the data plane for one assembled group (reference averager.py:514-562)."""
         logger.debug(f"{self.prefix}@{self.peer_id}: entering aggregation, group={group_info.group_id.hex()[:8]}")
         try:
-            num_peers = len(group_info.peer_ids)
-            # unpack per-peer metadata: [bandwidth, mode, user_gather, dist_info]
-            bandwidths, modes, user_gathered_raw, dist_infos, weights = [], [], [], [], []
-            for raw in group_info.gathered:
-                meta = MSGPackSerializer.loads(raw) if raw else {}
-                bandwidths.append(meta.get("bandwidth"))
-                mode_val = meta.get("mode", AveragingMode.NODE.value)
-                modes.append(AveragingMode(mode_val))
-                user_gathered_raw.append(meta.get("gather"))
-                dist_infos.append(tuple(meta["dist"]) if meta.get("dist") else None)
-                weights.append(float(meta.get("weight", 1.0)))
-            user_gathered = dict(zip(group_info.peer_ids, user_gathered_raw))
-
-            my_index = group_info.peer_ids.index(self.peer_id)
-            weights[my_index] = weight  # trust our own latest weight
-
-            if (
-                self.use_rccl_when_available
-                and group_matches_world(dist_infos)
-                and all(m == AveragingMode.NODE for m in modes)
-            ):
-                self.last_data_plane = "rccl"
-                async with enter_asynchronously(self.lock_averaged_tensors):
-                    await asyncio.get_event_loop().run_in_executor(
-                        None,
-                        DistributedAllReduceRunner(
-                            self._averaged_tensors,
-                            weight,
-                            wire_dtype=self.allreduce_wire_dtype,
-                            averaging_alpha=self.averaging_alpha,
-                        ).run,
-                    )
-                self._state_updated.set()
-                return user_gathered
-
-            # RPC butterfly path
-            self.last_data_plane = "rpc"
-            download_bandwidths = [
-                (0.0 if mode == AveragingMode.CLIENT else bw) for mode, bw in zip(modes, bandwidths)
-            ]
-            peer_fractions = await asyncio.get_event_loop().run_in_executor(
-                None, load_balance_peers, self.total_size, download_bandwidths, self.part_size_bytes
-            )
-            async with enter_asynchronously(self.lock_averaged_tensors):
-                await self._run_allreduce_inplace_(
-                    self._averaged_tensors, group_info, peer_fractions=peer_fractions, weight=weight, modes=modes
-                )
-            return user_gathered
+            meta = self._parse_group_metadata(group_info, weight)
+            await self._aggregate_tensors_with_group(group_info, weight, meta)
+            self._state_updated.set()
+            return meta.user_gathered
         except BaseException as e:
             if isinstance(e, asyncio.CancelledError):
                 raise
             logger.debug(f"aggregation failed: {e!r}")
             raise AveragingError(f"aggregation stage failed: {e!r}") from e
-        finally:
-            pass
+
+    async def _aggregate_tensors_with_group(self, group_info: GroupInfo, weight: float, meta: "GroupMetadata"):
+        """Default round: average the averager's own tensors. Override point for
+        multi-phase schemes (PowerSGD runs two chained rounds)."""
+        await self._average_tensors_with_group(self._averaged_tensors, group_info, weight, meta)
 
     async def _run_allreduce_inplace_(
         self,
