@@ -104,6 +104,7 @@ class DecentralizedAverager(ServicerBase):
     """Averages a fixed-schema list of tensors with dynamically matched groups of peers."""
 
     _matchmaking: Matchmaking
+    _allreduce_runner_class = AllReduceRunner  # override point for fault-injection tests
 
     def __init__(
         self,
@@ -448,7 +449,7 @@ class DecentralizedAverager(ServicerBase):
     ):
         """Run one allreduce round on the RPC butterfly (reference averager.py:537-562)."""
         group_id = group_id if group_id is not None else group_info.group_id
-        runner = AllReduceRunner(
+        runner = self._allreduce_runner_class(
             p2p=self._p2p,
             servicer_type=type(self),
             namespace=self.prefix,

@@ -1,0 +1,78 @@
+"""CLI smoke tests (reference tests/test_cli_scripts.py, test_start_server.py)."""
+
+import re
+import subprocess
+import sys
+import time
+
+import pytest
+
+
+def test_run_dht_starts_and_peers_connect():
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "hivemind_amd.hivemind_cli.run_dht", "--refresh_period", "1"],
+        stderr=subprocess.PIPE,
+        text=True,
+        encoding="utf-8",
+    )
+    try:
+        endpoint = None
+        deadline = time.monotonic() + 30
+        first_lines = []
+        while time.monotonic() < deadline:
+            line = proc.stderr.readline()
+            first_lines.append(line)
+            match = re.search(r"Running a DHT instance at (\S+)", line)
+            if match:
+                endpoint = match.group(1)
+                break
+        assert endpoint is not None, f"no readiness line in: {first_lines}"
+
+        # a second peer joins via --initial_peers
+        proc2 = subprocess.Popen(
+            [sys.executable, "-m", "hivemind_amd.hivemind_cli.run_dht", "--initial_peers", endpoint,
+             "--refresh_period", "1"],
+            stderr=subprocess.PIPE,
+            text=True,
+            encoding="utf-8",
+        )
+        try:
+            saw_peer = False
+            deadline = time.monotonic() + 30
+            while time.monotonic() < deadline:
+                line = proc2.stderr.readline()
+                if re.search(r"DHT status: [1-9]\d* known peers", line):
+                    saw_peer = True
+                    break
+            assert saw_peer, "second peer never saw the first"
+        finally:
+            proc2.terminate()
+            proc2.wait(10)
+    finally:
+        proc.terminate()
+        proc.wait(10)
+
+
+def test_run_server_starts():
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "hivemind_amd.hivemind_cli.run_server",
+         "--num_experts", "2", "--expert_pattern", "clitest.[0:16]",
+         "--expert_cls", "ffn", "--hidden_dim", "16", "--device", "cpu"],
+        stderr=subprocess.PIPE,
+        text=True,
+        encoding="utf-8",
+    )
+    try:
+        ready = False
+        deadline = time.monotonic() + 45
+        lines = []
+        while time.monotonic() < deadline:
+            line = proc.stderr.readline()
+            lines.append(line)
+            if "Server started with 2 experts" in line:
+                ready = True
+                break
+        assert ready, f"server never became ready: {lines[-5:]}"
+    finally:
+        proc.terminate()
+        proc.wait(10)
