@@ -1,0 +1,101 @@
+#!/usr/bin/env python3
+"""Averaging round benchmark (reference benchmarks/benchmark_averaging.py:38-80;
+BASELINE.json config: 2 CPU peers, 1M-param tensor, no compression).
+
+  python benchmarks/benchmark_averaging.py --num_peers 2 --num_params 1000000
+  python benchmarks/benchmark_averaging.py --num_peers 16 --target_group_size 4 --num_rounds 3
+"""
+
+import argparse
+import json
+import os
+import sys
+import threading
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch
+
+
+def main():
+    parser = argparse.ArgumentParser()
+    parser.add_argument("--num_peers", type=int, default=2)
+    parser.add_argument("--target_group_size", type=int, default=None)
+    parser.add_argument("--num_params", type=int, default=1_000_000)
+    parser.add_argument("--num_rounds", type=int, default=3)
+    parser.add_argument("--compression", choices=["none", "fp16", "blockwise8"], default="none")
+    parser.add_argument("--matchmaking_time", type=float, default=2.0)
+    args = parser.parse_args()
+    target_group_size = args.target_group_size or args.num_peers
+
+    from hivemind_amd import DHT
+    from hivemind_amd.averaging import DecentralizedAverager
+    from hivemind_amd.compression import BlockwiseQuantization, Float16Compression, NoCompression
+
+    compression = {"none": NoCompression(), "fp16": Float16Compression(), "blockwise8": BlockwiseQuantization()}[
+        args.compression
+    ]
+
+    peers = [DHT(start=True)]
+    for _ in range(args.num_peers - 1):
+        peers.append(DHT(initial_peers=[peers[0].endpoint], start=True))
+
+    torch.manual_seed(42)
+    averagers = [
+        DecentralizedAverager(
+            [torch.randn(args.num_params // 4) for _ in range(4)],
+            dht,
+            start=True,
+            prefix="bench_avg",
+            target_group_size=target_group_size,
+            min_group_size=2,
+            min_matchmaking_time=args.matchmaking_time,
+            request_timeout=1.0,
+            compression=compression,
+        )
+        for dht in peers
+    ]
+
+    round_times, successes = [], 0
+    for round_idx in range(args.num_rounds):
+        t0 = time.perf_counter()
+        controls = [avg.step(wait=False, timeout=120) for avg in averagers]
+        ok = 0
+        for c in controls:
+            try:
+                if c.result(150) is not None:
+                    ok += 1
+            except Exception:
+                pass
+        round_times.append(time.perf_counter() - t0)
+        successes += ok
+        print(f"round {round_idx}: {ok}/{args.num_peers} peers succeeded in {round_times[-1]:.2f}s", file=sys.stderr)
+
+    bytes_per_peer = args.num_params * 4
+    result = {
+        "metric": "averaging round wall time",
+        "num_peers": args.num_peers,
+        "target_group_size": target_group_size,
+        "num_params": args.num_params,
+        "compression": args.compression,
+        "mean_round_s": round(sum(round_times) / len(round_times), 3),
+        "success_rate": round(successes / (args.num_rounds * args.num_peers), 4),
+        "effective_goodput_MBps": round(bytes_per_peer / (sum(round_times) / len(round_times)) / 1e6, 1),
+    }
+    print(json.dumps(result), flush=True)
+    for avg in averagers:
+        avg.shutdown()
+    for p in peers:
+        p.shutdown()
+
+
+if __name__ == "__main__":
+    try:
+        main()
+    except BaseException:
+        import traceback
+
+        traceback.print_exc()
+        os._exit(1)
+    os._exit(0)

@@ -131,7 +131,9 @@ async def traverse_dht(
                     return
                 _, main_query, peer = best
                 heapq.heappop(candidates[main_query])
-                # pack additional active queries that would also like this peer
+                # pack additional active queries that would also like this peer;
+                # the peer counts as visited for every packed query so no other
+                # worker re-queries it on their behalf
                 packed = [main_query]
                 for q in queries:
                     if len(packed) >= queries_per_call:
@@ -139,6 +141,11 @@ async def traverse_dht(
                     if q is main_query or q in finished:
                         continue
                     packed.append(q)
+                for q in packed:
+                    visited[q].add(peer)
+                    # lazily drop the peer from q's own candidate heap
+                    candidates[q] = [(d, uid) for d, uid in candidates[q] if uid != peer]
+                    heapq.heapify(candidates[q])
             try:
                 responses = await get_neighbors(peer, packed)
             except Exception:
