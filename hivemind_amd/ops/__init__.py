@@ -120,6 +120,93 @@ def fused_bias_gelu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# Llama-family fused ops: RMSNorm, SwiGLU, RoPE
+# ---------------------------------------------------------------------------
+
+
+class _FusedRMSNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, gamma, eps):
+        y, rstd = hip_ops().rmsnorm_fwd(x.contiguous(), gamma.contiguous(), eps)
+        ctx.save_for_backward(x, gamma, rstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, gamma, rstd = ctx.saved_tensors
+        dx, dgamma = hip_ops().rmsnorm_bwd(dy.contiguous(), x.contiguous(), gamma, rstd)
+        return dx, dgamma.to(gamma.dtype), None
+
+
+def fused_rmsnorm(x: torch.Tensor, gamma: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    """y = x * gamma / sqrt(mean(x^2) + eps), one HIP kernel on GPU."""
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _FusedRMSNorm.apply(x, gamma, eps)
+    xf = x.float()
+    rstd = torch.rsqrt(xf.pow(2).mean(-1, keepdim=True) + eps)
+    return (xf * rstd * gamma.float()).to(x.dtype)
+
+
+class _FusedSwiGLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, gate, up):
+        ctx.save_for_backward(gate, up)
+        return hip_ops().swiglu_fwd(gate.contiguous(), up.contiguous())
+
+    @staticmethod
+    def backward(ctx, dy):
+        gate, up = ctx.saved_tensors
+        dgate, dup = hip_ops().swiglu_bwd(dy.contiguous(), gate.contiguous(), up.contiguous())
+        return dgate, dup
+
+
+def fused_swiglu(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    """silu(gate) * up -- the Llama MLP activation, fused."""
+    if gate.is_cuda and gate.dtype == torch.bfloat16:
+        return _FusedSwiGLU.apply(gate, up)
+    return torch.nn.functional.silu(gate.float()).mul(up.float()).to(gate.dtype)
+
+
+class _FusedRoPE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, cos_table, sin_table):
+        ctx.save_for_backward(cos_table, sin_table)
+        return hip_ops().rope_apply(x.contiguous(), cos_table, sin_table, 1.0)
+
+    @staticmethod
+    def backward(ctx, dy):
+        cos_table, sin_table = ctx.saved_tensors
+        return hip_ops().rope_apply(dy.contiguous(), cos_table, sin_table, -1.0), None, None
+
+
+def fused_rope(x: torch.Tensor, cos_table: torch.Tensor, sin_table: torch.Tensor) -> torch.Tensor:
+    """Rotate-half RoPE with host-precomputed fp32 cos/sin tables [tokens, head_dim/2].
+
+    x: [..., tokens, heads, head_dim] flattened so that the trailing dims are
+    (heads, head_dim) and tokens iterate over everything before them.
+    """
+    if x.is_cuda and x.dtype == torch.bfloat16:
+        return _FusedRoPE.apply(x, cos_table, sin_table)
+    half = x.shape[-1] // 2
+    x1, x2 = x[..., :half].float(), x[..., half:].float()
+    shape = [1] * (x.ndim - 3) + [cos_table.shape[0], 1, half]
+    cos = cos_table.reshape(shape)
+    sin = sin_table.reshape(shape)
+    out1 = x1 * cos - x2 * sin
+    out2 = x2 * cos + x1 * sin
+    return torch.cat([out1, out2], dim=-1).to(x.dtype)
+
+
+def build_rope_tables(seq_len: int, head_dim: int, base: float = 500000.0, device="cpu"):
+    """Host-precomputed RoPE tables (guide: never compute trig on-device)."""
+    half = head_dim // 2
+    inv_freq = 1.0 / (base ** (torch.arange(0, half, dtype=torch.float32) / half))
+    t = torch.arange(seq_len, dtype=torch.float32)
+    freqs = torch.outer(t, inv_freq)  # [seq_len, half]
+    return freqs.cos().to(device), freqs.sin().to(device)
+
+
+# ---------------------------------------------------------------------------
 # averaging primitives
 # ---------------------------------------------------------------------------
 

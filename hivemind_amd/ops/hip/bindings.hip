@@ -7,6 +7,7 @@
 #include <hip/hip_runtime.h>
 
 #include "elementwise.hip"
+#include "llama_ops.hip"
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -248,7 +249,94 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torc
   return {dx, dgamma, dbeta};
 }
 
+// ------------------------------------------------------------ llama kernels
+
+std::vector<torch::Tensor> rmsnorm_fwd(torch::Tensor x, torch::Tensor gamma, double eps) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(gamma);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && gamma.scalar_type() == torch::kFloat32);
+  int cols = (int)x.size(-1);
+  TORCH_CHECK(cols % 2 == 0, "rmsnorm: cols must be even");
+  long long rows = x.numel() / cols;
+  auto y = torch::empty_like(x);
+  auto rstd = torch::empty({rows}, x.options().dtype(torch::kFloat32));
+  int blocks = (int)std::min<long long>(rows, 4096);
+  hipLaunchKernelGGL(rmsnorm_fwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
+                     (const unsigned short*)x.data_ptr(), gamma.data_ptr<float>(),
+                     (unsigned short*)y.data_ptr(), rstd.data_ptr<float>(), (float)eps, rows, cols);
+  return {y, rstd};
+}
+
+std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch::Tensor gamma, torch::Tensor rstd) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(x);
+  int cols = (int)dy.size(-1);
+  TORCH_CHECK(cols % 2 == 0 && cols <= 2 * 16 * 256, "rmsnorm_bwd: cols must be even and <= 8192");
+  long long rows = dy.numel() / cols;
+  auto dx = torch::empty_like(dy);
+  auto dgamma = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  int blocks = (int)std::min<long long>(rows, 2048);
+  int pairs = (cols / 2 + 255) / 256;
+  auto launch = [&](auto kernel) {
+    hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
+                       (const unsigned short*)dy.data_ptr(), (const unsigned short*)x.data_ptr(),
+                       gamma.data_ptr<float>(), rstd.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), rows, cols);
+  };
+  switch (pairs) {
+    case 1: launch(rmsnorm_bwd_bf16_t<1>); break;
+    case 2: launch(rmsnorm_bwd_bf16_t<2>); break;
+    case 3: case 4: launch(rmsnorm_bwd_bf16_t<4>); break;
+    case 5: case 6: case 7: case 8: launch(rmsnorm_bwd_bf16_t<8>); break;
+    default: launch(rmsnorm_bwd_bf16_t<16>); break;
+  }
+  return {dx, dgamma};
+}
+
+torch::Tensor swiglu_fwd(torch::Tensor gate, torch::Tensor up) {
+  CHECK_GPU(gate); CHECK_CONTIG(gate); CHECK_CONTIG(up);
+  TORCH_CHECK(gate.numel() == up.numel() && gate.numel() % 2 == 0);
+  auto out = torch::empty_like(gate);
+  long long n2 = gate.numel() / 2;
+  hipLaunchKernelGGL(swiglu_fwd_bf16, dim3(grid_1d(n2)), dim3(256), 0, current_stream(),
+                     (const unsigned short*)gate.data_ptr(), (const unsigned short*)up.data_ptr(),
+                     (unsigned short*)out.data_ptr(), n2);
+  return out;
+}
+
+std::vector<torch::Tensor> swiglu_bwd(torch::Tensor dy, torch::Tensor gate, torch::Tensor up) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(gate); CHECK_CONTIG(up);
+  auto dgate = torch::empty_like(gate);
+  auto dup = torch::empty_like(up);
+  long long n2 = gate.numel() / 2;
+  hipLaunchKernelGGL(swiglu_bwd_bf16, dim3(grid_1d(n2)), dim3(256), 0, current_stream(),
+                     (const unsigned short*)dy.data_ptr(), (const unsigned short*)gate.data_ptr(),
+                     (const unsigned short*)up.data_ptr(), (unsigned short*)dgate.data_ptr(),
+                     (unsigned short*)dup.data_ptr(), n2);
+  return {dgate, dup};
+}
+
+torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cos_table, torch::Tensor sin_table, double direction) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(cos_table); CHECK_CONTIG(sin_table);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16);
+  int head_dim = (int)x.size(-1);
+  int heads = (int)x.size(-2);
+  long long tokens = x.numel() / ((long long)heads * head_dim);
+  TORCH_CHECK(head_dim % 2 == 0);
+  auto out = torch::empty_like(x);
+  int seq_len = (int)cos_table.size(0);
+  long long total = tokens * heads * (head_dim / 2);
+  hipLaunchKernelGGL(rope_bf16, dim3(grid_1d(total)), dim3(256), 0, current_stream(),
+                     (const unsigned short*)x.data_ptr(), cos_table.data_ptr<float>(),
+                     sin_table.data_ptr<float>(), (unsigned short*)out.data_ptr(),
+                     (float)direction, tokens, seq_len, heads, head_dim);
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
+  m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");
+  m.def("swiglu_fwd", &swiglu_fwd, "silu(gate) * up");
+  m.def("swiglu_bwd", &swiglu_bwd, "SwiGLU backward");
+  m.def("rope_apply", &rope_apply, "rotary embedding (direction=+1 fwd, -1 bwd)");
   m.def("apply_delta_", &apply_delta_, "tensor += alpha * delta (in place)");
   m.def("weighted_accumulate_", &weighted_accumulate_, "acc += w * x (in place)");
   m.def("compress_fp16", &compress_fp16_gpu, "clamp + cast to fp16");

@@ -197,3 +197,107 @@ def test_albert_model_step():
         losses.append(loss.item())
     assert all(torch.isfinite(torch.tensor(losses))), losses
     assert losses[-1] < losses[0], f"loss did not decrease: {losses}"
+
+
+@requires_gpu
+def test_rmsnorm_forward_backward():
+    from hivemind_amd.ops import fused_rmsnorm
+
+    torch.manual_seed(0)
+    for H in (768, 4096):
+        B = 32
+        x = torch.randn(B, H, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+        gamma = (torch.rand(H, device="cuda", dtype=torch.float32) + 0.5).requires_grad_(True)
+        y = fused_rmsnorm(x, gamma, eps=1e-5)
+
+        x_ref = x.detach().float().requires_grad_(True)
+        g_ref = gamma.detach().clone().requires_grad_(True)
+        rstd = torch.rsqrt(x_ref.pow(2).mean(-1, keepdim=True) + 1e-5)
+        y_ref = x_ref * rstd * g_ref
+        assert torch.allclose(y.float(), y_ref, atol=3e-2), (H, (y.float() - y_ref).abs().max())
+
+        dy = torch.randn_like(y_ref)
+        y_ref.backward(dy)
+        y.backward(dy.bfloat16())
+        assert torch.allclose(x.grad.float(), x_ref.grad, atol=5e-2), (H, (x.grad.float() - x_ref.grad).abs().max())
+        assert torch.allclose(gamma.grad, g_ref.grad, atol=0.2, rtol=2e-2), (H, (gamma.grad - g_ref.grad).abs().max())
+
+
+@requires_gpu
+def test_swiglu_forward_backward():
+    from hivemind_amd.ops import fused_swiglu
+
+    torch.manual_seed(0)
+    gate = torch.randn(64, 1024, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    up = torch.randn(64, 1024, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    out = fused_swiglu(gate, up)
+
+    g_ref = gate.detach().float().requires_grad_(True)
+    u_ref = up.detach().float().requires_grad_(True)
+    out_ref = torch.nn.functional.silu(g_ref) * u_ref
+    assert torch.allclose(out.float(), out_ref, atol=3e-2)
+
+    dy = torch.randn_like(out_ref)
+    out_ref.backward(dy)
+    out.backward(dy.bfloat16())
+    assert torch.allclose(gate.grad.float(), g_ref.grad, atol=5e-2)
+    assert torch.allclose(up.grad.float(), u_ref.grad, atol=5e-2)
+
+
+@requires_gpu
+def test_rope_forward_backward():
+    from hivemind_amd.ops import build_rope_tables, fused_rope
+
+    torch.manual_seed(0)
+    B, S, heads, hd = 2, 64, 8, 128
+    cos, sin = build_rope_tables(S, hd, device="cuda")
+    x = torch.randn(B, S, heads, hd, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    y = fused_rope(x, cos, sin)
+
+    x_ref = x.detach().float().requires_grad_(True)
+    half = hd // 2
+    x1, x2 = x_ref[..., :half], x_ref[..., half:]
+    c = cos.view(1, S, 1, half)
+    s = sin.view(1, S, 1, half)
+    y_ref = torch.cat([x1 * c - x2 * s, x2 * c + x1 * s], dim=-1)
+    assert torch.allclose(y.float(), y_ref, atol=2e-2), (y.float() - y_ref).abs().max()
+
+    dy = torch.randn_like(y_ref)
+    y_ref.backward(dy)
+    y.backward(dy.bfloat16())
+    assert torch.allclose(x.grad.float(), x_ref.grad, atol=3e-2)
+    # rotation preserves norms
+    assert torch.allclose(y.float().norm(), x.detach().float().norm(), rtol=1e-2)
+
+
+@requires_gpu
+def test_llama_model_step():
+    from hivemind_amd.models import LlamaConfig, LlamaForCausalLM
+    from hivemind_amd.ops import FusedAdamW, bind_grad
+
+    torch.manual_seed(0)
+    config = LlamaConfig.llama_1b()
+    config.num_hidden_layers = 4  # keep the smoke fast
+    model = LlamaForCausalLM(config).cuda()
+    masters, params = {}, []
+    for p in model.parameters():
+        master = torch.nn.Parameter(p.detach().float().clone())
+        masters[master] = p
+        params.append(master)
+    opt = FusedAdamW(params, lr=5e-5)
+    for master, live in masters.items():
+        opt.set_mirror(master, live.data)
+
+    ids = torch.randint(0, config.vocab_size, (2, 256), device="cuda")
+    losses = []
+    for _ in range(4):
+        loss, _ = model(ids, labels=ids)
+        loss.backward()
+        for master, live in masters.items():
+            bind_grad(master, live.grad)
+        opt.step()
+        for live in model.parameters():
+            live.grad = None
+        losses.append(loss.item())
+    assert all(torch.isfinite(torch.tensor(losses))), losses
+    assert losses[-1] < losses[0], losses
