@@ -36,7 +36,10 @@ def main():
     parser.add_argument("--batch", type=int, default=128, help="per-GPU samples per step")
     parser.add_argument("--seq-len", type=int, default=512)
     parser.add_argument("--target-batch-size", type=int, default=4096)
-    parser.add_argument("--model", type=str, default="albert-base", choices=["albert-base", "albert-large", "tiny"])
+    parser.add_argument("--model", type=str, default="albert-base",
+                        choices=["albert-base", "albert-large", "tiny", "llama-8b", "llama-1b", "llama-tiny"])
+    parser.add_argument("--dpu", action="store_true",
+                        help="delayed parameter updates: overlap averaging + optimizer step with compute")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -57,23 +60,31 @@ def main():
         device = torch.device("cpu")
 
     from hivemind_amd import DHT, Optimizer
-    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM
+    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM, LlamaConfig, LlamaForCausalLM
     from hivemind_amd.ops import FusedAdamW, hip_available
 
     if use_gpu and not hip_available():
         raise RuntimeError("HIP ops extension missing on a GPU node -- refusing to run a fallback bench")
 
+    is_llama = args.model.startswith("llama")
     if args.model == "albert-base":
         config = AlbertConfig.base()
     elif args.model == "albert-large":
         config = AlbertConfig.large()
-    else:
+    elif args.model == "tiny":
         config = AlbertConfig.tiny()
+    elif args.model == "llama-8b":
+        config = LlamaConfig.llama_3_8b()
+    elif args.model == "llama-1b":
+        config = LlamaConfig.llama_1b()
+    else:
+        config = LlamaConfig.tiny()
     if not use_gpu:
         config.dtype = torch.float32  # CPU bf16 matmuls are pathologically slow
 
     torch.manual_seed(1234 + rank)
-    model = AlbertForMaskedLM(config).to(device)
+    model_cls = LlamaForCausalLM if is_llama else AlbertForMaskedLM
+    model = model_cls(config).to(device)
     log(f"model: {args.model}, {model.num_parameters()/1e6:.1f}M params, device={device}")
 
     # ---------------------------------------------------------------- swarm
@@ -101,6 +112,8 @@ def main():
         matchmaking_time=1.0 if world_size > 1 else 0.5,
         averaging_timeout=120.0,
         reuse_grad_buffers=False,
+        delay_optimizer_step=args.dpu,
+        delay_grad_averaging=args.dpu,
         average_state_every=4,
         averager_opts=dict(
             request_timeout=0.5,
@@ -114,9 +127,10 @@ def main():
     def make_batch():
         input_ids = torch.randint(0, config.vocab_size, (args.batch, args.seq_len), device=device)
         labels = input_ids.clone()
-        # mask 15% of positions for MLM, rest ignored in the loss
-        mask = torch.rand(labels.shape, device=device) > 0.15
-        labels[mask] = -100
+        if not is_llama:
+            # mask 15% of positions for MLM, rest ignored in the loss
+            mask = torch.rand(labels.shape, device=device) > 0.15
+            labels[mask] = -100
         return input_ids, labels
 
     def one_step():
@@ -157,7 +171,8 @@ def main():
 
     if rank == 0:
         result = {
-            "metric": "samples/sec (whole swarm) ALBERT-base hivemind.Optimizer",
+            "metric": ("samples/sec (whole swarm) ALBERT-base hivemind.Optimizer" if not is_llama
+                       else f"samples/sec (whole swarm) {args.model} hivemind.Optimizer + DPU"),
             "value": round(samples_per_sec, 2),
             "unit": "samples/s",
             "n_gpus": n_gpus,
@@ -171,6 +186,7 @@ def main():
             "data": "synthetic",
             "config": {
                 "model": args.model,
+                "dpu": args.dpu,
                 "global_batch": args.target_batch_size,
                 "per_gpu_batch": args.batch,
                 "seq_len": args.seq_len,
