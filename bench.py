@@ -50,9 +50,12 @@ def main():
 
     import torch.distributed as dist
 
+    bench_pg = None  # gloo subgroup for timing barriers: never touches the RCCL
+    # communicators used by the (possibly background) averaging collectives
     if world_size > 1:
         backend = "nccl" if use_gpu else "gloo"
         dist.init_process_group(backend, rank=rank, world_size=world_size)
+        bench_pg = dist.new_group(backend="gloo")
     if use_gpu:
         torch.cuda.set_device(local_rank)
         device = torch.device("cuda", local_rank)
@@ -145,7 +148,7 @@ def main():
         if use_gpu:
             torch.cuda.synchronize(device)
         if world_size > 1:
-            dist.barrier()
+            dist.barrier(group=bench_pg)
 
     log(f"warmup: {args.warmup} steps")
     for i in range(args.warmup):
@@ -159,10 +162,11 @@ def main():
     sync()
     elapsed = time.perf_counter() - t0
 
-    # take the max elapsed across ranks (slowest peer defines the swarm rate)
+    # take the max elapsed across ranks (slowest peer defines the swarm rate);
+    # runs on the gloo subgroup so it cannot interleave with RCCL averaging
     if world_size > 1:
-        t = torch.tensor([elapsed], dtype=torch.float64, device=device if use_gpu else "cpu")
-        dist.all_reduce(t, op=dist.ReduceOp.MAX)
+        t = torch.tensor([elapsed], dtype=torch.float64)
+        dist.all_reduce(t, op=dist.ReduceOp.MAX, group=bench_pg)
         elapsed = float(t.item())
 
     total_samples = args.steps * args.batch * n_gpus

@@ -178,6 +178,14 @@ class DecentralizedAverager(ServicerBase):
         self._initial_group_bits = initial_group_bits
         self._running_groups: Dict[bytes, asyncio.Future] = {}
         self.last_data_plane: Optional[str] = None  # "rccl" | "rpc" after a round
+        # dedicated communicator: concurrent averagers (grad + state) must never
+        # interleave collectives on one communicator -- cross-rank ordering of
+        # threads is not deterministic (SURVEY.md §7 "hard parts"). new_group()
+        # is collective, so every rank must construct its averagers in the same
+        # order (true for Optimizer and for symmetric training scripts).
+        import torch.distributed as dist
+
+        self._dist_process_group = dist.new_group() if (dist.is_available() and dist.is_initialized()) else None
         self._state_updated = asyncio.Event()
         self._declare_state_task: Optional[asyncio.Task] = None
         self._ready = concurrent.futures.Future()
@@ -400,6 +408,7 @@ class DecentralizedAverager(ServicerBase):
                     DistributedAllReduceRunner(
                         tensors,
                         weight,
+                        process_group=self._dist_process_group,
                         wire_dtype=self.allreduce_wire_dtype,
                         averaging_alpha=self.averaging_alpha,
                     ).run,
