@@ -8,6 +8,7 @@
 
 #include "elementwise.hip"
 #include "llama_ops.hip"
+#include "mfma_gemm.hip"
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -331,7 +332,49 @@ torch::Tensor rope_apply(torch::Tensor x, torch::Tensor cos_table, torch::Tensor
   return out;
 }
 
+// ------------------------------------------------------- hand-written GEMM
+
+std::vector<torch::Tensor> mfma_linear_bf16(torch::Tensor x, torch::Tensor weight,
+                                            c10::optional<torch::Tensor> bias, bool gelu,
+                                            bool save_pre_act) {
+  CHECK_GPU(x); CHECK_CONTIG(x); CHECK_CONTIG(weight);
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 && weight.scalar_type() == torch::kBFloat16);
+  long long K = x.size(-1);
+  long long M = x.numel() / K;
+  long long N = weight.size(0);
+  TORCH_CHECK(weight.size(1) == K, "weight must be [N, K]");
+  TORCH_CHECK(M % 128 == 0 && N % 128 == 0 && K % 32 == 0,
+              "mfma_linear: M%128, N%128, K%32 must be 0 (wrapper pads M)");
+  auto out_sizes = x.sizes().vec();
+  out_sizes.back() = N;
+  auto out = torch::empty(out_sizes, x.options());
+  torch::Tensor pre_act;
+  unsigned short* pre_ptr = nullptr;
+  const unsigned short* bias_ptr = nullptr;
+  if (gelu) {
+    TORCH_CHECK(bias.has_value() && bias->is_contiguous() && bias->scalar_type() == torch::kBFloat16);
+    bias_ptr = (const unsigned short*)bias->data_ptr();
+    if (save_pre_act) {
+      pre_act = torch::empty(out_sizes, x.options());
+      pre_ptr = (unsigned short*)pre_act.data_ptr();
+    }
+  }
+  int grid = (int)((M / 128) * (N / 128));
+  if (gelu) {
+    hipLaunchKernelGGL(gemm_bt_bf16_t<1>, dim3(grid), dim3(256), 0, current_stream(),
+                       (const unsigned short*)x.data_ptr(), (const unsigned short*)weight.data_ptr(),
+                       bias_ptr, (unsigned short*)out.data_ptr(), pre_ptr, (int)M, (int)N, (int)K);
+  } else {
+    hipLaunchKernelGGL(gemm_bt_bf16_t<0>, dim3(grid), dim3(256), 0, current_stream(),
+                       (const unsigned short*)x.data_ptr(), (const unsigned short*)weight.data_ptr(),
+                       nullptr, (unsigned short*)out.data_ptr(), nullptr, (int)M, (int)N, (int)K);
+  }
+  if (save_pre_act && gelu) return {out, pre_act};
+  return {out};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mfma_linear_bf16", &mfma_linear_bf16, "hand-written MFMA GEMM: x @ W^T (+bias+gelu)");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");
   m.def("swiglu_fwd", &swiglu_fwd, "silu(gate) * up");

@@ -1,0 +1,105 @@
+// Hand-written CDNA4 MFMA GEMM: C[M,N] = A[M,K] @ W[N,K]^T (+ fused bias+GELU
+// epilogue). Structure follows the measured gfx950 ladder in
+// cdna_hip_programming.md §5 ("m97"): 128x128 output tile, 4 waves computing
+// 64x64 each as 4x4 fragments of v_mfma_f32_16x16x32_bf16, A/B staged through
+// LDS with __builtin_amdgcn_global_load_lds width 16 (the +67% lever, common
+// mistake #1), XCD-aware blockIdx swizzle for L2 locality (technique T1).
+//
+// W is the torch nn.Linear weight, stored [N, K] row-major -- exactly the
+// "B^T input" layout the ladder uses, so both A and W stage with coalesced
+// contiguous rows.
+//
+// Constraints (checked in the binding; the python wrapper pads M):
+//   M % 128 == 0, N % 128 == 0, K % 32 == 0.
+#pragma once
+
+typedef __attribute__((ext_vector_type(8))) short bf16x8_frag;   // MFMA A/B operand
+typedef __attribute__((ext_vector_type(4))) float f32x4_acc;     // MFMA C/D accumulator
+
+// epilogue selector: 0 = none (plain GEMM), 1 = bias + gelu_tanh
+template <int EPILOGUE>
+__global__ __launch_bounds__(256) void gemm_bt_bf16_t(
+    const ushort_t* __restrict__ A,   // [M, K] bf16 row-major
+    const ushort_t* __restrict__ W,   // [N, K] bf16 row-major (linear weight)
+    const ushort_t* __restrict__ bias,  // [N] bf16 (EPILOGUE==1) or nullptr
+    ushort_t* __restrict__ C,         // [M, N] bf16 row-major
+    ushort_t* __restrict__ pre_act,   // [M, N] saved x+bias for backward; may be null
+    int M, int N, int K) {
+  constexpr int BM = 128, BN = 128, BK = 32;
+  __shared__ ushort_t lds_a[BM * BK];  // [128][32] row-major, matches gload_lds linear writes
+  __shared__ ushort_t lds_b[BN * BK];
+
+  // XCD-aware swizzle (bijective form, guide ERRATA #11): consecutive
+  // workgroups share W panels; keep them on one XCD's L2.
+  int nwg = gridDim.x;
+  int wg = blockIdx.x;
+  {
+    constexpr int NXCD = 8;
+    int q = nwg / NXCD, r = nwg % NXCD;
+    int xcd = wg % NXCD, idx = wg / NXCD;
+    wg = (xcd < r) ? (xcd * (q + 1) + idx) : (r * (q + 1) + (xcd - r) * q + idx);
+  }
+  const int tiles_n = N / BN;
+  const int tile_m = wg / tiles_n, tile_n = wg % tiles_n;
+  const int m0 = tile_m * BM, n0 = tile_n * BN;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int wr = (wave >> 1) * 64, wc = (wave & 1) * 64;  // wave's 64x64 sub-tile
+
+  f32x4_acc acc[4][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
+
+  // staging addresses: thread t loads 8 bf16 (16 B) of row (t/4), segment (t%4)
+  const int row_of_t = tid >> 2, seg_of_t = tid & 3;
+  const long long a_src_base = (long long)(m0 + row_of_t) * K + seg_of_t * 8;
+  const long long b_src_base = (long long)(n0 + row_of_t) * K + seg_of_t * 8;
+
+  for (int k0 = 0; k0 < K; k0 += BK) {
+    __syncthreads();
+    // 128x32 tile = 256 threads x 8 bf16 each; LDS dest is linear in thread
+    // order (wave-uniform base + lane*16 -- the gload_lds contract, §5 caveat)
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(A + a_src_base + k0),
+                                     (__attribute__((address_space(3))) void*)(lds_a + tid * 8), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(W + b_src_base + k0),
+                                     (__attribute__((address_space(3))) void*)(lds_b + tid * 8), 16, 0, 0);
+    __syncthreads();
+
+    // fragment loads: lane l reads row (16-block + l&15), k = (l>>4)*8 .. +8
+    const int fr = lane & 15, fq = lane >> 4;
+#pragma unroll
+    for (int mi = 0; mi < 4; ++mi) {
+      bf16x8_frag a_frag = *reinterpret_cast<const bf16x8_frag*>(&lds_a[(wr + mi * 16 + fr) * BK + fq * 8]);
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        bf16x8_frag b_frag = *reinterpret_cast<const bf16x8_frag*>(&lds_b[(wc + ni * 16 + fr) * BK + fq * 8]);
+        acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc[mi][ni], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: C/D layout col = lane&15, row = (lane>>4)*4 + reg (guide §3)
+  const int fr = lane & 15, fq = lane >> 4;
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int reg = 0; reg < 4; ++reg) {
+        int row = m0 + wr + mi * 16 + fq * 4 + reg;
+        int col = n0 + wc + ni * 16 + fr;
+        float v = acc[mi][ni][reg];
+        long long out_idx = (long long)row * N + col;
+        if (EPILOGUE == 1) {
+          v += bf16_to_f32(bias[col]);
+          if (pre_act != nullptr) pre_act[out_idx] = f32_to_bf16(v);
+          v = gelu_tanh(v);
+        }
+        C[out_idx] = f32_to_bf16(v);
+      }
+    }
+  }
+}
