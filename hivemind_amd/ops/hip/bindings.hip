@@ -373,7 +373,17 @@ std::vector<torch::Tensor> mfma_linear_bf16(torch::Tensor x, torch::Tensor weigh
   return {out};
 }
 
+torch::Tensor mfma_probe(torch::Tensor a_vals, torch::Tensor b_vals) {
+  CHECK_GPU(a_vals); CHECK_CONTIG(a_vals); CHECK_CONTIG(b_vals);
+  auto c = torch::zeros({16, 16}, a_vals.options().dtype(torch::kFloat32));
+  hipLaunchKernelGGL(mfma_probe_16x16x32_bf16, dim3(1), dim3(64), 0, current_stream(),
+                     (const unsigned short*)a_vals.data_ptr(), (const unsigned short*)b_vals.data_ptr(),
+                     c.data_ptr<float>());
+  return c;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 operand layout probe");
   m.def("mfma_linear_bf16", &mfma_linear_bf16, "hand-written MFMA GEMM: x @ W^T (+bias+gelu)");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
   m.def("rmsnorm_bwd", &rmsnorm_bwd, "RMSNorm backward");

@@ -53,19 +53,26 @@ __global__ __launch_bounds__(256) void gemm_bt_bf16_t(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // staging addresses: thread t loads 8 bf16 (16 B) of row (t/4), segment (t%4)
+  // staging: thread t loads 16 B of row (t/4), segment (t%4); one issue covers
+  // 64 rows (256 threads x 8 bf16), so each 128x32 tile takes two issues.
+  // LDS dest is linear in thread order (wave-uniform base + lane*16 --
+  // the gload_lds contract, §5 caveat).
   const int row_of_t = tid >> 2, seg_of_t = tid & 3;
   const long long a_src_base = (long long)(m0 + row_of_t) * K + seg_of_t * 8;
   const long long b_src_base = (long long)(n0 + row_of_t) * K + seg_of_t * 8;
+  const long long half_rows_stride = 64LL * K;  // second issue: rows 64..127
+  constexpr int HALF_LDS = 64 * BK;             // 2048 bf16
 
   for (int k0 = 0; k0 < K; k0 += BK) {
     __syncthreads();
-    // 128x32 tile = 256 threads x 8 bf16 each; LDS dest is linear in thread
-    // order (wave-uniform base + lane*16 -- the gload_lds contract, §5 caveat)
     __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(A + a_src_base + k0),
                                      (__attribute__((address_space(3))) void*)(lds_a + tid * 8), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(A + a_src_base + half_rows_stride + k0),
+                                     (__attribute__((address_space(3))) void*)(lds_a + HALF_LDS + tid * 8), 16, 0, 0);
     __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(W + b_src_base + k0),
                                      (__attribute__((address_space(3))) void*)(lds_b + tid * 8), 16, 0, 0);
+    __builtin_amdgcn_global_load_lds((const __attribute__((address_space(1))) void*)(W + b_src_base + half_rows_stride + k0),
+                                     (__attribute__((address_space(3))) void*)(lds_b + HALF_LDS + tid * 8), 16, 0, 0);
     __syncthreads();
 
     // fragment loads: lane l reads row (16-block + l&15), k = (l>>4)*8 .. +8
@@ -101,5 +108,29 @@ __global__ __launch_bounds__(256) void gemm_bt_bf16_t(
         C[out_idx] = f32_to_bf16(v);
       }
     }
+  }
+}
+
+
+// Layout probe: lane l takes its 8 A/B operand values directly from
+// a_vals[l*8+j]; the resulting C (written with the verified C/D mapping)
+// reveals which (row, k) each (lane, j) slot corresponds to.
+extern "C" __global__ void mfma_probe_16x16x32_bf16(const ushort_t* __restrict__ a_vals,
+                                                    const ushort_t* __restrict__ b_vals,
+                                                    float* __restrict__ c_out) {
+  int lane = threadIdx.x & 63;
+  bf16x8_frag a_frag, b_frag;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    a_frag[j] = (short)a_vals[lane * 8 + j];
+    b_frag[j] = (short)b_vals[lane * 8 + j];
+  }
+  f32x4_acc acc = {0.f, 0.f, 0.f, 0.f};
+  acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a_frag, b_frag, acc, 0, 0, 0);
+#pragma unroll
+  for (int reg = 0; reg < 4; ++reg) {
+    int row = (lane >> 4) * 4 + reg;
+    int col = lane & 15;
+    c_out[row * 16 + col] = acc[reg];
   }
 }
