@@ -8,7 +8,10 @@
 
 #include "elementwise.hip"
 #include "llama_ops.hip"
-#include "mfma_gemm.hip"
+
+extern "C" void launch_gemm_bt_bf16(const void*, const void*, const void*, void*, void*, int, int, int, int, void*);
+extern "C" void launch_mfma_probe(const void*, const void*, void*, void*);
+extern "C" void launch_stage_probe(const void*, void*, int, void*);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -359,16 +362,9 @@ std::vector<torch::Tensor> mfma_linear_bf16(torch::Tensor x, torch::Tensor weigh
       pre_ptr = (unsigned short*)pre_act.data_ptr();
     }
   }
-  int grid = (int)((M / 128) * (N / 128));
-  if (gelu) {
-    hipLaunchKernelGGL(gemm_bt_bf16_t<1>, dim3(grid), dim3(256), 0, current_stream(),
-                       (const unsigned short*)x.data_ptr(), (const unsigned short*)weight.data_ptr(),
-                       bias_ptr, (unsigned short*)out.data_ptr(), pre_ptr, (int)M, (int)N, (int)K);
-  } else {
-    hipLaunchKernelGGL(gemm_bt_bf16_t<0>, dim3(grid), dim3(256), 0, current_stream(),
-                       (const unsigned short*)x.data_ptr(), (const unsigned short*)weight.data_ptr(),
-                       nullptr, (unsigned short*)out.data_ptr(), nullptr, (int)M, (int)N, (int)K);
-  }
+  launch_gemm_bt_bf16(x.data_ptr(), weight.data_ptr(), (const void*)bias_ptr,
+                      out.data_ptr(), (void*)pre_ptr, (int)M, (int)N, (int)K,
+                      gelu ? 1 : 0, (void*)current_stream());
   if (save_pre_act && gelu) return {out, pre_act};
   return {out};
 }
@@ -376,13 +372,20 @@ std::vector<torch::Tensor> mfma_linear_bf16(torch::Tensor x, torch::Tensor weigh
 torch::Tensor mfma_probe(torch::Tensor a_vals, torch::Tensor b_vals) {
   CHECK_GPU(a_vals); CHECK_CONTIG(a_vals); CHECK_CONTIG(b_vals);
   auto c = torch::zeros({16, 16}, a_vals.options().dtype(torch::kFloat32));
-  hipLaunchKernelGGL(mfma_probe_16x16x32_bf16, dim3(1), dim3(64), 0, current_stream(),
-                     (const unsigned short*)a_vals.data_ptr(), (const unsigned short*)b_vals.data_ptr(),
-                     c.data_ptr<float>());
+  launch_mfma_probe(a_vals.data_ptr(), b_vals.data_ptr(), c.data_ptr(), (void*)current_stream());
   return c;
 }
 
+torch::Tensor stage_probe(torch::Tensor A) {
+  CHECK_GPU(A); CHECK_CONTIG(A);
+  int K = (int)A.size(1);
+  auto out = torch::zeros({128 * 32}, A.options());
+  launch_stage_probe(A.data_ptr(), out.data_ptr(), K, (void*)current_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("stage_probe", &stage_probe, "gload_lds staging probe");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 operand layout probe");
   m.def("mfma_linear_bf16", &mfma_linear_bf16, "hand-written MFMA GEMM: x @ W^T (+bias+gelu)");
   m.def("rmsnorm_fwd", &rmsnorm_fwd, "RMSNorm forward (bf16)");
