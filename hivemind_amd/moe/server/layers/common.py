@@ -11,8 +11,14 @@ from __future__ import annotations
 import torch
 import torch.nn as nn
 
-from ....ops import fused_bias_gelu, fused_layernorm
+import os
+
+from ....ops import fused_bias_gelu, fused_layernorm, fused_linear_gelu
 from .custom_experts import register_expert_class
+
+# hand-written MFMA path for the expert up-projection (fused GEMM+bias+GELU);
+# set HIVEMIND_AMD_MFMA_EXPERT=0 to force the hipBLASLt + fused-epilogue path
+_USE_MFMA_EXPERT = os.environ.get("HIVEMIND_AMD_MFMA_EXPERT", "1") != "0"
 
 
 def sample_ffn_input(batch_size: int, hidden_dim: int) -> torch.Tensor:
@@ -31,8 +37,12 @@ class FeedforwardBlock(nn.Module):
         nn.init.normal_(self.ffn_up_weight, std=0.02)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        up = torch.nn.functional.linear(x, self.ffn_up_weight.to(x.dtype))
-        act = fused_bias_gelu(up, self.ffn_up_bias.to(x.dtype))
+        weight = self.ffn_up_weight.to(x.dtype)
+        bias = self.ffn_up_bias.to(x.dtype)
+        if _USE_MFMA_EXPERT and x.is_cuda and x.dtype == torch.bfloat16:
+            act = fused_linear_gelu(x, weight, bias)  # one MFMA kernel, fused epilogue
+        else:
+            act = fused_bias_gelu(torch.nn.functional.linear(x, weight), bias)
         down = self.ffn_down.to(x.dtype)(act) if self.ffn_down.weight.dtype != x.dtype else self.ffn_down(act)
         return fused_layernorm(down, self.layer_norm_weight, self.layer_norm_bias, residual=x)
 
