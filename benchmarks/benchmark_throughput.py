@@ -30,20 +30,29 @@ def _client_proc_main(proc_idx, dht_endpoint, uid_list, cfg, clients_per_proc, e
 
         client_dht = DHT(initial_peers=[dht_endpoint], start=True, client_mode=True)
         client_experts = get_experts(client_dht, uid_list)
+        assert all(e is not None for e in client_experts), "client could not resolve all experts"
         threads = []
+        thread_errors = []
 
         def one_client(client_idx):
-            torch.manual_seed(client_idx)
-            for b in range(cfg["batches_per_client"]):
-                expert = client_experts[(client_idx + b) % len(client_experts)]
-                x = torch.randn(cfg["batch_size"], cfg["hidden_dim"])
-                if cfg["backprop"]:
-                    x.requires_grad_(True)
-                    out = expert(x)
-                    out.sum().backward()
-                else:
-                    with torch.no_grad():
-                        expert(x)
+            try:
+                torch.manual_seed(client_idx)
+                for b in range(cfg["batches_per_client"]):
+                    expert = client_experts[(client_idx + b) % len(client_experts)]
+                    x = torch.randn(cfg["batch_size"], cfg["hidden_dim"])
+                    if cfg["backprop"]:
+                        x.requires_grad_(True)
+                        out = expert(x)
+                        out.sum().backward()
+                        assert x.grad is not None
+                    else:
+                        with torch.no_grad():
+                            out = expert(x)
+                    assert out.shape == x.shape
+            except BaseException:
+                import traceback
+
+                thread_errors.append(traceback.format_exc())
 
         for i in range(clients_per_proc):
             t = threading.Thread(target=one_client, args=(proc_idx * clients_per_proc + i,))
@@ -51,8 +60,10 @@ def _client_proc_main(proc_idx, dht_endpoint, uid_list, cfg, clients_per_proc, e
             threads.append(t)
         for t in threads:
             t.join()
+        if thread_errors:
+            error_queue.put(thread_errors[0])
         client_dht.shutdown()
-    except Exception:
+    except BaseException:
         import traceback
 
         error_queue.put(traceback.format_exc())
@@ -128,6 +139,13 @@ def main():
         p.join()
     elapsed = time.perf_counter() - t0
     assert error_queue.empty(), f"client errors: {error_queue.get()}"
+    # cross-check: the server must have actually processed every sample
+    processed = sum(
+        backend.forward_pool.total_processed + backend.backward_pool.total_processed
+        for backend in server.module_backends.values()
+    )
+    expected = total_samples * (2 if cfg["backprop"] else 1)  # backward implies a forward too
+    assert processed >= total_samples, f"server processed {processed} < {total_samples} samples"
 
     result = {
         "metric": "MoE server throughput (samples/s)",

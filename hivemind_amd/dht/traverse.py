@@ -94,6 +94,8 @@ async def traverse_dht(
         for uid in initial_nodes:
             visited[q].add(uid)
 
+    in_flight: Dict[DHTID, int] = {q: 0 for q in queries}  # RPCs that may still add candidates
+
     def upper_bound(q: DHTID) -> int:
         if len(nearest[q]) < beam_size:
             return DHTID.MAX + 1
@@ -102,6 +104,8 @@ async def traverse_dht(
     def maybe_finish(q: DHTID):
         if q in finished:
             return
+        if in_flight.get(q, 0) > 0:
+            return  # an outstanding request may still bring candidates or the value
         if not candidates[q] or candidates[q][ROOT][0] > upper_bound(q):
             finished.add(q)
             if found_callback is not None:
@@ -110,47 +114,67 @@ async def traverse_dht(
                 active_tasks.append(task)
 
     search_lock = asyncio.Lock()
+    progress_event = asyncio.Event()
 
     async def worker():
         while True:
             async with search_lock:
                 # pick the (query, candidate) pair with the smallest distance among active queries
                 best: Optional[Tuple[int, DHTID, DHTID]] = None  # (distance, query, peer)
+                idle_wait = False
                 for q in queries:
                     if q in finished:
                         continue
                     while candidates[q] and candidates[q][ROOT][0] > upper_bound(q):
                         heapq.heappop(candidates[q])
                     maybe_finish(q)
-                    if q in finished or not candidates[q]:
+                    if q in finished:
+                        continue
+                    if not candidates[q]:
+                        idle_wait = True  # in-flight elsewhere may refill this query
                         continue
                     d, uid = candidates[q][ROOT]
                     if best is None or d < best[0]:
                         best = (d, q, uid)
                 if best is None:
+                    if not idle_wait:
+                        return
+                    progress_event.clear()
+                else:
+                    _, main_query, peer = best
+                    heapq.heappop(candidates[main_query])
+                    # pack additional active queries that would also like this peer;
+                    # the peer counts as visited for every packed query so no other
+                    # worker re-queries it on their behalf
+                    packed = [main_query]
+                    for q in queries:
+                        if len(packed) >= queries_per_call:
+                            break
+                        if q is main_query or q in finished:
+                            continue
+                        packed.append(q)
+                    for q in packed:
+                        visited[q].add(peer)
+                        in_flight[q] += 1
+                        # lazily drop the peer from q's own candidate heap
+                        candidates[q] = [(d, uid) for d, uid in candidates[q] if uid != peer]
+                        heapq.heapify(candidates[q])
+            if best is None:
+                # no candidates right now but other workers are mid-RPC: wait for progress
+                try:
+                    await asyncio.wait_for(progress_event.wait(), timeout=1.0)
+                except asyncio.TimeoutError:
+                    pass
+                if all(q in finished for q in queries):
                     return
-                _, main_query, peer = best
-                heapq.heappop(candidates[main_query])
-                # pack additional active queries that would also like this peer;
-                # the peer counts as visited for every packed query so no other
-                # worker re-queries it on their behalf
-                packed = [main_query]
-                for q in queries:
-                    if len(packed) >= queries_per_call:
-                        break
-                    if q is main_query or q in finished:
-                        continue
-                    packed.append(q)
-                for q in packed:
-                    visited[q].add(peer)
-                    # lazily drop the peer from q's own candidate heap
-                    candidates[q] = [(d, uid) for d, uid in candidates[q] if uid != peer]
-                    heapq.heapify(candidates[q])
+                continue
             try:
                 responses = await get_neighbors(peer, packed)
             except Exception:
                 responses = {}
             async with search_lock:
+                for q in packed:
+                    in_flight[q] -= 1
                 for q, (neighbors, should_stop) in responses.items():
                     if q not in candidates:
                         continue
@@ -174,6 +198,7 @@ async def traverse_dht(
                                 heapq.heappushpop(nearest[q], (-distance, uid))
                 for q in packed:
                     maybe_finish(q)
+                progress_event.set()
 
     workers = [asyncio.create_task(worker()) for _ in range(max(1, num_workers))]
     try:

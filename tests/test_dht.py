@@ -145,3 +145,19 @@ def test_dht_expiration():
     time.sleep(0.7)
     assert dht.get("ephemeral") is None
     dht.shutdown()
+
+
+def test_key_stored_before_peer_joins_is_reachable():
+    """Regression: a key stored while the swarm had one node must be findable
+    by peers that join later (exposed a traverse in-flight accounting race)."""
+
+    async def main():
+        alice = await DHTNode.create()
+        assert await alice.store("early_key", MSGPackSerializer.dumps(7), get_dht_time() + 60)
+        bob = await DHTNode.create(initial_peers=[alice.p2p.endpoint])
+        result = await bob.get("early_key")
+        assert result is not None and MSGPackSerializer.loads(result.value) == 7
+        await bob.shutdown()
+        await alice.shutdown()
+
+    run(main())
