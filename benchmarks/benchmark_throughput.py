@@ -139,13 +139,14 @@ def main():
         p.join()
     elapsed = time.perf_counter() - t0
     assert error_queue.empty(), f"client errors: {error_queue.get()}"
-    # cross-check: the server must have actually processed every sample
-    processed = sum(
+    # cross-check: the server must have actually processed every client call
+    # (pools count tasks; each client batch is one forward task + one backward task)
+    processed_tasks = sum(
         backend.forward_pool.total_processed + backend.backward_pool.total_processed
         for backend in server.module_backends.values()
     )
-    expected = total_samples * (2 if cfg["backprop"] else 1)  # backward implies a forward too
-    assert processed >= total_samples, f"server processed {processed} < {total_samples} samples"
+    expected_tasks = cfg["num_clients"] * cfg["batches_per_client"] * (2 if cfg["backprop"] else 1)
+    assert processed_tasks >= expected_tasks, f"server processed {processed_tasks} of {expected_tasks} tasks"
 
     result = {
         "metric": "MoE server throughput (samples/s)",
