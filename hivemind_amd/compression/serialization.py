@@ -19,7 +19,7 @@ from .base import CompressionBase, CompressionInfo, CompressionType, NoCompressi
 from .floating import Float16Compression, ScaledFloat16Compression
 from .quantization import BlockwiseQuantization, Quantile8BitQuantization, Uniform8BitQuantization
 
-STREAMING_CHUNK_SIZE_BYTES = 2**18
+STREAMING_CHUNK_SIZE_BYTES = 2**22
 
 BASE_COMPRESSION_TYPES = {
     CompressionType.NONE: NoCompression(),

@@ -26,7 +26,11 @@ from .module_backend import ModuleBackend
 
 logger = get_logger(__name__)
 
-MAX_UNARY_PAYLOAD_SIZE = 2 * 1024 * 1024  # same cutoff as reference (control.py:36-39)
+# The reference caps unary payloads at 2 MB because its p2pd control channel
+# requires it (control.py:36-39); this transport has no such constraint, and on
+# a node-local/xGMI control network larger unary messages skip per-chunk
+# envelope overhead entirely.
+MAX_UNARY_PAYLOAD_SIZE = 32 * 1024 * 1024
 
 
 @dataclass

@@ -92,18 +92,24 @@ class _Connection:
         self.closed = asyncio.Event()
 
     async def send_frame(self, ftype: int, call_id: bytes, handler: str = "", payload: bytes = b""):
-        env = MSGPackSerializer.dumps((ftype, call_id, handler, payload))
+        # header and payload travel as separate buffers: a multi-MB tensor part
+        # is never re-wrapped through msgpack (zero extra copies on the hot path)
+        header = MSGPackSerializer.dumps((ftype, call_id, handler))
         async with self.send_lock:
-            self.writer.write(struct.pack(">I", len(env)) + env)
+            self.writer.write(struct.pack(">II", len(header), len(payload)))
+            self.writer.write(header)
+            if payload:
+                self.writer.write(payload)
             await self.writer.drain()
 
     async def recv_frame(self) -> Tuple[int, bytes, str, bytes]:
-        header = await self.reader.readexactly(4)
-        (length,) = struct.unpack(">I", header)
-        if length > MAX_FRAME_SIZE:
-            raise P2PDaemonError(f"frame too large: {length}")
-        env = await self.reader.readexactly(length)
-        ftype, call_id, handler, payload = MSGPackSerializer.loads(env)
+        lengths = await self.reader.readexactly(8)
+        hlen, plen = struct.unpack(">II", lengths)
+        if hlen + plen > MAX_FRAME_SIZE:
+            raise P2PDaemonError(f"frame too large: {hlen + plen}")
+        header = await self.reader.readexactly(hlen)
+        payload = await self.reader.readexactly(plen) if plen else b""
+        ftype, call_id, handler = MSGPackSerializer.loads(header)
         return ftype, call_id, handler, payload
 
     def fail_all(self, exc: Exception):
