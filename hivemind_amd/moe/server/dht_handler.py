@@ -26,13 +26,22 @@ logger = get_logger(__name__)
 
 
 class DHTHandlerThread(threading.Thread):
-    def __init__(self, module_backends, dht: DHT, update_period: float = 30.0, expiration: Optional[float] = None, endpoint: str = ""):
+    def __init__(
+        self,
+        module_backends,
+        dht: DHT,
+        update_period: float = 30.0,
+        expiration: Optional[float] = None,
+        endpoint: str = "",
+        uid_to_peer: Optional[Dict[ExpertUID, Tuple[str, str]]] = None,
+    ):
         super().__init__(name="moe-dht-handler", daemon=True)
         if expiration is None:
             expiration = max(2 * update_period, 60.0)
         self.module_backends = module_backends
         self.dht, self.update_period, self.expiration = dht, update_period, expiration
         self.endpoint = endpoint
+        self.uid_to_peer = uid_to_peer  # uid -> (peer_id_b58, endpoint): balanced handlers
         self.stop_event = threading.Event()
 
     def run(self) -> None:
@@ -43,6 +52,7 @@ class DHTHandlerThread(threading.Thread):
                     list(self.module_backends.keys()),
                     expiration_time=get_dht_time() + self.expiration,
                     endpoint=self.endpoint,
+                    uid_to_peer=self.uid_to_peer,
                 )
             except Exception as e:
                 logger.warning(f"expert declaration failed: {e!r}")
@@ -53,21 +63,32 @@ class DHTHandlerThread(threading.Thread):
         self.stop_event.set()
 
 
-def declare_experts(dht: DHT, uids: Sequence[ExpertUID], expiration_time: DHTExpiration, endpoint: str = "") -> Dict:
-    """Declare experts + grid prefixes to the DHT (reference dht_handler.py:41-79)."""
+def declare_experts(
+    dht: DHT,
+    uids: Sequence[ExpertUID],
+    expiration_time: DHTExpiration,
+    endpoint: str = "",
+    uid_to_peer: Optional[Dict[ExpertUID, Tuple[str, str]]] = None,
+) -> Dict:
+    """Declare experts + grid prefixes to the DHT (reference dht_handler.py:41-79).
+
+    ``uid_to_peer`` maps uids to (peer_id_b58, endpoint) when experts are served
+    by several balanced connection-handler listeners (reference's ``balanced``
+    p2pd flag, expressed as expert sharding across handler loops).
+    """
     for uid in uids:
         assert is_valid_uid(uid), f"invalid expert uid: {uid}"
-    peer_b58 = dht.peer_id.to_base58()
-    endpoint = endpoint or dht.endpoint
+    default_peer = (dht.peer_id.to_base58(), endpoint or dht.endpoint)
 
     async def _declare(dht_obj, node):
         from ...utils.serializer import MSGPackSerializer
 
         keys, subkeys, values = [], [], []
         for uid in uids:
+            peer_b58, ep = (uid_to_peer or {}).get(uid, default_peer)
             keys.append(uid)
             subkeys.append(None)
-            values.append(MSGPackSerializer.dumps([peer_b58, endpoint]))
+            values.append(MSGPackSerializer.dumps([peer_b58, ep]))
             # all proper prefixes: "ffn.3.5" -> ("ffn.3", 5), ("ffn", 3)
             prefix = uid
             while UID_DELIMITER in prefix:
@@ -75,7 +96,7 @@ def declare_experts(dht: DHT, uids: Sequence[ExpertUID], expiration_time: DHTExp
                 parent = parent.rstrip(UID_DELIMITER)
                 keys.append(parent)
                 subkeys.append(coord)
-                values.append(MSGPackSerializer.dumps([uid, peer_b58, endpoint]))
+                values.append(MSGPackSerializer.dumps([uid, peer_b58, ep]))
                 prefix = parent
         return await node.store_many(keys, values, expiration_time, subkeys=subkeys)
 

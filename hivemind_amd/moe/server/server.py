@@ -41,6 +41,7 @@ class Server(threading.Thread):
         module_backends: Dict[str, ModuleBackend],
         *,
         device: Optional[torch.device] = None,
+        num_connection_handlers: int = 4,
         update_period: float = 30.0,
         expiration: Optional[float] = None,
         checkpoint_dir: Optional[Path] = None,
@@ -50,21 +51,53 @@ class Server(threading.Thread):
         super().__init__(name="moe-server", daemon=True)
         self.dht, self.module_backends = dht, module_backends
         self.device = device
+        self.num_connection_handlers = max(1, num_connection_handlers)
         self.conn_handler = ConnectionHandler(module_backends)
         self.runtime = Runtime(module_backends, device=device, stats_report_interval=stats_report_interval)
-        self.dht_handler_thread = DHTHandlerThread(module_backends, dht, update_period, expiration)
+        self._handler_loops: list = []
+        self._handler_p2ps: list = []
         self.checkpoint_saver = (
             CheckpointSaver(module_backends, checkpoint_dir, update_period) if checkpoint_dir is not None else None
         )
+        self._update_period, self._expiration = update_period, expiration
+        self.dht_handler_thread: Optional[DHTHandlerThread] = None
         self.ready = threading.Event()
         self._stop_requested = threading.Event()
         if start:
             self.run_in_background(await_ready=True)
 
     def run(self):
-        """Start all components and serve until shutdown (reference server.py:231-263)."""
-        p2p = self.dht.replicate_p2p()
-        asyncio.run_coroutine_threadsafe(self.conn_handler.add_handlers(p2p), self.dht.loop).result(15)
+        """Start all components and serve until shutdown (reference server.py:231-263).
+
+        Experts are sharded across ``num_connection_handlers`` listener loops,
+        each its own thread + P2P endpoint: tensor (de)serialization for
+        different experts runs in parallel (the reference achieves the same
+        with N forked ConnectionHandler processes + the p2pd ``balanced`` flag).
+        """
+        from ...p2p import P2P
+        from ...utils.asyncio_utils import EventLoopThread
+
+        # handler 0 lives on the DHT loop (keeps single-loop deployments simple)
+        p2p0 = self.dht.replicate_p2p()
+        asyncio.run_coroutine_threadsafe(self.conn_handler.add_handlers(p2p0), self.dht.loop).result(15)
+        self._handler_p2ps.append(p2p0)
+        handler_loops = [self.dht.loop]
+        for i in range(1, self.num_connection_handlers):
+            loop_thread = EventLoopThread(name=f"moe-handler-{i}")
+            loop_thread.start_and_wait()
+            p2p_i = asyncio.run_coroutine_threadsafe(P2P.create(), loop_thread.loop).result(15)
+            asyncio.run_coroutine_threadsafe(self.conn_handler.add_handlers(p2p_i), loop_thread.loop).result(15)
+            self._handler_loops.append(loop_thread)
+            self._handler_p2ps.append(p2p_i)
+            handler_loops.append(loop_thread.loop)
+
+        uid_to_peer = {}
+        for idx, uid in enumerate(sorted(self.module_backends.keys())):
+            p2p_i = self._handler_p2ps[idx % len(self._handler_p2ps)]
+            uid_to_peer[uid] = (p2p_i.peer_id.to_base58(), p2p_i.endpoint)
+        self.dht_handler_thread = DHTHandlerThread(
+            self.module_backends, self.dht, self._update_period, self._expiration, uid_to_peer=uid_to_peer
+        )
         self.dht_handler_thread.start()
         if self.checkpoint_saver is not None:
             self.checkpoint_saver.start()
@@ -85,7 +118,13 @@ class Server(threading.Thread):
             self.conn_handler.remove_p2p_handlers(self.dht.replicate_p2p())
         except Exception:
             pass
-        self.dht_handler_thread.shutdown()
+        for loop_thread in self._handler_loops:
+            try:
+                loop_thread.shutdown()
+            except Exception:
+                pass
+        if self.dht_handler_thread is not None:
+            self.dht_handler_thread.shutdown()
         if self.checkpoint_saver is not None:
             self.checkpoint_saver.shutdown()
         self.runtime.shutdown()
@@ -108,6 +147,7 @@ class Server(threading.Thread):
         clip_grad_norm: Optional[float] = None,
         min_batch_size: int = 1,
         max_batch_size: int = 16384,
+        num_connection_handlers: int = 4,
         device: Optional[str] = None,
         checkpoint_dir: Optional[Path] = None,
         load_experts_from_dir: bool = False,
@@ -162,6 +202,7 @@ class Server(threading.Thread):
             dht,
             module_backends,
             device=torch.device(device),
+            num_connection_handlers=num_connection_handlers,
             update_period=update_period,
             expiration=expiration,
             checkpoint_dir=checkpoint_dir,

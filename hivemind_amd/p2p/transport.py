@@ -47,6 +47,9 @@ _T_CANCEL = 8
 _T_PING = 9
 
 MAX_FRAME_SIZE = 256 * 1024 * 1024  # control plane sanity bound
+STREAM_BUFFER_LIMIT = 64 * 1024 * 1024  # asyncio StreamReader high-water mark:
+# the default 64 KiB throttles multi-MB tensor frames with hundreds of
+# flow-control round-trips per read
 STREAM_CHUNK_SIZE = 512 * 1024  # aligns with averaging part size
 
 HandlerType = Union[
@@ -171,7 +174,7 @@ class P2P:
         self._listen_host = listen_host
         self._listen = listen
         if listen:
-            self._server = await asyncio.start_server(self._on_accept, listen_host, port)
+            self._server = await asyncio.start_server(self._on_accept, listen_host, port, limit=STREAM_BUFFER_LIMIT)
             self._port = self._server.sockets[0].getsockname()[1]
         return self
 
@@ -238,7 +241,7 @@ class P2P:
     async def connect_endpoint(self, endpoint: str) -> PeerInfo:
         """Dial a bare endpoint and learn the peer's identity (bootstrap helper)."""
         host, port = split_endpoint(endpoint)
-        reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port), timeout=10)
+        reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
         conn = _Connection(self, reader, writer)
         await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
         ftype, _, _, payload = await conn.recv_frame()
@@ -279,7 +282,7 @@ class P2P:
             for ep in endpoints:
                 try:
                     host, port = split_endpoint(ep)
-                    reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port), timeout=10)
+                    reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
                     conn = _Connection(self, reader, writer)
                     await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
                     ftype, _, _, payload = await conn.recv_frame()
