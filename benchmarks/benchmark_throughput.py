@@ -107,7 +107,7 @@ def main():
     server = Server.create(
         dht=dht, expert_uids=uids, expert_cls="ffn", hidden_dim=cfg["hidden_dim"],
         optim_cls=(torch.optim.Adam if cfg["backprop"] else None),
-        max_batch_size=cfg["max_batch_size"], num_connection_handlers=8, device=device, start=True,
+        max_batch_size=cfg["max_batch_size"], device=device, start=True,
     )
     t_server_ready = time.perf_counter()
 
@@ -118,7 +118,7 @@ def main():
 
     # clients run in separate processes (their serialization must not share the
     # server's GIL) -- reference benchmark_throughput.py uses client processes too
-    num_client_procs = max(1, min(16, cfg["num_clients"] // 2)) if cfg["num_clients"] >= 4 else 1
+    num_client_procs = max(1, min(8, cfg["num_clients"] // 4)) if cfg["num_clients"] >= 4 else 1
     clients_per_proc = cfg["num_clients"] // num_client_procs
 
     import torch.multiprocessing as mp
