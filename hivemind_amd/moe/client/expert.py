@@ -109,20 +109,36 @@ def _tensors_to_parts(uid: str, tensors: Sequence[torch.Tensor]) -> Tuple[List[W
     return parts, total
 
 
+RAW_PATH_MAX_BYTES = 128 * 1024 * 1024  # single-frame fast path bound
+
+
 async def expert_forward(uid: str, stub, tensors: Sequence[torch.Tensor], timeout: Optional[float] = None) -> List[torch.Tensor]:
-    """Unary below 2 MB, streamed above (reference expert.py:149-191)."""
-    parts, total = _tensors_to_parts(uid, tensors)
-    if total <= MAX_UNARY_PAYLOAD_SIZE:
-        raw = await asyncio.wait_for(stub.rpc_forward(ExpertRequest(uid=uid, tensors=parts)), timeout)
-        return [deserialize_torch_tensor(t) for t in ExpertResponse.loads(raw).tensors]
+    """Raw single-frame fast path (moe/wire.py) below 128 MB, chunk-streamed
+    above (reference's unary/stream split at expert.py:149-191)."""
+    from ..wire import pack_tensors, unpack_tensors
+
+    total = sum(t.numel() * t.element_size() for t in tensors)
+    if total <= RAW_PATH_MAX_BYTES:
+        payload = pack_tensors(uid, tensors)
+        raw = await asyncio.wait_for(
+            stub._p2p.call_unary(stub._peer, "expert::fwd_raw", payload, timeout=timeout), timeout
+        )
+        return unpack_tensors(raw)[1]
+    parts, _ = _tensors_to_parts(uid, tensors)
     return await _expert_stream_call(stub.rpc_forward_stream, uid, parts, timeout)
 
 
 async def expert_backward(uid: str, stub, tensors: Sequence[torch.Tensor], timeout: Optional[float] = None) -> List[torch.Tensor]:
-    parts, total = _tensors_to_parts(uid, tensors)
-    if total <= MAX_UNARY_PAYLOAD_SIZE:
-        raw = await asyncio.wait_for(stub.rpc_backward(ExpertRequest(uid=uid, tensors=parts)), timeout)
-        return [deserialize_torch_tensor(t) for t in ExpertResponse.loads(raw).tensors]
+    from ..wire import pack_tensors, unpack_tensors
+
+    total = sum(t.numel() * t.element_size() for t in tensors)
+    if total <= RAW_PATH_MAX_BYTES:
+        payload = pack_tensors(uid, tensors)
+        raw = await asyncio.wait_for(
+            stub._p2p.call_unary(stub._peer, "expert::bwd_raw", payload, timeout=timeout), timeout
+        )
+        return unpack_tensors(raw)[1]
+    parts, _ = _tensors_to_parts(uid, tensors)
     return await _expert_stream_call(stub.rpc_backward_stream, uid, parts, timeout)
 
 

@@ -60,6 +60,24 @@ class ConnectionHandler(ServicerBase):
 
     async def add_handlers(self, p2p: P2P):
         await self.add_p2p_handlers(p2p)
+        # fast path: tensors ride the raw frame payload (moe/wire.py), one copy
+        # per direction instead of the generic msgpack-wrapped route
+        p2p.add_unary_handler("expert::fwd_raw", self._raw_forward)
+        p2p.add_unary_handler("expert::bwd_raw", self._raw_backward)
+
+    async def _raw_forward(self, payload: bytes, context: RpcContext) -> bytes:
+        from ..wire import pack_tensors, unpack_tensors
+
+        uid, inputs = await self._in_executor(unpack_tensors, payload)
+        outputs = await self._process(uid, inputs, backward=False)
+        return await self._in_executor(pack_tensors, uid, outputs)
+
+    async def _raw_backward(self, payload: bytes, context: RpcContext) -> bytes:
+        from ..wire import pack_tensors, unpack_tensors
+
+        uid, inputs = await self._in_executor(unpack_tensors, payload)
+        outputs = await self._process(uid, inputs, backward=True)
+        return await self._in_executor(pack_tensors, uid, outputs)
 
     def _backend(self, uid: str) -> ModuleBackend:
         if uid not in self.module_backends:
