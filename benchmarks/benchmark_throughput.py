@@ -107,7 +107,7 @@ def main():
     server = Server.create(
         dht=dht, expert_uids=uids, expert_cls="ffn", hidden_dim=cfg["hidden_dim"],
         optim_cls=(torch.optim.Adam if cfg["backprop"] else None),
-        max_batch_size=cfg["max_batch_size"], num_connection_handlers=6, device=device, start=True,
+        max_batch_size=cfg["max_batch_size"], device=device, start=True,
     )
     t_server_ready = time.perf_counter()
 
