@@ -314,13 +314,27 @@ class DHTNode:
             tasks = []
             for nid, peer in nearest.items():
                 if nid == self.node_id:
-                    # store locally
+                    # store locally: sign + validate exactly as a remote store would
+                    # (reference node.py stores through the same record validators)
+                    from .validation import DHTRecord
+
                     for c in composites:
                         _, subkey = c
+                        value = binary_values[c]
+                        if isinstance(value, bytes) and self.protocol.record_validator is not None:
+                            packed_subkey = (
+                                MSGPackSerializer.dumps(subkey) if subkey is not None else b""
+                            )
+                            record = DHTRecord(key_id.to_bytes(), packed_subkey, value, expirations[c])
+                            signed = self.protocol.record_validator.sign_value(record)
+                            record = DHTRecord(key_id.to_bytes(), packed_subkey, signed, expirations[c])
+                            if not self.protocol.record_validator.validate(record):
+                                continue  # forged update of a protected key
+                            value = self.protocol.record_validator.strip_value(record)
                         if subkey is None:
-                            ok = self.protocol.storage.store(key_id, binary_values[c], expirations[c])
+                            ok = self.protocol.storage.store(key_id, value, expirations[c])
                         else:
-                            ok = self.protocol.storage.store_subkey(key_id, subkey, binary_values[c], expirations[c])
+                            ok = self.protocol.storage.store_subkey(key_id, subkey, value, expirations[c])
                         store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or ok
                     continue
                 ks = [key_id] * len(composites)

@@ -1,0 +1,216 @@
+"""Coverage for reference behaviors: client/aux averaging modes, local-updates
+optimizer, GradScaler, DHT record validators, expert checkpoints, custom
+expert classes (reference tests: test_averaging client/aux cases,
+test_dht_crypto/validation, test_custom_experts, test_start_server)."""
+
+import os
+import tempfile
+import threading
+import time
+from pathlib import Path
+
+import pytest
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from hivemind_amd import DHT
+from hivemind_amd.averaging import DecentralizedAverager
+
+
+def make_dht_swarm(n):
+    root = DHT(start=True)
+    return [root] + [DHT(initial_peers=[root.endpoint], start=True) for _ in range(n - 1)]
+
+
+def test_client_mode_averager():
+    """A client-mode peer participates in averaging but owns no vector fraction
+    (reference test_averaging.py:58-125 client/aux mixes)."""
+    dhts = make_dht_swarm(3)
+    tensors = [[torch.full((64,), float(i + 1))] for i in range(3)]
+    expected = sum(t[0] for t in tensors) / 3
+    averagers = []
+    for i in range(3):
+        averagers.append(
+            DecentralizedAverager(
+                [t.clone() for t in tensors[i]], dhts[i], start=True, prefix="climode",
+                target_group_size=3, min_group_size=3, min_matchmaking_time=1.0,
+                request_timeout=0.5, client_mode=(i == 2),
+            )
+        )
+    futures = [avg.step(wait=False, timeout=60) for avg in averagers]
+    results = [f.result(90) for f in futures]
+    assert all(r is not None for r in results)
+    for avg in averagers:
+        with avg.get_tensors() as ts:
+            assert torch.allclose(ts[0], expected, atol=1e-4), ts[0][:4]
+    for avg in averagers:
+        avg.shutdown()
+    for d in dhts:
+        d.shutdown()
+
+
+def test_aux_averager_contributes_zero_weight():
+    """Aux peers assist averaging without contributing data (weight 0)."""
+    dhts = make_dht_swarm(3)
+    values = [2.0, 4.0, 999.0]  # aux peer's value must not affect the average
+    averagers = []
+    for i in range(3):
+        averagers.append(
+            DecentralizedAverager(
+                [torch.full((32,), values[i])], dhts[i], start=True, prefix="auxmode",
+                target_group_size=3, min_group_size=3, min_matchmaking_time=1.0,
+                request_timeout=0.5, auxiliary=(i == 2),
+            )
+        )
+    futures = [avg.step(wait=False, timeout=60) for avg in averagers]
+    results = [f.result(90) for f in futures]
+    assert all(r is not None for r in results)
+    for avg in averagers[:2]:
+        with avg.get_tensors() as ts:
+            assert torch.allclose(ts[0], torch.full((32,), 3.0), atol=1e-4), ts[0][:4]
+    for avg in averagers:
+        avg.shutdown()
+    for d in dhts:
+        d.shutdown()
+
+
+def test_optimizer_local_updates_mode():
+    """use_local_updates: inner optimizer steps every batch, periodic param
+    averaging only (reference optimizer.py:143-145)."""
+    from hivemind_amd import Optimizer
+
+    dhts = make_dht_swarm(2)
+    results = [None, None]
+
+    def run_peer(idx):
+        torch.manual_seed(idx)
+        model = nn.Linear(8, 2)
+        opt = Optimizer(
+            dht=dhts[idx], run_id="localupd", target_batch_size=64, batch_size_per_step=16,
+            optimizer=lambda pg: torch.optim.SGD(pg, lr=0.1),
+            params=[{"params": list(model.parameters())}],
+            use_local_updates=True, matchmaking_time=1.0, averaging_timeout=30.0,
+            averager_opts=dict(request_timeout=0.5, min_group_size=2),
+            tracker_opts=dict(min_refresh_period=0.2, default_refresh_period=0.5),
+        )
+        X = torch.randn(64, 8)
+        w_before = model.weight.detach().clone()
+        for step in range(30):
+            loss = model(X).pow(2).mean()
+            loss.backward()
+            opt.step()
+            opt.zero_grad()
+            if opt.local_epoch >= 2:
+                break
+        results[idx] = (opt.local_epoch, not torch.allclose(model.weight.detach(), w_before))
+        opt.shutdown()
+
+    threads = [threading.Thread(target=run_peer, args=(i,)) for i in range(2)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(120)
+    assert all(r is not None and r[0] >= 2 and r[1] for r in results), results
+    for d in dhts:
+        d.shutdown()
+
+
+def test_grad_scaler_accumulation_semantics():
+    """GradScaler only unscales/steps inside the global step (reference grad_scaler.py)."""
+    from hivemind_amd.optim.grad_scaler import GradScaler
+
+    scaler = GradScaler(enabled=True)
+    p = torch.nn.Parameter(torch.ones(4))
+    opt = torch.optim.SGD([p], lr=0.1)
+    loss = (p * 2).sum()
+    scaler.scale(loss).backward()
+    # outside the global step: no unscale, no inner step
+    assert scaler.unscale_(opt) is False
+    assert scaler.step(opt) is False
+    assert scaler.update() is False
+    # inside a global step: everything runs once
+    with scaler.running_global_step():
+        assert scaler.unscale_(opt) is True
+        assert scaler.step(opt) is True
+    assert scaler.update() is True
+
+
+def test_dht_signature_validator_blocks_forgery():
+    """Keys with an [owner:...] marker may only be updated by the owner
+    (reference test_dht_crypto.py)."""
+    from hivemind_amd.dht.crypto import SignatureValidator
+    from hivemind_amd.utils.crypto import PrivateKey
+    from hivemind_amd.utils.timed_storage import get_dht_time
+
+    owner_validator = SignatureValidator(PrivateKey())
+    mallory_validator = SignatureValidator(PrivateKey())
+    dht_owner = DHT(start=True, record_validators=[owner_validator])
+    dht_mallory = DHT(initial_peers=[dht_owner.endpoint], start=True, record_validators=[mallory_validator])
+    time.sleep(0.2)
+
+    protected_key = b"epoch" + owner_validator.local_public_key
+    now = get_dht_time()
+    assert dht_owner.store(protected_key, 42, now + 30)
+    time.sleep(0.3)
+    result = dht_mallory.get(protected_key, latest=True)
+    assert result is not None and result.value == 42
+    # mallory cannot overwrite the owner's protected key with a forged value
+    assert not dht_mallory.store(protected_key, 666, now + 60)
+    result = dht_owner.get(protected_key, latest=True)
+    assert result is not None and result.value == 42
+    dht_mallory.shutdown()
+    dht_owner.shutdown()
+
+
+def test_expert_checkpoints_roundtrip():
+    """CheckpointSaver writes checkpoint_last.pt per expert; load_experts restores
+    (reference moe/server/checkpoints.py:36-75)."""
+    from hivemind_amd.moe.server.checkpoints import load_experts, store_experts
+    from hivemind_amd.moe.server.layers import name_to_block
+    from hivemind_amd.moe.server.module_backend import ModuleBackend
+    from hivemind_amd.utils.tensor_descr import BatchTensorDescriptor
+
+    with tempfile.TemporaryDirectory() as tmpdir:
+        tmpdir = Path(tmpdir)
+        expert = name_to_block["ffn"](16)
+        backend = ModuleBackend(
+            name="ckpt_test.0", module=expert, optimizer=torch.optim.SGD(expert.parameters(), lr=0.1),
+            args_schema=(BatchTensorDescriptor(16),),
+        )
+        backends = {"ckpt_test.0": backend}
+        store_experts(backends, tmpdir)
+        assert (tmpdir / "ckpt_test.0" / "checkpoint_last.pt").exists()
+        original = expert.ffn_up_weight.detach().clone()
+        with torch.no_grad():
+            expert.ffn_up_weight.mul_(0.0)
+        load_experts(backends, tmpdir)
+        assert torch.allclose(expert.ffn_up_weight.detach(), original)
+
+
+def test_register_custom_expert_class():
+    """register_expert_class + hosting a custom expert (reference custom_experts.py:17)."""
+    from hivemind_amd.moe import Server, get_experts, register_expert_class
+
+    @register_expert_class("double_mlp_test", lambda batch, hid: torch.empty((batch, hid)))
+    class DoubleMLP(nn.Module):
+        def __init__(self, hid_dim):
+            super().__init__()
+            self.lin = nn.Linear(hid_dim, hid_dim)
+
+        def forward(self, x):
+            return 2 * self.lin(x)
+
+    dht = DHT(start=True)
+    server = Server.create(
+        dht=dht, expert_uids=["dbl.0"], expert_cls="double_mlp_test", hidden_dim=8,
+        optim_cls=None, device="cpu", start=True,
+    )
+    try:
+        (expert,) = get_experts(dht, ["dbl.0"])
+        assert expert is not None
+        out = expert(torch.randn(3, 8))
+        assert out.shape == (3, 8)
+    finally:
+        server.shutdown()
+        dht.shutdown()
