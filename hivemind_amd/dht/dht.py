@@ -151,7 +151,7 @@ class DHT:
 
     async def _get(self, key: DHTKey, latest: bool, **kwargs) -> Optional[ValueWithExpiration]:
         result = await self._node.get(key, latest=latest, **kwargs)
-        return _deserialize_result(result)
+        return _deserialize_result(result, self._record_validator)
 
     def store(
         self,
@@ -204,18 +204,34 @@ class DHT:
         self._record_validator.extend(record_validators)
 
 
-def _deserialize_result(result: Optional[ValueWithExpiration]) -> Optional[ValueWithExpiration]:
+def _deserialize_result(
+    result: Optional[ValueWithExpiration], validator: Optional[RecordValidatorBase] = None
+) -> Optional[ValueWithExpiration]:
+    """Record validators sign stored values in place; strip signatures before
+    deserializing (values served from local storage arrive signed)."""
+    from .validation import DHTRecord
+
+    def _strip(raw: bytes, expiration: float) -> bytes:
+        if validator is None or not isinstance(raw, bytes):
+            return raw
+        try:
+            return validator.strip_value(DHTRecord(b"", b"", raw, expiration))
+        except Exception:
+            return raw
+
     if result is None:
         return None
     value, expiration = result
     if isinstance(value, DictionaryDHTValue):
         out: Dict[Any, ValueWithExpiration] = {}
         for subkey, (sub_value, sub_expiration) in value.items():
+            sub_value = _strip(sub_value, sub_expiration)
             try:
                 out[subkey] = ValueWithExpiration(MSGPackSerializer.loads(sub_value), sub_expiration)
             except Exception:
                 out[subkey] = ValueWithExpiration(sub_value, sub_expiration)
         return ValueWithExpiration(out, expiration)
+    value = _strip(value, expiration)
     try:
         return ValueWithExpiration(MSGPackSerializer.loads(value), expiration)
     except Exception:

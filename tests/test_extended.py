@@ -149,16 +149,18 @@ def test_dht_signature_validator_blocks_forgery():
     dht_mallory = DHT(initial_peers=[dht_owner.endpoint], start=True, record_validators=[mallory_validator])
     time.sleep(0.2)
 
-    protected_key = b"epoch" + owner_validator.local_public_key
+    # ownership markers protect SUBKEYS (DHT keys travel hashed; this is how the
+    # progress tracker publishes signed per-peer records)
+    protected_subkey = b"peer_state" + owner_validator.local_public_key
     now = get_dht_time()
-    assert dht_owner.store(protected_key, 42, now + 30)
+    assert dht_owner.store("run_epoch", 42, now + 30, subkey=protected_subkey)
     time.sleep(0.3)
-    result = dht_mallory.get(protected_key, latest=True)
-    assert result is not None and result.value == 42
-    # mallory cannot overwrite the owner's protected key with a forged value
-    assert not dht_mallory.store(protected_key, 666, now + 60)
-    result = dht_owner.get(protected_key, latest=True)
-    assert result is not None and result.value == 42
+    result = dht_mallory.get("run_epoch", latest=True)
+    assert result is not None and result.value[protected_subkey].value == 42
+    # mallory cannot overwrite the owner's protected subkey with a forged value
+    assert not dht_mallory.store("run_epoch", 666, now + 60, subkey=protected_subkey)
+    result = dht_owner.get("run_epoch", latest=True)
+    assert result is not None and result.value[protected_subkey].value == 42
     dht_mallory.shutdown()
     dht_owner.shutdown()
 
