@@ -352,6 +352,59 @@ def dequantize_blockwise(q: torch.Tensor, absmax: torch.Tensor) -> torch.Tensor:
 # ---------------------------------------------------------------------------
 
 
+class Lamb(torch.optim.Optimizer):
+    """LAMB (You et al.): Adam statistics + layer-wise trust-ratio scaling --
+    the optimizer of the reference's collaborative ALBERT recipe
+    (examples/albert/run_trainer.py LAMB + clipping). Runs on fp32 masters with
+    an optional bf16 mirror like FusedAdamW; norms and the update run as torch
+    (rocBLAS) ops on-device.
+    """
+
+    def __init__(self, params, lr=1e-3, betas=(0.9, 0.999), eps=1e-6, weight_decay=0.01,
+                 clamp_trust_ratio=(0.0, 10.0), mirrors=None):
+        defaults = dict(lr=lr, betas=betas, eps=eps, weight_decay=weight_decay)
+        super().__init__(params, defaults)
+        self.clamp_trust_ratio = clamp_trust_ratio
+        self._mirrors = mirrors or {}
+
+    def set_mirror(self, param: torch.Tensor, mirror: torch.Tensor):
+        self._mirrors[param] = mirror
+
+    @torch.no_grad()
+    def step(self, closure=None):
+        loss = closure() if closure is not None else None
+        for group in self.param_groups:
+            beta1, beta2 = group["betas"]
+            for p in group["params"]:
+                if p.grad is None:
+                    continue
+                g = p.grad.float()
+                state = self.state[p]
+                if len(state) == 0:
+                    state["step"] = 0
+                    state["exp_avg"] = torch.zeros_like(p, dtype=torch.float32)
+                    state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
+                state["step"] += 1
+                m, v = state["exp_avg"], state["exp_avg_sq"]
+                m.mul_(beta1).add_(g, alpha=1 - beta1)
+                v.mul_(beta2).addcmul_(g, g, value=1 - beta2)
+                m_hat = m / (1 - beta1 ** state["step"])
+                v_hat = v / (1 - beta2 ** state["step"])
+                update = m_hat / (v_hat.sqrt() + group["eps"])
+                if group["weight_decay"]:
+                    update = update + group["weight_decay"] * p.float()
+                p_norm = p.detach().float().norm()
+                u_norm = update.norm()
+                trust_ratio = torch.where(
+                    (p_norm > 0) & (u_norm > 0), p_norm / u_norm, torch.ones_like(p_norm)
+                ).clamp(*self.clamp_trust_ratio)
+                p.data.add_(update.to(p.dtype), alpha=-group["lr"] * float(trust_ratio))
+                mirror = self._mirrors.get(p)
+                if mirror is not None:
+                    mirror.copy_(p.data.to(mirror.dtype))
+        return loss
+
+
 class FusedAdamW(torch.optim.Optimizer):
     """AdamW whose step is one HIP kernel per parameter on the GPU.
 
