@@ -400,8 +400,14 @@ __global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
   int cols2 = cols >> 1;
 
   float accg0[PAIRS], accg1[PAIRS], accb0[PAIRS], accb1[PAIRS];
+  float gam0[PAIRS], gam1[PAIRS];  // each lane's gamma columns, loaded once
 #pragma unroll
-  for (int p = 0; p < PAIRS; ++p) accg0[p] = accg1[p] = accb0[p] = accb1[p] = 0.f;
+  for (int p = 0; p < PAIRS; ++p) {
+    accg0[p] = accg1[p] = accb0[p] = accb1[p] = 0.f;
+    int c = lane + p * 64;
+    gam0[p] = (c < cols2) ? gamma[2 * c] : 0.f;
+    gam1[p] = (c < cols2) ? gamma[2 * c + 1] : 0.f;
+  }
 
   for (long long row = (long long)blockIdx.x * waves_per_block + wave; row < rows;
        row += (long long)gridDim.x * waves_per_block) {
@@ -420,7 +426,7 @@ __global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
       float dy1 = bf16_to_f32((ushort_t)(dyv >> 16));
       float xh0 = (bf16_to_f32((ushort_t)(hv & 0xffff)) - mu) * rs;
       float xh1 = (bf16_to_f32((ushort_t)(hv >> 16)) - mu) * rs;
-      float dg0 = dy0 * gamma[2 * c], dg1 = dy1 * gamma[2 * c + 1];
+      float dg0 = dy0 * gam0[p], dg1 = dy1 * gam1[p];
       c1 += dg0 * xh0 + dg1 * xh1;
       c2 += dg0 + dg1;
       accg0[p] += dy0 * xh0;
@@ -441,8 +447,8 @@ __global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
       uint32_t dyv = dyr2[c], hv = hr2[c];
       float xh0 = (bf16_to_f32((ushort_t)(hv & 0xffff)) - mu) * rs;
       float xh1 = (bf16_to_f32((ushort_t)(hv >> 16)) - mu) * rs;
-      float dg0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gamma[2 * c];
-      float dg1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gamma[2 * c + 1];
+      float dg0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gam0[p];
+      float dg1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gam1[p];
       float o0 = (dg0 - c2 - xh0 * c1) * rs;
       float o1 = (dg1 - c2 - xh1 * c1) * rs;
       dxr2[c] = (uint32_t)f32_to_bf16(o0) | ((uint32_t)f32_to_bf16(o1) << 16);

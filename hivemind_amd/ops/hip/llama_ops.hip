@@ -62,8 +62,14 @@ __global__ void rmsnorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
   const int cols2 = cols >> 1;
 
   float accg0[PAIRS], accg1[PAIRS];
+  float gam0[PAIRS], gam1[PAIRS];  // per-thread gamma columns, loaded once
 #pragma unroll
-  for (int p = 0; p < PAIRS; ++p) accg0[p] = accg1[p] = 0.f;
+  for (int p = 0; p < PAIRS; ++p) {
+    accg0[p] = accg1[p] = 0.f;
+    int c = tid + p * blockDim.x;
+    gam0[p] = (c < cols2) ? gamma[2 * c] : 0.f;
+    gam1[p] = (c < cols2) ? gamma[2 * c + 1] : 0.f;
+  }
 
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     const uint32_t* dyr2 = reinterpret_cast<const uint32_t*>(dy + row * cols);
@@ -81,7 +87,7 @@ __global__ void rmsnorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
       float dy1 = bf16_to_f32((ushort_t)(dyv >> 16));
       float x0 = bf16_to_f32((ushort_t)(xv & 0xffff));
       float x1 = bf16_to_f32((ushort_t)(xv >> 16));
-      dot += dy0 * gamma[2 * c] * x0 + dy1 * gamma[2 * c + 1] * x1;
+      dot += dy0 * gam0[p] * x0 + dy1 * gam1[p] * x1;
       accg0[p] += dy0 * x0 * rs;
       accg1[p] += dy1 * x1 * rs;
     }
@@ -99,8 +105,8 @@ __global__ void rmsnorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
       float dy1 = bf16_to_f32((ushort_t)(dyv >> 16));
       float x0 = bf16_to_f32((ushort_t)(xv & 0xffff));
       float x1 = bf16_to_f32((ushort_t)(xv >> 16));
-      float o0 = gamma[2 * c] * dy0 * rs - x0 * k;
-      float o1 = gamma[2 * c + 1] * dy1 * rs - x1 * k;
+      float o0 = gam0[p] * dy0 * rs - x0 * k;
+      float o1 = gam1[p] * dy1 * rs - x1 * k;
       dxr2[c] = (uint32_t)f32_to_bf16(o0) | ((uint32_t)f32_to_bf16(o1) << 16);
     }
     __syncthreads();
