@@ -84,6 +84,10 @@ def main():
         config = LlamaConfig.tiny()
     if not use_gpu:
         config.dtype = torch.float32  # CPU bf16 matmuls are pathologically slow
+    max_pos = getattr(config, "max_position_embeddings", None)
+    if max_pos is not None and args.seq_len > max_pos:
+        log(f"clamping seq_len {args.seq_len} -> {max_pos} (model max positions)")
+        args.seq_len = max_pos
 
     torch.manual_seed(1234 + rank)
     model_cls = LlamaForCausalLM if is_llama else AlbertForMaskedLM
@@ -214,4 +218,12 @@ if __name__ == "__main__":
         traceback.print_exc()
         # daemon threads (DHT loop, tracker) must not keep a failed bench alive
         os._exit(1)
-    os._exit(0)
+    # Exit normally so profilers (rocprofv3) can finalize their databases in
+    # C-level destructors, but arm a watchdog in case a lingering non-daemon
+    # thread would otherwise hang the process after a successful run.
+    import threading
+
+    watchdog = threading.Timer(30.0, lambda: os._exit(0))
+    watchdog.daemon = True
+    watchdog.start()
+    sys.exit(0)
