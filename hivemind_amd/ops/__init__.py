@@ -119,6 +119,38 @@ def fused_bias_gelu(x: torch.Tensor, bias: torch.Tensor) -> torch.Tensor:
     return torch.nn.functional.gelu((x.float() + bias.float()), approximate="tanh").to(x.dtype)
 
 
+class _FusedCrossEntropy(torch.autograd.Function):
+    """Masked-LM cross-entropy that never materializes the [N, V] log-softmax:
+    forward saves only the per-row logsumexp (fp32 [N]); backward recomputes
+    softmax from the bf16 logits in one streaming pass. The mean is taken over
+    rows with label != ignore_index, matching F.cross_entropy(ignore_index=-100).
+    All reductions stay on device -- no host sync in either direction."""
+
+    @staticmethod
+    def forward(ctx, logits, labels):
+        loss_sum, valid, lse = hip_ops().cross_entropy_fwd(logits.contiguous(), labels.contiguous())
+        ctx.save_for_backward(logits, labels, lse, valid)
+        return loss_sum / valid.clamp_min(1).to(torch.float32)
+
+    @staticmethod
+    def backward(ctx, dloss):
+        logits, labels, lse, valid = ctx.saved_tensors
+        dlogits = hip_ops().cross_entropy_bwd(
+            logits.contiguous(), labels.contiguous(), lse,
+            dloss.reshape(()).to(torch.float32).contiguous(), valid,
+        )
+        return dlogits, None
+
+
+def fused_cross_entropy(logits: torch.Tensor, labels: torch.Tensor) -> torch.Tensor:
+    """Mean cross-entropy over labels != -100 (the MLM loss of the ALBERT
+    example). GPU bf16 path runs the fused kernels; everything else falls back
+    to F.cross_entropy in fp32."""
+    if logits.is_cuda and logits.dtype == torch.bfloat16 and logits.dim() == 2:
+        return _FusedCrossEntropy.apply(logits, labels)
+    return torch.nn.functional.cross_entropy(logits.float(), labels, ignore_index=-100)
+
+
 # ---------------------------------------------------------------------------
 # hand-written MFMA GEMM (guide §5 structure): x @ W^T with fused bias+GELU
 # ---------------------------------------------------------------------------

@@ -141,6 +141,39 @@ void fused_adamw_(torch::Tensor param, torch::Tensor grad, torch::Tensor exp_avg
   }
 }
 
+// ------------------------------------------------------------ cross-entropy
+
+std::vector<torch::Tensor> cross_entropy_fwd(torch::Tensor logits, torch::Tensor labels) {
+  CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_CONTIG(labels);
+  TORCH_CHECK(logits.scalar_type() == torch::kBFloat16, "cross_entropy expects bf16 logits");
+  TORCH_CHECK(labels.scalar_type() == torch::kInt64, "cross_entropy expects int64 labels");
+  TORCH_CHECK(logits.dim() == 2 && labels.dim() == 1 && labels.size(0) == logits.size(0),
+              "cross_entropy: logits [N, V], labels [N]");
+  int n_rows = (int)logits.size(0), n_cols = (int)logits.size(1);
+  auto lse = torch::empty({n_rows}, logits.options().dtype(torch::kFloat32));
+  auto loss_sum = torch::zeros({}, logits.options().dtype(torch::kFloat32));
+  auto valid = torch::zeros({}, logits.options().dtype(torch::kInt32));
+  hipLaunchKernelGGL(cross_entropy_fwd_bf16, dim3(n_rows), dim3(256), 0, current_stream(),
+                     (const unsigned short*)logits.data_ptr(), (const long long*)labels.data_ptr(),
+                     lse.data_ptr<float>(), loss_sum.data_ptr<float>(), valid.data_ptr<int>(),
+                     n_rows, n_cols);
+  return {loss_sum, valid, lse};
+}
+
+torch::Tensor cross_entropy_bwd(torch::Tensor logits, torch::Tensor labels, torch::Tensor lse,
+                                torch::Tensor upstream, torch::Tensor valid) {
+  CHECK_GPU(logits); CHECK_CONTIG(logits); CHECK_CONTIG(labels); CHECK_CONTIG(lse);
+  TORCH_CHECK(upstream.scalar_type() == torch::kFloat32 && upstream.numel() == 1,
+              "cross_entropy_bwd: upstream must be a single fp32 value on device");
+  int n_rows = (int)logits.size(0), n_cols = (int)logits.size(1);
+  auto dlogits = torch::empty_like(logits);
+  hipLaunchKernelGGL(cross_entropy_bwd_bf16, dim3(n_rows), dim3(256), 0, current_stream(),
+                     (const unsigned short*)logits.data_ptr(), (const long long*)labels.data_ptr(),
+                     lse.data_ptr<float>(), upstream.data_ptr<float>(), valid.data_ptr<int>(),
+                     (unsigned short*)dlogits.data_ptr(), n_rows, n_cols);
+  return dlogits;
+}
+
 // --------------------------------------------------------------- activations
 
 std::vector<torch::Tensor> bias_gelu_fwd(torch::Tensor x, torch::Tensor bias, bool save_pre_act) {
@@ -400,6 +433,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("quantize_blockwise", &quantize_blockwise_gpu, "blockwise int8 quantize -> (q, absmax)");
   m.def("dequantize_blockwise", &dequantize_blockwise_gpu, "blockwise int8 dequantize");
   m.def("fused_adamw_", &fused_adamw_, "fused AdamW step on fp32 master params");
+  m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused MLM cross-entropy fwd -> (loss_sum, valid, lse)");
+  m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused cross-entropy bwd -> dlogits");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "out = gelu(x + bias)");
   m.def("bias_gelu_bwd", &bias_gelu_bwd, "backward of bias+gelu");
   m.def("layernorm_fwd", &layernorm_fwd, "fused (residual+)layernorm forward");

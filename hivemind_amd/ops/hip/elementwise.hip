@@ -464,3 +464,105 @@ __global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
     atomicAdd(&dbeta[2 * c + 1], accb1[p]);
   }
 }
+
+// ---------------------------------------------------------------------------
+// Fused masked-LM cross-entropy over a large vocab (reference: the ALBERT
+// example's MLM loss; torch's F.cross_entropy materializes a [N, V] log-softmax
+// for backward -- 3.9 GB at N=65536, V=30000). This version saves only the
+// per-row logsumexp (fp32 [N]) and recomputes softmax in backward: two
+// streaming passes over the bf16 logits, no giant activation.
+//
+// fwd: one block (256 threads) per row; online max/sum-exp merge in registers,
+//      then wave shuffles + LDS. Rows with label < 0 (ignore_index) contribute
+//      nothing. Emits sum of losses + count of valid rows via global atomics.
+// bwd: dlogits = (softmax - onehot(label)) * (*upstream) / max(*valid, 1),
+//      zero for ignored rows. upstream/valid stay on device (no host sync).
+// ---------------------------------------------------------------------------
+extern "C" __global__ void cross_entropy_fwd_bf16(const ushort_t* __restrict__ logits,
+                                                  const long long* __restrict__ labels,
+                                                  float* __restrict__ lse,
+                                                  float* __restrict__ loss_sum,
+                                                  int* __restrict__ valid_count,
+                                                  int n_rows, int n_cols) {
+  int row = blockIdx.x;
+  if (row >= n_rows) return;
+  const ushort_t* lrow = logits + (long long)row * n_cols;
+  const uint32_t* l2 = reinterpret_cast<const uint32_t*>(lrow);
+  int tid = threadIdx.x;
+  int n2 = n_cols >> 1;
+  float m = -__builtin_inff(), s = 0.f;
+  for (int i = tid; i < n2; i += blockDim.x) {
+    uint32_t v = l2[i];
+    float a = bf16_to_f32((ushort_t)(v & 0xffff));
+    float b = bf16_to_f32((ushort_t)(v >> 16));
+    float mn = fmaxf(m, fmaxf(a, b));
+    s = s * __expf(m - mn) + __expf(a - mn) + __expf(b - mn);
+    m = mn;
+  }
+  if ((n_cols & 1) && tid == 0) {
+    float a = bf16_to_f32(lrow[n_cols - 1]);
+    float mn = fmaxf(m, a);
+    s = s * __expf(m - mn) + __expf(a - mn);
+    m = mn;
+  }
+  // merge partial (m, s) across the wave, then across waves via LDS
+  for (int off = 32; off > 0; off >>= 1) {
+    float mo = __shfl_down(m, off), so = __shfl_down(s, off);
+    float mn = fmaxf(m, mo);
+    s = s * __expf(m - mn) + so * __expf(mo - mn);
+    m = mn;
+  }
+  __shared__ float sm[4], ss[4];
+  int lane = tid & 63, wave = tid >> 6;
+  if (lane == 0) { sm[wave] = m; ss[wave] = s; }
+  __syncthreads();
+  if (tid == 0) {
+    float M = sm[0], S = ss[0];
+    for (int w = 1; w < (int)(blockDim.x >> 6); ++w) {
+      float mn = fmaxf(M, sm[w]);
+      S = S * __expf(M - mn) + ss[w] * __expf(sm[w] - mn);
+      M = mn;
+    }
+    float l = __logf(S) + M;
+    lse[row] = l;
+    long long lab = labels[row];
+    if (lab >= 0 && lab < n_cols) {
+      atomicAdd(loss_sum, l - bf16_to_f32(lrow[lab]));
+      atomicAdd(valid_count, 1);
+    }
+  }
+}
+
+extern "C" __global__ void cross_entropy_bwd_bf16(const ushort_t* __restrict__ logits,
+                                                  const long long* __restrict__ labels,
+                                                  const float* __restrict__ lse,
+                                                  const float* __restrict__ upstream,
+                                                  const int* __restrict__ valid_count,
+                                                  ushort_t* __restrict__ dlogits,
+                                                  int n_rows, int n_cols) {
+  int row = blockIdx.x;
+  if (row >= n_rows) return;
+  const ushort_t* lrow = logits + (long long)row * n_cols;
+  const uint32_t* l2 = reinterpret_cast<const uint32_t*>(lrow);
+  uint32_t* d2 = reinterpret_cast<uint32_t*>(dlogits + (long long)row * n_cols);
+  int tid = threadIdx.x;
+  int n2 = n_cols >> 1;
+  long long lab = labels[row];
+  float l = lse[row];
+  int nvalid = *valid_count;
+  float gs = (lab >= 0) ? (*upstream) / (float)(nvalid > 0 ? nvalid : 1) : 0.f;
+  for (int i = tid; i < n2; i += blockDim.x) {
+    uint32_t v = l2[i];
+    int c0 = 2 * i, c1 = 2 * i + 1;
+    float p0 = __expf(bf16_to_f32((ushort_t)(v & 0xffff)) - l);
+    float p1 = __expf(bf16_to_f32((ushort_t)(v >> 16)) - l);
+    float g0 = (p0 - (c0 == lab ? 1.f : 0.f)) * gs;
+    float g1 = (p1 - (c1 == lab ? 1.f : 0.f)) * gs;
+    d2[i] = (uint32_t)f32_to_bf16(g0) | ((uint32_t)f32_to_bf16(g1) << 16);
+  }
+  if ((n_cols & 1) && tid == 0) {
+    int c = n_cols - 1;
+    float p = __expf(bf16_to_f32(lrow[c]) - l);
+    dlogits[(long long)row * n_cols + c] = f32_to_bf16((p - (c == lab ? 1.f : 0.f)) * gs);
+  }
+}

@@ -343,3 +343,40 @@ def test_mfma_linear_gelu_forward_backward():
     assert (x.grad.float() - x_ref.grad).abs().max() / scale < 5e-2
     assert (w.grad.float() - w_ref.grad).abs().max() / w_ref.grad.abs().max() < 5e-2
     assert (b.grad.float() - b_ref.grad).abs().max() / (b_ref.grad.abs().max() + 1) < 5e-2
+
+
+def test_fused_cross_entropy():
+    """Fused MLM cross-entropy (loss + dlogits) vs fp32 F.cross_entropy with
+    ignore_index=-100, including an all-ignored batch edge case."""
+    from hivemind_amd.ops import fused_cross_entropy
+
+    torch.manual_seed(7)
+    N, V = 4096, 3000
+    logits = (torch.randn(N, V, device="cuda") * 3).bfloat16().requires_grad_(True)
+    labels = torch.randint(0, V, (N,), device="cuda")
+    labels[torch.rand(N, device="cuda") > 0.15] = -100  # MLM-style sparsity
+
+    loss = fused_cross_entropy(logits, labels)
+    loss.backward()
+
+    ref_logits = logits.detach().float().requires_grad_(True)
+    ref_loss = torch.nn.functional.cross_entropy(ref_logits, labels, ignore_index=-100)
+    ref_loss.backward()
+
+    assert torch.allclose(loss.float(), ref_loss, atol=2e-2, rtol=1e-3), (loss.item(), ref_loss.item())
+    assert torch.allclose(logits.grad.float(), ref_logits.grad, atol=2e-4), (
+        (logits.grad.float() - ref_logits.grad).abs().max()
+    )
+
+    # upstream gradient scaling must propagate (loss * 2).backward()
+    logits2 = logits.detach().clone().requires_grad_(True)
+    (fused_cross_entropy(logits2, labels) * 2).backward()
+    assert torch.allclose(logits2.grad.float(), 2 * ref_logits.grad, atol=4e-4)
+
+    # every label ignored: loss 0-safe, zero grads
+    logits3 = logits.detach().clone().requires_grad_(True)
+    all_ignored = torch.full((N,), -100, dtype=torch.int64, device="cuda")
+    loss3 = fused_cross_entropy(logits3, all_ignored)
+    loss3.backward()
+    assert loss3.item() == 0.0
+    assert logits3.grad.abs().max().item() == 0.0
