@@ -203,14 +203,15 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act
   TORCH_CHECK(cols % 2 == 0, "bias_gelu_bwd requires an even column count");
   long long rows = dy.numel() / cols;
   auto dx = torch::empty_like(dy);
-  auto dbias = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  auto dbias = torch::empty({cols}, dy.options().dtype(torch::kFloat32));
   TORCH_CHECK(cols <= 2 * 8 * 256, "bias_gelu_bwd: cols must be <= 4096");
   int blocks = (int)std::min<long long>(rows, 2048);
+  auto partial = torch::empty({blocks, cols}, dy.options().dtype(torch::kFloat32));
   int pairs = (int)((cols / 2 + 255) / 256);
   auto launch = [&](auto kernel) {
     hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
                        (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
-                       (unsigned short*)dx.data_ptr(), dbias.data_ptr<float>(), rows, cols);
+                       (unsigned short*)dx.data_ptr(), partial.data_ptr<float>(), rows, cols);
   };
   switch (pairs) {
     case 1: launch(bias_gelu_bwd_bf16_t<1>); break;
@@ -220,6 +221,8 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act
     case 5: case 6: launch(bias_gelu_bwd_bf16_t<6>); break;
     default: launch(bias_gelu_bwd_bf16_t<8>); break;
   }
+  hipLaunchKernelGGL(reduce_block_partials_f32, dim3((int)((cols + 255) / 256)), dim3(256), 0,
+                     current_stream(), partial.data_ptr<float>(), dbias.data_ptr<float>(), blocks, cols);
   return {dx, dbias};
 }
 
@@ -262,16 +265,16 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torc
   int cols = (int)dy.size(-1);
   long long rows = dy.numel() / cols;
   auto dx = torch::empty_like(dy);
-  auto dgamma = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
-  auto dbeta = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  auto fused = torch::empty({2 * (long long)cols}, dy.options().dtype(torch::kFloat32));
   TORCH_CHECK(cols % 2 == 0 && cols <= 2 * 64 * 16, "layernorm_bwd: cols must be even and <= 2048");
   int blocks = (int)std::min<long long>((rows + 3) / 4, 2048);
+  auto partial = torch::empty({blocks, 2 * (long long)cols}, dy.options().dtype(torch::kFloat32));
   int pairs = (cols / 2 + 63) / 64;
   auto launch = [&](auto kernel) {
     hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
                        (const unsigned short*)dy.data_ptr(), (const unsigned short*)h.data_ptr(),
                        gamma.data_ptr<float>(), mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                       (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),
+                       (unsigned short*)dx.data_ptr(), partial.data_ptr<float>(),
                        rows, cols);
   };
   switch (pairs) {
@@ -283,6 +286,11 @@ std::vector<torch::Tensor> layernorm_bwd(torch::Tensor dy, torch::Tensor h, torc
     case 9: case 10: case 11: case 12: launch(layernorm_bwd_bf16_t<12>); break;
     default: launch(layernorm_bwd_bf16_t<16>); break;
   }
+  hipLaunchKernelGGL(reduce_block_partials_f32, dim3((int)((2 * cols + 255) / 256)), dim3(256), 0,
+                     current_stream(), partial.data_ptr<float>(), fused.data_ptr<float>(),
+                     blocks, 2 * (long long)cols);
+  auto dgamma = fused.narrow(0, 0, cols);
+  auto dbeta = fused.narrow(0, cols, cols);
   return {dx, dgamma, dbeta};
 }
 
@@ -309,14 +317,15 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch:
   TORCH_CHECK(cols % 2 == 0 && cols <= 2 * 16 * 256, "rmsnorm_bwd: cols must be even and <= 8192");
   long long rows = dy.numel() / cols;
   auto dx = torch::empty_like(dy);
-  auto dgamma = torch::zeros({cols}, dy.options().dtype(torch::kFloat32));
+  auto dgamma = torch::empty({cols}, dy.options().dtype(torch::kFloat32));
   int blocks = (int)std::min<long long>(rows, 2048);
+  auto partial = torch::empty({blocks, (long long)cols}, dy.options().dtype(torch::kFloat32));
   int pairs = (cols / 2 + 255) / 256;
   auto launch = [&](auto kernel) {
     hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
                        (const unsigned short*)dy.data_ptr(), (const unsigned short*)x.data_ptr(),
                        gamma.data_ptr<float>(), rstd.data_ptr<float>(),
-                       (unsigned short*)dx.data_ptr(), dgamma.data_ptr<float>(), rows, cols);
+                       (unsigned short*)dx.data_ptr(), partial.data_ptr<float>(), rows, cols);
   };
   switch (pairs) {
     case 1: launch(rmsnorm_bwd_bf16_t<1>); break;
@@ -325,6 +334,9 @@ std::vector<torch::Tensor> rmsnorm_bwd(torch::Tensor dy, torch::Tensor x, torch:
     case 5: case 6: case 7: case 8: launch(rmsnorm_bwd_bf16_t<8>); break;
     default: launch(rmsnorm_bwd_bf16_t<16>); break;
   }
+  hipLaunchKernelGGL(reduce_block_partials_f32, dim3((int)((cols + 255) / 256)), dim3(256), 0,
+                     current_stream(), partial.data_ptr<float>(), dgamma.data_ptr<float>(),
+                     blocks, (long long)cols);
   return {dx, dgamma};
 }
 

@@ -270,7 +270,7 @@ template <int PAIRS>
 __global__ void bias_gelu_bwd_bf16_t(const ushort_t* __restrict__ dy,
                                      const ushort_t* __restrict__ pre_act,
                                      ushort_t* __restrict__ dx,
-                                     float* __restrict__ dbias,
+                                     float* __restrict__ partial,  // [gridDim, cols] per-block dbias
                                      long long rows, long long cols) {
   const uint32_t* dy2 = reinterpret_cast<const uint32_t*>(dy);
   const uint32_t* pre2 = reinterpret_cast<const uint32_t*>(pre_act);
@@ -295,12 +295,17 @@ __global__ void bias_gelu_bwd_bf16_t(const ushort_t* __restrict__ dy,
       acc1[p] += g1;
     }
   }
+  // each column is owned by exactly one thread in this block: store the
+  // block's partial sums contiguously; a tiny second kernel reduces over
+  // blocks. Global float atomics here would serialize gridDim-deep per
+  // address (measured ~200 us of pure atomic tail at gridDim=2048).
+  float* prow = partial + (long long)blockIdx.x * cols;
 #pragma unroll
   for (int p = 0; p < PAIRS; ++p) {
     long long c = threadIdx.x + (long long)p * blockDim.x;
     if (c >= cols2) break;
-    atomicAdd(&dbias[2 * c], acc0[p]);
-    atomicAdd(&dbias[2 * c + 1], acc1[p]);
+    prow[2 * c] = acc0[p];
+    prow[2 * c + 1] = acc1[p];
   }
 }
 
@@ -391,8 +396,7 @@ __global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
                                      const float* __restrict__ mean,
                                      const float* __restrict__ rstd,
                                      ushort_t* __restrict__ dx,
-                                     float* __restrict__ dgamma,
-                                     float* __restrict__ dbeta,
+                                     float* __restrict__ partial,  // [gridDim, 2*cols]: dgamma then dbeta
                                      long long rows, int cols) {
   int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
@@ -454,15 +458,26 @@ __global__ void layernorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
       dxr2[c] = (uint32_t)f32_to_bf16(o0) | ((uint32_t)f32_to_bf16(o1) << 16);
     }
   }
+  // columns are owned per WAVE here (4 waves share each column): merge the
+  // waves' register accumulators through LDS (depth-4 LDS atomics), then
+  // store ONE per-block partial row; a second kernel reduces over blocks.
+  // Global atomics would be 4*gridDim-deep per address (~8192 at gridDim
+  // 2048) and dominated this kernel's runtime.
+  __shared__ float smem[2 * PAIRS * 128];  // [dgamma(cols) | dbeta(cols)], cols <= PAIRS*128
+  for (int c = threadIdx.x; c < 2 * cols; c += blockDim.x) smem[c] = 0.f;
+  __syncthreads();
 #pragma unroll
   for (int p = 0; p < PAIRS; ++p) {
     int c = lane + p * 64;
     if (c >= cols2) break;
-    atomicAdd(&dgamma[2 * c], accg0[p]);
-    atomicAdd(&dbeta[2 * c], accb0[p]);
-    atomicAdd(&dgamma[2 * c + 1], accg1[p]);
-    atomicAdd(&dbeta[2 * c + 1], accb1[p]);
+    atomicAdd(&smem[2 * c], accg0[p]);
+    atomicAdd(&smem[2 * c + 1], accg1[p]);
+    atomicAdd(&smem[cols + 2 * c], accb0[p]);
+    atomicAdd(&smem[cols + 2 * c + 1], accb1[p]);
   }
+  __syncthreads();
+  float* prow = partial + (long long)blockIdx.x * (2 * cols);
+  for (int c = threadIdx.x; c < 2 * cols; c += blockDim.x) prow[c] = smem[c];
 }
 
 // ---------------------------------------------------------------------------
@@ -565,4 +580,20 @@ extern "C" __global__ void cross_entropy_bwd_bf16(const ushort_t* __restrict__ l
     float p = __expf(bf16_to_f32(lrow[c]) - l);
     dlogits[(long long)row * n_cols + c] = f32_to_bf16((p - (c == lab ? 1.f : 0.f)) * gs);
   }
+}
+
+
+// ---------------------------------------------------------------------------
+// Reduce per-block partial rows [nblocks, width] -> out[width] (column sums).
+// Deterministic replacement for contended global fp32 atomics in the norm /
+// bias backward kernels: the partial matrix is a few MB, one streaming pass.
+// ---------------------------------------------------------------------------
+extern "C" __global__ void reduce_block_partials_f32(const float* __restrict__ partial,
+                                                     float* __restrict__ out,
+                                                     int nblocks, long long width) {
+  long long c = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= width) return;
+  float s = 0.f;
+  for (int b = 0; b < nblocks; ++b) s += partial[(long long)b * width + c];
+  out[c] = s;
 }

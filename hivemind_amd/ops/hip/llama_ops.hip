@@ -55,7 +55,7 @@ __global__ void rmsnorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
                                    const float* __restrict__ gamma,
                                    const float* __restrict__ rstd,
                                    ushort_t* __restrict__ dx,
-                                   float* __restrict__ dgamma,
+                                   float* __restrict__ partial,  // [gridDim, cols] per-block dgamma
                                    long long rows, int cols) {
   __shared__ float wave_sums[4];
   const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
@@ -111,12 +111,16 @@ __global__ void rmsnorm_bwd_bf16_t(const ushort_t* __restrict__ dy,
     }
     __syncthreads();
   }
+  // one owner thread per column in this block: store the block partial row;
+  // reduce_block_partials_f32 sums over blocks (contended global fp32 atomics
+  // serialize gridDim-deep per address and dominated this kernel).
+  float* prow = partial + (long long)blockIdx.x * cols;
 #pragma unroll
   for (int p = 0; p < PAIRS; ++p) {
     int c = tid + p * blockDim.x;
     if (c >= cols2) break;
-    atomicAdd(&dgamma[2 * c], accg0[p]);
-    atomicAdd(&dgamma[2 * c + 1], accg1[p]);
+    prow[2 * c] = accg0[p];
+    prow[2 * c + 1] = accg1[p];
   }
 }
 
