@@ -582,18 +582,3 @@ extern "C" __global__ void cross_entropy_bwd_bf16(const ushort_t* __restrict__ l
   }
 }
 
-
-// ---------------------------------------------------------------------------
-// Reduce per-block partial rows [nblocks, width] -> out[width] (column sums).
-// Deterministic replacement for contended global fp32 atomics in the norm /
-// bias backward kernels: the partial matrix is a few MB, one streaming pass.
-// ---------------------------------------------------------------------------
-extern "C" __global__ void reduce_block_partials_f32(const float* __restrict__ partial,
-                                                     float* __restrict__ out,
-                                                     int nblocks, long long width) {
-  long long c = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= width) return;
-  float s = 0.f;
-  for (int b = 0; b < nblocks; ++b) s += partial[(long long)b * width + c];
-  out[c] = s;
-}
