@@ -98,17 +98,22 @@ def fused_layernorm(
 
 
 class _FusedBiasGelu(torch.autograd.Function):
+    """Forward writes only gelu(x + bias); backward recomputes x + bias from
+    the saved input (x is retained by autograd for the producing GEMM's
+    backward anyway), so no pre-activation tensor is ever materialized."""
+
     @staticmethod
     def forward(ctx, x, bias):
-        out, pre_act = hip_ops().bias_gelu_fwd(x.contiguous(), bias.contiguous(), True)
-        ctx.save_for_backward(pre_act)
+        x, bias = x.contiguous(), bias.contiguous()
+        (out,) = hip_ops().bias_gelu_fwd(x, bias, False)
+        ctx.save_for_backward(x, bias)
         ctx.bias_dtype = bias.dtype
         return out
 
     @staticmethod
     def backward(ctx, dy):
-        (pre_act,) = ctx.saved_tensors
-        dx, dbias = hip_ops().bias_gelu_bwd(dy.contiguous(), pre_act)
+        x, bias = ctx.saved_tensors
+        dx, dbias = hip_ops().bias_gelu_bwd_xb(dy.contiguous(), x, bias)
         return dx, dbias.to(ctx.bias_dtype)
 
 

@@ -197,8 +197,11 @@ std::vector<torch::Tensor> bias_gelu_fwd(torch::Tensor x, torch::Tensor bias, bo
   return {out};
 }
 
-std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act) {
-  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(pre_act);
+// shared driver: pre_or_x is either a materialized pre-activation (MFMA
+// fused-epilogue path) or the raw GEMM output with `bias` added in-kernel.
+static std::vector<torch::Tensor> bias_gelu_bwd_impl(torch::Tensor dy, torch::Tensor pre_or_x,
+                                                     const unsigned short* bias_ptr) {
+  CHECK_GPU(dy); CHECK_CONTIG(dy); CHECK_CONTIG(pre_or_x);
   long long cols = dy.size(-1);
   TORCH_CHECK(cols % 2 == 0, "bias_gelu_bwd requires an even column count");
   long long rows = dy.numel() / cols;
@@ -209,21 +212,42 @@ std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act
   int pairs = (int)((cols / 2 + 255) / 256);
   auto launch = [&](auto kernel) {
     hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
-                       (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_act.data_ptr(),
-                       (unsigned short*)dx.data_ptr(), partial.data_ptr<float>(), rows, cols);
+                       (const unsigned short*)dy.data_ptr(), (const unsigned short*)pre_or_x.data_ptr(),
+                       bias_ptr, (unsigned short*)dx.data_ptr(), partial.data_ptr<float>(), rows, cols);
   };
-  switch (pairs) {
-    case 1: launch(bias_gelu_bwd_bf16_t<1>); break;
-    case 2: launch(bias_gelu_bwd_bf16_t<2>); break;
-    case 3: launch(bias_gelu_bwd_bf16_t<3>); break;
-    case 4: launch(bias_gelu_bwd_bf16_t<4>); break;
-    case 5: case 6: launch(bias_gelu_bwd_bf16_t<6>); break;
-    default: launch(bias_gelu_bwd_bf16_t<8>); break;
+  if (bias_ptr) {
+    switch (pairs) {
+      case 1: launch(bias_gelu_bwd_bf16_t<1, true>); break;
+      case 2: launch(bias_gelu_bwd_bf16_t<2, true>); break;
+      case 3: launch(bias_gelu_bwd_bf16_t<3, true>); break;
+      case 4: launch(bias_gelu_bwd_bf16_t<4, true>); break;
+      case 5: case 6: launch(bias_gelu_bwd_bf16_t<6, true>); break;
+      default: launch(bias_gelu_bwd_bf16_t<8, true>); break;
+    }
+  } else {
+    switch (pairs) {
+      case 1: launch(bias_gelu_bwd_bf16_t<1, false>); break;
+      case 2: launch(bias_gelu_bwd_bf16_t<2, false>); break;
+      case 3: launch(bias_gelu_bwd_bf16_t<3, false>); break;
+      case 4: launch(bias_gelu_bwd_bf16_t<4, false>); break;
+      case 5: case 6: launch(bias_gelu_bwd_bf16_t<6, false>); break;
+      default: launch(bias_gelu_bwd_bf16_t<8, false>); break;
+    }
   }
   // column-reduce the per-block partials with ATen's tuned reducer (a naive
   // one-thread-per-column kernel here measured 535 us -- latency-bound)
   auto dbias = partial.sum(0);
   return {dx, dbias};
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd(torch::Tensor dy, torch::Tensor pre_act) {
+  return bias_gelu_bwd_impl(dy, pre_act, nullptr);
+}
+
+std::vector<torch::Tensor> bias_gelu_bwd_xb(torch::Tensor dy, torch::Tensor x, torch::Tensor bias) {
+  CHECK_CONTIG(bias);
+  TORCH_CHECK(bias.scalar_type() == torch::kBFloat16 && bias.numel() == dy.size(-1));
+  return bias_gelu_bwd_impl(dy, x, (const unsigned short*)bias.data_ptr());
 }
 
 // ----------------------------------------------------------------- layernorm
@@ -249,12 +273,24 @@ std::vector<torch::Tensor> layernorm_fwd(torch::Tensor x, c10::optional<torch::T
     h_ptr = (unsigned short*)h.data_ptr();
   }
   int blocks = (int)((rows + 3) / 4);
-  hipLaunchKernelGGL(layernorm_fwd_bf16, dim3(blocks), dim3(256), 0, current_stream(),
-                     (const unsigned short*)x.data_ptr(), res_ptr,
-                     gamma.data_ptr<float>(), beta.data_ptr<float>(),
-                     (unsigned short*)y.data_ptr(), h_ptr,
-                     mean.data_ptr<float>(), rstd.data_ptr<float>(),
-                     (float)eps, rows, cols);
+  int pairs_fwd = (cols / 2 + 63) / 64;
+  TORCH_CHECK(pairs_fwd <= 16, "layernorm_fwd: cols must be <= 2048");
+  auto launch_fwd = [&](auto kernel) {
+    hipLaunchKernelGGL(kernel, dim3(blocks), dim3(256), 0, current_stream(),
+                       (const unsigned short*)x.data_ptr(), res_ptr,
+                       gamma.data_ptr<float>(), beta.data_ptr<float>(),
+                       (unsigned short*)y.data_ptr(), h_ptr,
+                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                       (float)eps, rows, cols);
+  };
+  switch (pairs_fwd) {
+    case 1: launch_fwd(layernorm_fwd_bf16_t<1>); break;
+    case 2: launch_fwd(layernorm_fwd_bf16_t<2>); break;
+    case 3: case 4: launch_fwd(layernorm_fwd_bf16_t<4>); break;
+    case 5: case 6: launch_fwd(layernorm_fwd_bf16_t<6>); break;
+    case 7: case 8: launch_fwd(layernorm_fwd_bf16_t<8>); break;
+    default: launch_fwd(layernorm_fwd_bf16_t<16>); break;
+  }
   if (residual.has_value()) return {y, mean, rstd, h};
   return {y, mean, rstd};
 }
@@ -442,7 +478,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cross_entropy_fwd", &cross_entropy_fwd, "fused MLM cross-entropy fwd -> (loss_sum, valid, lse)");
   m.def("cross_entropy_bwd", &cross_entropy_bwd, "fused cross-entropy bwd -> dlogits");
   m.def("bias_gelu_fwd", &bias_gelu_fwd, "out = gelu(x + bias)");
-  m.def("bias_gelu_bwd", &bias_gelu_bwd, "backward of bias+gelu");
+  m.def("bias_gelu_bwd", &bias_gelu_bwd, "backward of bias+gelu (from saved pre-activation)");
+  m.def("bias_gelu_bwd_xb", &bias_gelu_bwd_xb, "backward of bias+gelu (recomputes x+bias)");
   m.def("layernorm_fwd", &layernorm_fwd, "fused (residual+)layernorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "layernorm backward");
 }

@@ -266,20 +266,39 @@ extern "C" __global__ void bias_gelu_fwd_bf16(const ushort_t* __restrict__ x,
 // PAIRS is compile-time: runtime-indexed register arrays spill to scratch
 // (cdna_hip_programming.md rule #20). Dispatch in bindings covers
 // cols <= 2 * PAIRS_MAX * 256.
-template <int PAIRS>
+// FROM_XB: `pre` is the raw GEMM output x and the bias is added here (the
+// x tensor is saved by autograd for the GEMM backward anyway, so the forward
+// pass never writes a separate pre_act -- 384 MB less traffic per call at
+// ALBERT-base batch 128). FROM_XB=false: `pre` is a materialized pre_act
+// (the MFMA fused-epilogue path, where x never exists in memory).
+template <int PAIRS, bool FROM_XB>
 __global__ void bias_gelu_bwd_bf16_t(const ushort_t* __restrict__ dy,
-                                     const ushort_t* __restrict__ pre_act,
+                                     const ushort_t* __restrict__ pre,
+                                     const ushort_t* __restrict__ bias,  // null unless FROM_XB
                                      ushort_t* __restrict__ dx,
                                      float* __restrict__ partial,  // [gridDim, cols] per-block dbias
                                      long long rows, long long cols) {
   const uint32_t* dy2 = reinterpret_cast<const uint32_t*>(dy);
-  const uint32_t* pre2 = reinterpret_cast<const uint32_t*>(pre_act);
+  const uint32_t* pre2 = reinterpret_cast<const uint32_t*>(pre);
+  const uint32_t* b2 = reinterpret_cast<const uint32_t*>(bias);
   uint32_t* dx2 = reinterpret_cast<uint32_t*>(dx);
   long long cols2 = cols >> 1;
 
   float acc0[PAIRS], acc1[PAIRS];
+  float bia0[PAIRS], bia1[PAIRS];  // per-thread bias columns, loaded once
 #pragma unroll
-  for (int p = 0; p < PAIRS; ++p) acc0[p] = acc1[p] = 0.f;
+  for (int p = 0; p < PAIRS; ++p) {
+    acc0[p] = acc1[p] = 0.f;
+    bia0[p] = bia1[p] = 0.f;
+    if (FROM_XB) {
+      long long c = threadIdx.x + (long long)p * blockDim.x;
+      if (c < cols2) {
+        uint32_t bv = b2[c];
+        bia0[p] = bf16_to_f32((ushort_t)(bv & 0xffff));
+        bia1[p] = bf16_to_f32((ushort_t)(bv >> 16));
+      }
+    }
+  }
 
   for (long long row = blockIdx.x; row < rows; row += gridDim.x) {
     long long base = row * cols2;
@@ -288,8 +307,11 @@ __global__ void bias_gelu_bwd_bf16_t(const ushort_t* __restrict__ dy,
       long long c = threadIdx.x + (long long)p * blockDim.x;
       if (c >= cols2) break;
       uint32_t dyv = dy2[base + c], pv = pre2[base + c];
-      float g0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gelu_tanh_grad(bf16_to_f32((ushort_t)(pv & 0xffff)));
-      float g1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gelu_tanh_grad(bf16_to_f32((ushort_t)(pv >> 16)));
+      float p0 = bf16_to_f32((ushort_t)(pv & 0xffff));
+      float p1 = bf16_to_f32((ushort_t)(pv >> 16));
+      if (FROM_XB) { p0 += bia0[p]; p1 += bia1[p]; }
+      float g0 = bf16_to_f32((ushort_t)(dyv & 0xffff)) * gelu_tanh_grad(p0);
+      float g1 = bf16_to_f32((ushort_t)(dyv >> 16)) * gelu_tanh_grad(p1);
       dx2[base + c] = (uint32_t)f32_to_bf16(g0) | ((uint32_t)f32_to_bf16(g1) << 16);
       acc0[p] += g0;
       acc1[p] += g1;
@@ -310,46 +332,53 @@ __global__ void bias_gelu_bwd_bf16_t(const ushort_t* __restrict__ dy,
 }
 
 // ---------------------------------------------------------------------------
-// LayerNorm forward: one wave per row (cols <= 8192), bf16 in/out, fp32 stats.
-// Optionally fuses a residual add: h = x + residual; y = ln(h).
+// LayerNorm forward: one wave per row, bf16 in/out, fp32 stats. The row's
+// values are STASHED IN REGISTERS between the stats pass and the normalize
+// pass (PAIRS column-pairs per lane, compile-time -- rule #20), so the input
+// is read from HBM exactly once. Optionally fuses a residual add:
+// h = x + residual; y = ln(h). cols <= 2 * PAIRS * 64.
 // ---------------------------------------------------------------------------
-extern "C" __global__ void layernorm_fwd_bf16(const ushort_t* __restrict__ x,
-                                              const ushort_t* __restrict__ residual,  // may be null
-                                              const float* __restrict__ gamma,
-                                              const float* __restrict__ beta,
-                                              ushort_t* __restrict__ y,
-                                              ushort_t* __restrict__ h_out,  // saved normalized input source (x+res); may be null
-                                              float* __restrict__ mean_out,
-                                              float* __restrict__ rstd_out,
-                                              float eps, long long rows, int cols) {
+template <int PAIRS>
+__global__ void layernorm_fwd_bf16_t(const ushort_t* __restrict__ x,
+                                     const ushort_t* __restrict__ residual,  // may be null
+                                     const float* __restrict__ gamma,
+                                     const float* __restrict__ beta,
+                                     ushort_t* __restrict__ y,
+                                     ushort_t* __restrict__ h_out,  // saved normalized input source (x+res); may be null
+                                     float* __restrict__ mean_out,
+                                     float* __restrict__ rstd_out,
+                                     float eps, long long rows, int cols) {
   // blockDim.x = 256 -> 4 waves; each wave owns one row
   int wave = threadIdx.x >> 6;
   int lane = threadIdx.x & 63;
   long long row = (long long)blockIdx.x * 4 + wave;
   if (row >= rows) return;
-  const ushort_t* xr = x + row * cols;
-  const ushort_t* rr = residual ? residual + row * cols : nullptr;
-  ushort_t* hr = h_out ? h_out + row * cols : nullptr;
-  ushort_t* yr = y + row * cols;
-
-  // paired bf16 loads: 4 B/lane coalesced (scalar bf16 halves throughput, G13)
-  const uint32_t* xr2 = reinterpret_cast<const uint32_t*>(xr);
-  const uint32_t* rr2 = reinterpret_cast<const uint32_t*>(rr);
-  uint32_t* hr2 = reinterpret_cast<uint32_t*>(hr);
+  const uint32_t* xr2 = reinterpret_cast<const uint32_t*>(x + row * cols);
+  const uint32_t* rr2 = residual ? reinterpret_cast<const uint32_t*>(residual + row * cols) : nullptr;
+  uint32_t* hr2 = h_out ? reinterpret_cast<uint32_t*>(h_out + row * cols) : nullptr;
+  uint32_t* yr2 = reinterpret_cast<uint32_t*>(y + row * cols);
   int cols2 = cols >> 1;
+
+  float v0[PAIRS], v1[PAIRS];
   float sum = 0.f, sumsq = 0.f;
-  for (int c = lane; c < cols2; c += 64) {
-    uint32_t xv = xr2[c];
-    float v0 = bf16_to_f32((ushort_t)(xv & 0xffff));
-    float v1 = bf16_to_f32((ushort_t)(xv >> 16));
-    if (rr) {
-      uint32_t rv = rr2[c];
-      v0 += bf16_to_f32((ushort_t)(rv & 0xffff));
-      v1 += bf16_to_f32((ushort_t)(rv >> 16));
+#pragma unroll
+  for (int p = 0; p < PAIRS; ++p) {
+    int c = lane + p * 64;
+    v0[p] = v1[p] = 0.f;
+    if (c < cols2) {
+      uint32_t xv = xr2[c];
+      float a = bf16_to_f32((ushort_t)(xv & 0xffff));
+      float b = bf16_to_f32((ushort_t)(xv >> 16));
+      if (rr2) {
+        uint32_t rv = rr2[c];
+        a += bf16_to_f32((ushort_t)(rv & 0xffff));
+        b += bf16_to_f32((ushort_t)(rv >> 16));
+      }
+      if (hr2) hr2[c] = (uint32_t)f32_to_bf16(a) | ((uint32_t)f32_to_bf16(b) << 16);
+      v0[p] = a; v1[p] = b;
+      sum += a + b;
+      sumsq += a * a + b * b;
     }
-    if (hr) hr2[c] = (uint32_t)f32_to_bf16(v0) | ((uint32_t)f32_to_bf16(v1) << 16);
-    sum += v0 + v1;
-    sumsq += v0 * v0 + v1 * v1;
   }
   for (int off = 32; off > 0; off >>= 1) {
     sum += __shfl_down(sum, off);
@@ -364,19 +393,14 @@ extern "C" __global__ void layernorm_fwd_bf16(const ushort_t* __restrict__ x,
     mean_out[row] = mean;
     rstd_out[row] = rstd;
   }
-  uint32_t* yr2 = reinterpret_cast<uint32_t*>(yr);
-  for (int c = lane; c < cols2; c += 64) {
-    uint32_t xv = xr2[c];
-    float v0 = bf16_to_f32((ushort_t)(xv & 0xffff));
-    float v1 = bf16_to_f32((ushort_t)(xv >> 16));
-    if (rr) {
-      uint32_t rv = rr2[c];
-      v0 += bf16_to_f32((ushort_t)(rv & 0xffff));
-      v1 += bf16_to_f32((ushort_t)(rv >> 16));
+#pragma unroll
+  for (int p = 0; p < PAIRS; ++p) {
+    int c = lane + p * 64;
+    if (c < cols2) {
+      float n0 = fmaf((v0[p] - mean) * rstd, gamma[2 * c], beta[2 * c]);
+      float n1 = fmaf((v1[p] - mean) * rstd, gamma[2 * c + 1], beta[2 * c + 1]);
+      yr2[c] = (uint32_t)f32_to_bf16(n0) | ((uint32_t)f32_to_bf16(n1) << 16);
     }
-    float n0 = fmaf((v0 - mean) * rstd, gamma[2 * c], beta[2 * c]);
-    float n1 = fmaf((v1 - mean) * rstd, gamma[2 * c + 1], beta[2 * c + 1]);
-    yr2[c] = (uint32_t)f32_to_bf16(n0) | ((uint32_t)f32_to_bf16(n1) << 16);
   }
 }
 
