@@ -159,6 +159,7 @@ class DecentralizedAverager(ServicerBase):
         self.shutdown_timeout = shutdown_timeout
         self.declare_state_period = declare_state_period
         self._allow_state_sharing = bool(allow_state_sharing if allow_state_sharing is not None else not (client_mode or auxiliary))
+        self._state_sharing_priority: Optional[float] = None
 
         self._averaged_tensors = tuple(averaged_tensors)
         self.lock_averaged_tensors = threading.Lock()
@@ -555,9 +556,16 @@ class DecentralizedAverager(ServicerBase):
 
     @property
     def state_sharing_priority(self) -> float:
-        """Donors with higher priority are tried first; subclasses override
+        """Donors with higher priority are tried first. Assignable (reference
+        test_load_state_priority); subclasses override the getter
         (TrainingStateAverager uses local epoch)."""
+        if self._state_sharing_priority is not None:
+            return float(self._state_sharing_priority)
         return float(self._allow_state_sharing)
+
+    @state_sharing_priority.setter
+    def state_sharing_priority(self, value: float):
+        self._state_sharing_priority = float(value)
 
     async def _get_current_state_from_host_process(self) -> Tuple[Any, Sequence[torch.Tensor]]:
         """Hook: collect the state to share. Default: metadata=None + averaged tensors."""
@@ -592,12 +600,16 @@ class DecentralizedAverager(ServicerBase):
             for part in split_for_streaming(serialized):
                 yield DownloadData(tensor=part)
 
-    def load_state_from_peers(self, wait: bool = True, timeout: Optional[float] = None):
-        """Download the latest state from the best donor (reference averager.py:668-736)."""
-        future = asyncio.run_coroutine_threadsafe(self._load_state_from_peers(timeout), self._loop)
+    def load_state_from_peers(self, wait: bool = True, timeout: Optional[float] = None, apply: bool = False):
+        """Download the latest state from the best donor (reference averager.py:668-736).
+
+        Returns (metadata, tensors) without modifying local state unless
+        ``apply=True`` (reference semantics: the base class only downloads;
+        TrainingStateAverager overrides the default to install the state)."""
+        future = asyncio.run_coroutine_threadsafe(self._load_state_from_peers(timeout, apply), self._loop)
         return future.result(timeout) if wait else future
 
-    async def _load_state_from_peers(self, timeout: Optional[float] = None) -> Optional[Tuple[Any, Sequence[torch.Tensor]]]:
+    async def _load_state_from_peers(self, timeout: Optional[float] = None, apply: bool = False) -> Optional[Tuple[Any, Sequence[torch.Tensor]]]:
         key = f"{self.prefix}.all_averagers"
         result = await asyncio.wrap_future(self.dht.get(key, latest=True, return_future=True))
         if result is None or not isinstance(result.value, dict):
@@ -637,7 +649,8 @@ class DecentralizedAverager(ServicerBase):
                 if tensor_parts:
                     tensors.append(deserialize_torch_tensor(combine_from_streaming(tensor_parts)))
                 logger.info(f"downloaded state from {donor}: {len(tensors)} tensors")
-                await asyncio.get_event_loop().run_in_executor(None, self.load_state, metadata, tensors)
+                if apply:
+                    await asyncio.get_event_loop().run_in_executor(None, self.load_state, metadata, tensors)
                 return metadata, tensors
             except asyncio.CancelledError:
                 raise

@@ -360,6 +360,11 @@ class TrainingStateAverager(DecentralizedAverager):
         all_tensors = list(chain(optimized_parameters, extra_tensors, optimizer_tensors))
         return metadata, all_tensors
 
+    def load_state_from_peers(self, wait: bool = True, timeout=None, apply: bool = True):
+        """Training peers install the downloaded state by default (the base
+        averager only returns it -- reference TrainingStateAverager)."""
+        return super().load_state_from_peers(wait=wait, timeout=timeout, apply=apply)
+
     def load_state(self, metadata: Any, tensors: Sequence[torch.Tensor]):
         """Restore a downloaded state (reference state_averager.py:658-704)."""
         if metadata is None or not isinstance(metadata, dict):

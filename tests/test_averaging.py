@@ -123,7 +123,13 @@ def test_state_sharing():
         min_matchmaking_time=1.0, request_timeout=0.5, declare_state_period=1.0,
     )
     time.sleep(2.0)  # let the donor declare itself
+    # plain download returns the state without touching local tensors
     result = target.load_state_from_peers(timeout=20)
+    assert result is not None
+    with target.get_tensors() as tensors:
+        assert torch.allclose(tensors[0], torch.zeros(10))
+    # apply=True installs it (what TrainingStateAverager does by default)
+    result = target.load_state_from_peers(timeout=20, apply=True)
     assert result is not None
     with target.get_tensors() as tensors:
         assert torch.allclose(tensors[0], torch.full((10,), 3.14))
@@ -196,6 +202,41 @@ def test_overcrowded():
     results = [f.result(90) for f in futures]
     n_grouped = sum(r is not None and len(r) >= 2 for r in results)
     assert n_grouped >= n - 1, f"only {n_grouped}/{n} peers averaged: {results}"
+    for avg in averagers:
+        avg.shutdown()
+    for d in dhts:
+        d.shutdown()
+
+
+def test_load_state_priority():
+    """Donors advertise a priority under {prefix}.all_averagers; downloads go to
+    the highest-priority peer that allows sharing (reference
+    test_averaging.py:416 test_load_state_priority)."""
+    dhts = make_dht_swarm(4)
+    averagers = []
+    for i in range(4):
+        avg = DecentralizedAverager(
+            [torch.randn(3), torch.tensor([float(i)])], dhts[i], start=True,
+            prefix="prio", target_group_size=2, min_matchmaking_time=1.0,
+            request_timeout=0.5, declare_state_period=0.5,
+            allow_state_sharing=(i != 1),  # peer 1 never shares
+        )
+        avg.state_sharing_priority = 5 - abs(2 - i)  # peer 2 highest (5), then 1 and 3 (4), then 0 (3)
+        averagers.append(avg)
+    time.sleep(1.2)  # let declare_state run with the assigned priorities
+
+    # peer 0 downloads: best donor is peer 2 (priority 5)
+    result = averagers[0].load_state_from_peers(timeout=20)
+    assert result is not None and result[1][-1].item() == 2.0
+    # peer 2 downloads: peer 1 (priority 4) does not share, so peer 3 (also 4)
+    # or peer 0 (3) serves; never the non-sharing peer 1
+    result = averagers[2].load_state_from_peers(timeout=20)
+    assert result is not None and result[1][-1].item() in (0.0, 3.0)
+
+    averagers[0].state_sharing_priority = 10
+    time.sleep(1.2)
+    result = averagers[2].load_state_from_peers(timeout=20)
+    assert result is not None and result[1][-1].item() == 0.0
     for avg in averagers:
         avg.shutdown()
     for d in dhts:
