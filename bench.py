@@ -65,6 +65,8 @@ def main():
         try:
             import torch.cuda.tunable as tunable
 
+            if os.environ.get("PYTORCH_TUNABLEOP_ENABLED", "1") == "0":
+                raise RuntimeError("disabled via PYTORCH_TUNABLEOP_ENABLED=0")
             tunable.enable(True)
             tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
                                  "profiles", "tunableop_albert_b128.csv")
