@@ -153,6 +153,11 @@ class _RemoteCallMany(torch.autograd.Function):
     ):
         num_samples, max_experts = len(experts_per_sample), max(len(e) for e in experts_per_sample)
 
+        if detect_anomalies:
+            for t in flat_inputs:
+                if not torch.isfinite(t).all():
+                    raise ValueError("One of the inputs has nan/inf values")
+
         async def _forward_all():
             tasks = {}
             for i, experts in enumerate(experts_per_sample):
@@ -163,6 +168,14 @@ class _RemoteCallMany(torch.autograd.Function):
             return await _collect_responses(tasks, num_samples, k_min, timeout_after_k_min, forward_timeout)
 
         results = RemoteExpertWorker.run_coroutine(_forward_all(), loop=loop)
+
+        if detect_anomalies:
+            # treat experts that returned non-finite outputs as dead
+            bad = [pair for pair, outs in results.items()
+                   if any(not torch.isfinite(o).all() for o in outs)]
+            for pair in bad:
+                logger.warning(f"dropping expert response {experts_per_sample[pair[0]][pair[1]].uid}: nonfinite output")
+                del results[pair]
 
         alive_counts = [0] * num_samples
         for (i, _j) in results:
@@ -202,6 +215,7 @@ class _RemoteCallMany(torch.autograd.Function):
             info=info,
             loop=loop,
             num_samples=num_samples,
+            detect_anomalies=detect_anomalies,
         )
         return (mask, *(t.requires_grad_(True) for t in stacked_outputs))
 
@@ -227,9 +241,16 @@ class _RemoteCallMany(torch.autograd.Function):
                 tasks, num_samples, saved["backward_k_min"], saved["timeout_after_k_min"], saved["backward_timeout"]
             )
 
+        if saved["detect_anomalies"]:
+            for g in grad_outputs_stacked:
+                if not torch.isfinite(g).all():
+                    raise ValueError("One of the gradients has nan/inf values")
+
         results = RemoteExpertWorker.run_coroutine(_backward_all(), loop=loop)
         grad_inputs = [torch.zeros_like(t) for t in flat_inputs]
         for (i, _j), grads in results.items():
+            if saved["detect_anomalies"] and any(not torch.isfinite(g).all() for g in grads):
+                continue  # dead reducer: exclude nonfinite expert gradients
             for gi, g in zip(grad_inputs, grads[: len(grad_inputs)]):
                 gi[i : i + 1] += g.to(gi.dtype)
         return (DUMMY, None, None, None, None, None, None, None, None, None, None, *grad_inputs)

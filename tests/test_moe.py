@@ -176,3 +176,49 @@ def test_moe_tolerates_dead_experts():
     finally:
         server.shutdown()
         dht.shutdown()
+
+
+def test_client_anomaly_detection():
+    """detect_anomalies: nan/inf inputs raise; experts that return non-finite
+    outputs are treated as dead (reference test_moe.py client anomaly test)."""
+    from hivemind_amd.moe import register_expert_class
+
+    @register_expert_class("nan_expert_test", lambda batch, hid: torch.empty((batch, hid)))
+    class NaNExpert(nn.Module):
+        def __init__(self, hid_dim):
+            super().__init__()
+            self.lin = nn.Linear(hid_dim, hid_dim)
+
+        def forward(self, x):
+            return self.lin(x) * float("nan")
+
+    torch.manual_seed(0)
+    dht = DHT(start=True)
+    good = Server.create(
+        dht=dht, expert_uids=["anom.0.0", "anom.1.0"], expert_cls="ffn", hidden_dim=16,
+        optim_cls=None, device="cpu", start=True,
+    )
+    bad = Server.create(
+        dht=DHT(initial_peers=[dht.endpoint], start=True),
+        expert_uids=["anom.2.0", "anom.3.0"], expert_cls="nan_expert_test", hidden_dim=16,
+        optim_cls=None, device="cpu", start=True,
+    )
+    try:
+        moe = RemoteMixtureOfExperts(
+            in_features=16, grid_size=(4, 1), dht=dht, uid_prefix="anom", k_best=4, k_min=1,
+            forward_timeout=15, backward_timeout=15, detect_anomalies=True,
+        )
+        # nan input is rejected outright
+        bad_input = torch.randn(2, 16)
+        bad_input[0, 0] = float("nan")
+        with pytest.raises(ValueError, match="nan/inf"):
+            moe(bad_input)
+        # nan-producing experts are dropped; the mixture output stays finite
+        out = moe(torch.randn(4, 16))
+        assert torch.isfinite(out).all()
+        out.sum().backward()
+        assert torch.isfinite(moe.proj.weight.grad).all()
+    finally:
+        good.shutdown()
+        bad.shutdown()
+        dht.shutdown()
