@@ -231,7 +231,10 @@ class DHTNode:
         queries = list(queries)
         k_nearest = k_nearest if k_nearest is not None else self.protocol.bucket_size
         beam_size = beam_size if beam_size is not None else max(self.beam_size, k_nearest)
-        num_workers = num_workers if num_workers is not None else self.num_workers
+        # scale worker count with the number of packed queries: a batched
+        # store (declare_experts stores uid + every grid prefix in one call)
+        # can carry ~100 queries, and 4 workers serialize it needlessly
+        num_workers = num_workers if num_workers is not None else min(max(self.num_workers, len(queries) // 2), 32)
         node_to_peer = dict(node_to_peer or {})
 
         initial: List[DHTID] = []
@@ -308,48 +311,61 @@ class DHTNode:
             list(dict.fromkeys(key_ids)), k_nearest=self.num_replicas, exclude_self=exclude_self
         )
 
-        async def store_on_peers(key_id: DHTID, nearest: Dict[DHTID, PeerInfo]):
-            composites = [c for c in binary_values if c[0] == key_id]
-            in_my_range = not exclude_self and self.node_id in nearest
-            tasks = []
+        # group work by DESTINATION PEER: one bulk call_store per peer carrying
+        # every (key, subkey) it should replicate, instead of one RPC per
+        # (key, peer) pair (a 32-uid declare_experts produced ~280 store RPCs
+        # that way; this sends at most one per peer in the replica union)
+        composites_by_key: Dict[DHTID, List[Tuple[DHTID, Any]]] = {}
+        for c in binary_values:
+            composites_by_key.setdefault(c[0], []).append(c)
+        per_peer: Dict[DHTID, Tuple[PeerInfo, List[Tuple[DHTID, Any]]]] = {}
+        local_composites: List[Tuple[DHTID, Any]] = []
+        for key_id, nearest in nearest_per_query.items():
             for nid, peer in nearest.items():
                 if nid == self.node_id:
-                    # store locally: sign + validate exactly as a remote store would
-                    # (reference node.py stores through the same record validators)
-                    from .validation import DHTRecord
+                    local_composites.extend(composites_by_key.get(key_id, ()))
+                else:
+                    entry = per_peer.setdefault(nid, (peer, []))
+                    entry[1].extend(composites_by_key.get(key_id, ()))
 
-                    for c in composites:
-                        _, subkey = c
-                        value = binary_values[c]
-                        if isinstance(value, bytes) and self.protocol.record_validator is not None:
-                            packed_subkey = (
-                                MSGPackSerializer.dumps(subkey) if subkey is not None else b""
-                            )
-                            record = DHTRecord(key_id.to_bytes(), packed_subkey, value, expirations[c])
-                            signed = self.protocol.record_validator.sign_value(record)
-                            record = DHTRecord(key_id.to_bytes(), packed_subkey, signed, expirations[c])
-                            if not self.protocol.record_validator.validate(record):
-                                continue  # forged update of a protected key
-                            value = self.protocol.record_validator.strip_value(record)
-                        if subkey is None:
-                            ok = self.protocol.storage.store(key_id, value, expirations[c])
-                        else:
-                            ok = self.protocol.storage.store_subkey(key_id, subkey, value, expirations[c])
-                        store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or ok
-                    continue
-                ks = [key_id] * len(composites)
-                vs = [binary_values[c] for c in composites]
-                exps = [expirations[c] for c in composites]
-                subs = [c[1] for c in composites]
-                tasks.append(self.protocol.call_store(peer, ks, vs, exps, subkeys=subs))
-            responses = await asyncio.gather(*tasks, return_exceptions=True)
-            for response in responses:
-                if isinstance(response, Exception) or response is None:
-                    continue
-                for c, ok in zip(composites, response):
-                    store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or bool(ok)
+        # store locally: sign + validate exactly as a remote store would
+        # (reference node.py stores through the same record validators)
+        from .validation import DHTRecord
 
-        await asyncio.gather(*(store_on_peers(key_id, nearest) for key_id, nearest in nearest_per_query.items()))
+        for c in local_composites:
+            key_id, subkey = c
+            value = binary_values[c]
+            if isinstance(value, bytes) and self.protocol.record_validator is not None:
+                packed_subkey = MSGPackSerializer.dumps(subkey) if subkey is not None else b""
+                record = DHTRecord(key_id.to_bytes(), packed_subkey, value, expirations[c])
+                signed = self.protocol.record_validator.sign_value(record)
+                record = DHTRecord(key_id.to_bytes(), packed_subkey, signed, expirations[c])
+                if not self.protocol.record_validator.validate(record):
+                    continue  # forged update of a protected key
+                value = self.protocol.record_validator.strip_value(record)
+            if subkey is None:
+                ok = self.protocol.storage.store(key_id, value, expirations[c])
+            else:
+                ok = self.protocol.storage.store_subkey(key_id, subkey, value, expirations[c])
+            store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or ok
+
+        async def store_on_peer(peer: PeerInfo, composites: List[Tuple[DHTID, Any]]):
+            ks = [c[0] for c in composites]
+            vs = [binary_values[c] for c in composites]
+            exps = [expirations[c] for c in composites]
+            subs = [c[1] for c in composites]
+            try:
+                response = await self.protocol.call_store(peer, ks, vs, exps, subkeys=subs)
+            except Exception:
+                return
+            if response is None:
+                return
+            for c, ok in zip(composites, response):
+                store_ok[id_to_original[c]] = store_ok[id_to_original[c]] or bool(ok)
+
+        await asyncio.gather(
+            *(store_on_peer(peer, comps) for peer, comps in per_peer.values()), return_exceptions=True
+        )
 
         if self.cache_on_store:
             for c, value in binary_values.items():

@@ -95,6 +95,10 @@ async def traverse_dht(
             visited[q].add(uid)
 
     in_flight: Dict[DHTID, int] = {q: 0 for q in queries}  # RPCs that may still add candidates
+    queried: Dict[DHTID, Set[DHTID]] = {q: set() for q in queries}  # peers this query was sent to
+    # (visited = "known, don't re-add as a candidate"; queried = "an RPC already
+    # asked this peer about this query" -- initial_nodes are visited but NOT
+    # queried, so packing can still batch them)
 
     def upper_bound(q: DHTID) -> int:
         if len(nearest[q]) < beam_size:
@@ -125,7 +129,9 @@ async def traverse_dht(
                 for q in queries:
                     if q in finished:
                         continue
-                    while candidates[q] and candidates[q][ROOT][0] > upper_bound(q):
+                    while candidates[q] and (
+                        candidates[q][ROOT][0] > upper_bound(q) or candidates[q][ROOT][1] in queried[q]
+                    ):
                         heapq.heappop(candidates[q])
                     maybe_finish(q)
                     if q in finished:
@@ -143,18 +149,23 @@ async def traverse_dht(
                 else:
                     _, main_query, peer = best
                     heapq.heappop(candidates[main_query])
-                    # pack additional active queries that would also like this peer;
-                    # the peer counts as visited for every packed query so no other
-                    # worker re-queries it on their behalf
+                    # pack additional active queries that have NOT queried this
+                    # peer yet; the peer counts as visited for every packed query
+                    # so no other worker re-queries it on their behalf. Packing
+                    # queries that already visited the peer is pure waste, and
+                    # skipping unvisited ones forces them to re-query the peer
+                    # as main queries later (measured 350 -> ~40 find RPCs for a
+                    # 65-query declare in an 8-peer swarm).
                     packed = [main_query]
                     for q in queries:
                         if len(packed) >= queries_per_call:
                             break
-                        if q is main_query or q in finished:
+                        if q is main_query or q in finished or peer in queried[q]:
                             continue
                         packed.append(q)
                     for q in packed:
                         visited[q].add(peer)
+                        queried[q].add(peer)
                         in_flight[q] += 1
                         # lazily drop the peer from q's own candidate heap
                         candidates[q] = [(d, uid) for d, uid in candidates[q] if uid != peer]
