@@ -161,3 +161,42 @@ def test_key_stored_before_peer_joins_is_reachable():
         await alice.shutdown()
 
     run(main())
+
+
+def test_batched_store_packs_rpcs():
+    """A batched declare (32 uids -> ~65 keys incl. grid prefixes) must pack
+    queries into few find RPCs and bulk per-peer store RPCs -- regression test
+    for the traverse packing filter (was ~350 find + ~280 store RPCs, now ~35+7)."""
+    import random
+
+    import hivemind_amd.dht.node as node_mod
+    from hivemind_amd.dht.protocol import DHTProtocol
+    from hivemind_amd.moe.server.dht_handler import declare_experts
+
+    counts = {"find": 0, "store": 0}
+    orig_find, orig_store = DHTProtocol.call_find, DHTProtocol.call_store
+
+    async def counted_find(self, *a, **kw):
+        counts["find"] += 1
+        return await orig_find(self, *a, **kw)
+
+    async def counted_store(self, *a, **kw):
+        counts["store"] += 1
+        return await orig_store(self, *a, **kw)
+
+    DHTProtocol.call_find, DHTProtocol.call_store = counted_find, counted_store
+    try:
+        root = DHT(start=True)
+        dhts = [root] + [DHT(initial_peers=[root.endpoint], start=True) for _ in range(7)]
+        time.sleep(0.5)
+        counts.update(find=0, store=0)
+        uids = [f"expert.{random.randint(0, 9999)}.{random.randint(0, 255)}" for _ in range(32)]
+        result = declare_experts(dhts[3], uids, get_dht_time() + 120)
+        assert all(result.values())
+        # ideal: ~5 packed find RPCs per peer + <= 1 bulk store per peer
+        assert counts["find"] <= 120, f"traverse packing regressed: {counts['find']} find RPCs"
+        assert counts["store"] <= 24, f"per-peer store batching regressed: {counts['store']} store RPCs"
+    finally:
+        DHTProtocol.call_find, DHTProtocol.call_store = orig_find, orig_store
+        for d in locals().get("dhts", []):
+            d.shutdown()
