@@ -80,6 +80,8 @@ def main():
     parser.add_argument("--batches-per-client", type=int, default=None)
     parser.add_argument("--batch-size", type=int, default=None)
     parser.add_argument("--max-batch-size", type=int, default=None)
+    parser.add_argument("--num-handlers", type=int, default=4,
+                        help="balanced connection-handler loops on the server")
     args = parser.parse_args()
 
     presets = {
@@ -108,6 +110,7 @@ def main():
         dht=dht, expert_uids=uids, expert_cls="ffn", hidden_dim=cfg["hidden_dim"],
         optim_cls=(torch.optim.Adam if cfg["backprop"] else None),
         max_batch_size=cfg["max_batch_size"], device=device, start=True,
+        num_connection_handlers=args.num_handlers,
     )
     t_server_ready = time.perf_counter()
 
