@@ -80,8 +80,8 @@ def main():
     parser.add_argument("--batches-per-client", type=int, default=None)
     parser.add_argument("--batch-size", type=int, default=None)
     parser.add_argument("--max-batch-size", type=int, default=None)
-    parser.add_argument("--num-handlers", type=int, default=4,
-                        help="balanced connection-handler loops on the server")
+    parser.add_argument("--num-handlers", type=int, default=8,
+                        help="balanced connection-handler loops on the server (8 measured best on a 256-core MI355X node)")
     args = parser.parse_args()
 
     presets = {
