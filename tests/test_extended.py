@@ -304,3 +304,40 @@ def test_optimizer_delayed_updates_mode():
     assert all(r is not None and r[0] >= 2 and r[1] for r in results), results
     for d in dhts:
         d.shutdown()
+
+
+def test_dht_schema_validator():
+    """SchemaValidator enforces pydantic-typed DHT records
+    (reference test_dht_schema.py)."""
+    from typing import Dict
+
+    import pydantic
+
+    from hivemind_amd.dht.schema import SchemaValidator
+
+    class TypedSchema(pydantic.BaseModel):
+        experiment_name: bytes
+        n_batches: Dict[bytes, int]
+
+    dht1 = DHT(start=True, record_validators=[SchemaValidator(TypedSchema, allow_extra_keys=False)])
+    dht2 = DHT(initial_peers=[dht1.endpoint], start=True,
+               record_validators=[SchemaValidator(TypedSchema, allow_extra_keys=False)])
+    from hivemind_amd.utils.timed_storage import get_dht_time
+
+    now = get_dht_time()
+    # valid records
+    assert dht1.store("experiment_name", b"exp1", now + 30)
+    assert dht2.store("n_batches", 777, now + 30, subkey=b"peerA")
+    time.sleep(0.3)
+    got = dht2.get("experiment_name", latest=True)
+    assert got is not None and got.value == b"exp1"
+    # wrong value type is rejected
+    assert not dht1.store("experiment_name", 12345, now + 60)
+    assert not dht2.store("n_batches", "not an int", now + 60, subkey=b"peerB")
+    # unknown keys are rejected when extra keys are disallowed
+    assert not dht1.store("unknown_key", 1, now + 30)
+    # the valid record survived the rejected overwrite
+    got = dht2.get("experiment_name", latest=True)
+    assert got is not None and got.value == b"exp1"
+    dht2.shutdown()
+    dht1.shutdown()
