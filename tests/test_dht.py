@@ -200,3 +200,34 @@ def test_batched_store_packs_rpcs():
         DHTProtocol.call_find, DHTProtocol.call_store = orig_find, orig_store
         for d in locals().get("dhts", []):
             d.shutdown()
+
+
+def test_routing_table_search_exact():
+    """get_nearest_neighbors must agree EXACTLY with brute-force XOR ranking
+    over the table's active nodes (reference test_routing.py:82)."""
+    import heapq
+    import random as _random
+    from itertools import chain, zip_longest
+
+    from hivemind_amd.p2p import PeerID
+
+    _random.seed(7)
+    node_id = DHTID.generate()
+    table = RoutingTable(node_id, bucket_size=20, depth_modulo=5)
+    for port in _random.sample(range(1_000_000), 2000):
+        table.add_or_update_node(DHTID.generate(), PeerID(port.to_bytes(4, "big")), f"127.0.0.1:{port % 65535}")
+
+    active = list(chain(*(bucket.nodes_to_peers.keys() for bucket in table.buckets)))
+    assert 100 <= len(active) <= 2000
+
+    for _ in range(200):
+        k = _random.randint(1, 50)
+        query = DHTID.generate()
+        ours = [uid for uid, _info in table.get_nearest_neighbors(query, k=k)]
+        reference = heapq.nsmallest(k, active, key=query.xor_distance)
+        assert all(a == b for a, b in zip_longest(ours, reference)), (k, query)
+
+    # exclusion: the excluded node never appears
+    victim = active[0]
+    ours = [uid for uid, _info in table.get_nearest_neighbors(victim, k=20, exclude=victim)]
+    assert victim not in ours and len(ours) == 20
