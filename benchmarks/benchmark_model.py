@@ -33,6 +33,8 @@ def main():
     config = {"albert-base": AlbertConfig.base, "albert-large": AlbertConfig.large, "tiny": AlbertConfig.tiny}[args.model]()
     if device.type == "cpu":
         config.dtype = torch.float32
+    if args.seq_len > config.max_position_embeddings:
+        args.seq_len = config.max_position_embeddings
     model = AlbertForMaskedLM(config).to(device)
 
     opt = None
