@@ -75,7 +75,11 @@ def main():
                     tunable.read_file(tuned)
                 except Exception as e:
                     log(f"tunableop table not loaded: {e}")
-            tunable.tuning_enable(True)
+            # tuning unknown shapes from scratch can add minutes of warmup
+            # (observed: llama-1b blew a 200 s window); shapes missing from the
+            # table just use hipBLASLt's default algorithm unless the caller
+            # explicitly opts in to tuning
+            tunable.tuning_enable(os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") == "1")
             tunable.write_file_on_exit(False)
         except Exception as e:
             log(f"TunableOp unavailable: {e}")
