@@ -205,7 +205,8 @@ def main():
     if rank == 0:
         result = {
             "metric": ("samples/sec (whole swarm) ALBERT-base hivemind.Optimizer" if not is_llama
-                       else f"samples/sec (whole swarm) {args.model} hivemind.Optimizer + DPU"),
+                       else f"samples/sec (whole swarm) {args.model} hivemind.Optimizer"
+                       + (" + DPU" if args.dpu else "")),
             "value": round(samples_per_sec, 2),
             "unit": "samples/s",
             "n_gpus": n_gpus,
