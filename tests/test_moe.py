@@ -222,3 +222,36 @@ def test_client_anomaly_detection():
         good.shutdown()
         bad.shutdown()
         dht.shutdown()
+
+
+def test_switch_moe_training():
+    """Switch-MoE (top-1 routing + load-balancing loss) learns a separable task
+    (reference test_training.py:112 shape)."""
+    torch.manual_seed(0)
+    dht = DHT(start=True)
+    server = Server.create(
+        dht=dht, expert_uids=[f"swtr.{i}.0" for i in range(4)], expert_cls="ffn", hidden_dim=16,
+        optim_cls=lambda p: torch.optim.SGD(p, lr=0.05), device="cpu", start=True,
+    )
+    try:
+        moe = RemoteSwitchMixtureOfExperts(
+            in_features=16, grid_size=(4, 1), dht=dht, uid_prefix="swtr",
+            forward_timeout=15, backward_timeout=15, jitter_eps=0.0,
+        )
+        head = nn.Linear(16, 2)
+        opt = torch.optim.SGD(list(moe.proj.parameters()) + list(head.parameters()), lr=0.05)
+        X = torch.randn(64, 16)
+        y = (X[:, 0] > 0).long()
+        initial_loss = None
+        for step in range(12):
+            out, balancing_loss = moe(X)
+            loss = torch.nn.functional.cross_entropy(head(out), y) + 0.01 * balancing_loss
+            if initial_loss is None:
+                initial_loss = loss.item()
+            opt.zero_grad()
+            loss.backward()
+            opt.step()
+        assert loss.item() < initial_loss, (initial_loss, loss.item())
+    finally:
+        server.shutdown()
+        dht.shutdown()
