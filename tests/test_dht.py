@@ -231,3 +231,31 @@ def test_routing_table_search_exact():
     victim = active[0]
     ours = [uid for uid, _info in table.get_nearest_neighbors(victim, k=20, exclude=victim)]
     assert victim not in ours and len(ours) == 20
+
+
+def test_dht_survives_peer_failures():
+    """Values stay retrievable after a minority of peers die abruptly
+    (reference test_dht_node fault tolerance: blacklist + num_replicas)."""
+    root = DHT(start=True)
+    dhts = [root] + [DHT(initial_peers=[root.endpoint], start=True) for _ in range(7)]
+    time.sleep(0.5)
+    now = get_dht_time()
+    for i in range(8):
+        assert dhts[i % 8].store(f"survive{i}", i, now + 120)
+    # kill 3 of 8 peers without any goodbye
+    for d in dhts[5:]:
+        d.shutdown()
+    time.sleep(0.3)
+    found = 0
+    for i in range(8):
+        res = dhts[i % 5].get(f"survive{i}", latest=False)
+        found += res is not None and res.value == i
+    # replication factor covers a 3/8 failure: most keys must survive
+    assert found >= 6, f"only {found}/8 keys survived peer failures"
+    # the surviving swarm still accepts new writes and reads them back
+    assert dhts[0].store("after_failure", 42, now + 60)
+    time.sleep(0.2)
+    res = dhts[3].get("after_failure", latest=True)
+    assert res is not None and res.value == 42
+    for d in dhts[:5]:
+        d.shutdown()
