@@ -134,6 +134,7 @@ class DHTNode:
         listen_host: str = "127.0.0.1",
         port: int = 0,
         client_mode: bool = False,
+        relay_endpoint: Optional[str] = None,
         record_validator: Optional[RecordValidatorBase] = None,
         blacklist_time: float = 5.0,
         backoff_rate: float = 2.0,
@@ -146,7 +147,11 @@ class DHTNode:
         self.reuse_get_requests = reuse_get_requests
         self.blacklist = Blacklist(blacklist_time, backoff_rate)
         if p2p is None:
-            p2p = await P2P.create(listen_host=listen_host, port=port, listen=not client_mode)
+            # a NATed node (client_mode + relay_endpoint) registers with a public
+            # relay peer and advertises a relay:// endpoint instead of listening
+            p2p = await P2P.create(
+                listen_host=listen_host, port=port, listen=not client_mode, relay_endpoint=relay_endpoint
+            )
             self._should_shutdown_p2p = True
         self.p2p = p2p
         self.protocol = await DHTProtocol.create(

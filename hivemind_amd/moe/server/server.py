@@ -46,11 +46,13 @@ class Server(threading.Thread):
         expiration: Optional[float] = None,
         checkpoint_dir: Optional[Path] = None,
         stats_report_interval: Optional[float] = None,
+        relay_endpoint: Optional[str] = None,
         start: bool = False,
     ):
         super().__init__(name="moe-server", daemon=True)
         self.dht, self.module_backends = dht, module_backends
         self.device = device
+        self.relay_endpoint = relay_endpoint
         self.num_connection_handlers = max(1, num_connection_handlers)
         self.conn_handler = ConnectionHandler(module_backends)
         self.runtime = Runtime(module_backends, device=device, stats_report_interval=stats_report_interval)
@@ -85,7 +87,12 @@ class Server(threading.Thread):
         for i in range(1, self.num_connection_handlers):
             loop_thread = EventLoopThread(name=f"moe-handler-{i}")
             loop_thread.start_and_wait()
-            p2p_i = asyncio.run_coroutine_threadsafe(P2P.create(), loop_thread.loop).result(15)
+            # behind NAT (relay_endpoint set): no inbound socket; each handler
+            # loop registers with the relay and advertises a relay:// endpoint
+            p2p_i = asyncio.run_coroutine_threadsafe(
+                P2P.create(listen=self.relay_endpoint is None, relay_endpoint=self.relay_endpoint),
+                loop_thread.loop,
+            ).result(15)
             asyncio.run_coroutine_threadsafe(self.conn_handler.add_handlers(p2p_i), loop_thread.loop).result(15)
             self._handler_loops.append(loop_thread)
             self._handler_p2ps.append(p2p_i)
@@ -154,6 +161,7 @@ class Server(threading.Thread):
         stats_report_interval: Optional[float] = None,
         update_period: float = 30.0,
         expiration: Optional[float] = None,
+        relay_endpoint: Optional[str] = None,
         start: bool = False,
         **kwargs,
     ) -> "Server":
@@ -207,6 +215,7 @@ class Server(threading.Thread):
             expiration=expiration,
             checkpoint_dir=checkpoint_dir,
             stats_report_interval=stats_report_interval,
+            relay_endpoint=relay_endpoint,
             start=start,
         )
 

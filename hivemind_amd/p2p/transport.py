@@ -45,6 +45,19 @@ _T_STREAM_END = 6  # callee finished (stream done)
 _T_ERROR = 7
 _T_CANCEL = 8
 _T_PING = 9
+# circuit relay (the reference's libp2p relay/NAT-traversal, re-expressed for
+# this transport): a NATed peer keeps an OUTBOUND connection to a public relay
+# peer and registers; a third party dials the relay, which asks the NATed peer
+# to dial back a fresh socket and then splices the two sockets byte-for-byte.
+# After the splice both ends run the ordinary HELLO handshake -- the RPC layer
+# never knows the connection is relayed.
+_T_RELAY_CONNECT = 10  # caller -> relay: payload = target peer id
+_T_RELAY_OK = 11  # relay -> caller: target dialed back, pipe is live
+_T_RELAY_OPEN = 12  # relay -> target (over its registered conn): call_id = channel
+_T_RELAY_ACCEPT = 13  # target -> relay (on a fresh socket): call_id = channel
+
+RELAY_SCHEME = "relay://"
+RELAY_REGISTER_HANDLER = "__relay__.register"
 
 MAX_FRAME_SIZE = 256 * 1024 * 1024  # control plane sanity bound
 STREAM_BUFFER_LIMIT = 64 * 1024 * 1024  # asyncio StreamReader high-water mark:
@@ -157,6 +170,11 @@ class P2P:
         self._endpoint_book: Dict[PeerID, str] = {}  # last known endpoint per peer
         self._alive = True
         self._listen = True
+        # relay state: as a RELAY -- which peers registered + sockets waiting to
+        # be spliced; as a NATED PEER -- the relay we are reachable through
+        self._relay_served: Dict[PeerID, float] = {}
+        self._relay_waiting: Dict[bytes, asyncio.Future] = {}
+        self._relay_endpoint: Optional[str] = None
 
     # ------------------------------------------------------------------ setup
 
@@ -167,6 +185,7 @@ class P2P:
         port: int = 0,
         identity: Optional[PrivateKey] = None,
         listen: bool = True,
+        relay_endpoint: Optional[str] = None,
     ) -> "P2P":
         self = cls()
         self._identity = identity if identity is not None else PrivateKey()
@@ -176,19 +195,122 @@ class P2P:
         if listen:
             self._server = await asyncio.start_server(self._on_accept, listen_host, port, limit=STREAM_BUFFER_LIMIT)
             self._port = self._server.sockets[0].getsockname()[1]
+            # any listening peer can serve as a circuit relay for NATed peers
+            self.add_unary_handler(RELAY_REGISTER_HANDLER, self._rpc_relay_register)
+        if relay_endpoint is not None:
+            await self.register_with_relay(relay_endpoint)
         return self
 
     @property
     def endpoint(self) -> str:
+        if not self._listen and self._relay_endpoint is not None:
+            return f"{RELAY_SCHEME}{self._relay_endpoint}/{self.peer_id.to_base58()}"
         return make_endpoint(self._listen_host, self._port)
 
     @property
     def peer_info(self) -> PeerInfo:
-        return PeerInfo(self.peer_id, (self.endpoint,) if self._listen else ())
+        if self._listen or self._relay_endpoint is not None:
+            return PeerInfo(self.peer_id, (self.endpoint,))
+        return PeerInfo(self.peer_id, ())
 
     def get_visible_maddrs(self) -> List[str]:
-        """API-compat: list of announce addresses (host:port strings)."""
-        return [self.endpoint] if self._listen else []
+        """API-compat: list of announce addresses (host:port or relay:// strings)."""
+        if self._listen or self._relay_endpoint is not None:
+            return [self.endpoint]
+        return []
+
+    # ----------------------------------------------------------------- relay
+
+    async def register_with_relay(self, relay_endpoint: str) -> None:
+        """NATed side: keep an outbound connection to a public relay peer and
+        become reachable at ``relay://<relay_endpoint>/<our peer id>``."""
+        info = await self.connect_endpoint(relay_endpoint)
+        await self.call_unary(info.peer_id, RELAY_REGISTER_HANDLER, b"")
+        self._relay_endpoint = relay_endpoint
+
+    async def _rpc_relay_register(self, _payload: bytes, ctx: "RpcContext") -> bytes:
+        import time as _time
+
+        self._relay_served[ctx.remote_id] = _time.monotonic()
+        logger.debug(f"serving as relay for {ctx.remote_id}")
+        return b""
+
+    async def _relay_dial_back(self, channel: bytes):
+        """NATed side: the relay asked us to open the second leg of a circuit."""
+        try:
+            host, port = split_endpoint(self._relay_endpoint)
+            reader, writer = await asyncio.wait_for(
+                asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10
+            )
+            conn = _Connection(self, reader, writer)
+            await conn.recv_frame()  # absorb the relay's eager HELLO
+            await conn.send_frame(_T_RELAY_ACCEPT, channel)
+            # from here the socket is spliced to the caller: ordinary handshake
+            await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+            ftype, _, _, payload = await conn.recv_frame()
+            if ftype != _T_HELLO:
+                raise P2PDaemonError("relayed caller did not say HELLO")
+            conn.remote_id = PeerID(payload)
+            self._register_connection(conn)
+            conn.reader_task = asyncio.create_task(self._connection_loop(conn))
+        except Exception as e:
+            logger.warning(f"relay dial-back failed: {e!r}")
+
+    async def _dial_via_relay(self, relay_url: str) -> _Connection:
+        """Caller side: connect to ``relay://<host:port>/<target b58>``."""
+        rest = relay_url[len(RELAY_SCHEME):]
+        relay_ep, _, target_b58 = rest.rpartition("/")
+        if not relay_ep or not target_b58:
+            raise P2PDaemonError(f"malformed relay endpoint: {relay_url}")
+        target = PeerID.from_base58(target_b58)
+        host, port = split_endpoint(relay_ep)
+        reader, writer = await asyncio.wait_for(
+            asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10
+        )
+        conn = _Connection(self, reader, writer)
+        try:
+            await conn.recv_frame()  # absorb the relay's eager HELLO
+            await conn.send_frame(_T_RELAY_CONNECT, b"", "", target.to_bytes())
+            ftype, _, _, payload = await conn.recv_frame()
+            if ftype == _T_ERROR:
+                raise P2PDaemonError(payload.decode(errors="replace"))
+            if ftype != _T_RELAY_OK:
+                raise P2PDaemonError(f"unexpected relay response type {ftype}")
+            # circuit is live: ordinary handshake with the target through the pipe
+            await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+            ftype, _, _, payload = await conn.recv_frame()
+            if ftype != _T_HELLO:
+                raise P2PDaemonError("relayed target did not say HELLO")
+            conn.remote_id = PeerID(payload)
+            return conn
+        except Exception:
+            try:
+                writer.close()
+            except Exception:
+                pass
+            raise
+
+    async def _relay_splice(self, conn_a: _Connection, conn_b: _Connection):
+        """Relay side: forward raw bytes between the two circuit legs."""
+
+        async def pump(src: _Connection, dst: _Connection):
+            try:
+                while True:
+                    chunk = await src.reader.read(STREAM_CHUNK_SIZE)
+                    if not chunk:
+                        break
+                    dst.writer.write(chunk)
+                    await dst.writer.drain()
+            except Exception:
+                pass
+            finally:
+                for c in (conn_a, conn_b):
+                    try:
+                        c.writer.close()
+                    except Exception:
+                        pass
+
+        await asyncio.gather(pump(conn_a, conn_b), pump(conn_b, conn_a))
 
     # -------------------------------------------------------------- handlers
 
@@ -228,7 +350,17 @@ class P2P:
         conn = _Connection(self, reader, writer)
         try:
             await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-            ftype, _, _, payload = await conn.recv_frame()
+            ftype, call_id, _, payload = await conn.recv_frame()
+            if ftype == _T_RELAY_CONNECT:
+                await self._serve_relay_connect(conn, PeerID(payload))
+                return
+            if ftype == _T_RELAY_ACCEPT:
+                fut = self._relay_waiting.pop(call_id, None)
+                if fut is not None and not fut.done():
+                    fut.set_result(conn)  # the splice task owns the socket now
+                else:
+                    await conn.close()
+                return
             if ftype != _T_HELLO:
                 raise P2PDaemonError("expected HELLO")
             conn.remote_id = PeerID(payload)
@@ -237,6 +369,31 @@ class P2P:
             return
         self._register_connection(conn)
         conn.reader_task = asyncio.create_task(self._connection_loop(conn))
+
+    async def _serve_relay_connect(self, caller_conn: _Connection, target: PeerID):
+        """Relay side: a caller asked to be piped to a registered NATed peer."""
+        target_conn = self._connections.get(target)
+        if target not in self._relay_served or target_conn is None or target_conn.closed.is_set():
+            try:
+                await caller_conn.send_frame(_T_ERROR, b"", "", f"no registered relay target {target}".encode())
+            finally:
+                await caller_conn.close()
+            return
+        channel = os.urandom(8)
+        fut: asyncio.Future = asyncio.get_event_loop().create_future()
+        self._relay_waiting[channel] = fut
+        try:
+            await target_conn.send_frame(_T_RELAY_OPEN, channel)
+            accept_conn: _Connection = await asyncio.wait_for(fut, timeout=15)
+        except Exception as e:
+            self._relay_waiting.pop(channel, None)
+            try:
+                await caller_conn.send_frame(_T_ERROR, b"", "", f"relay target did not dial back: {e!r}".encode())
+            finally:
+                await caller_conn.close()
+            return
+        await caller_conn.send_frame(_T_RELAY_OK, b"")
+        await self._relay_splice(caller_conn, accept_conn)
 
     async def connect_endpoint(self, endpoint: str) -> PeerInfo:
         """Dial a bare endpoint and learn the peer's identity (bootstrap helper)."""
@@ -281,14 +438,17 @@ class P2P:
                 return conn
             for ep in endpoints:
                 try:
-                    host, port = split_endpoint(ep)
-                    reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
-                    conn = _Connection(self, reader, writer)
-                    await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-                    ftype, _, _, payload = await conn.recv_frame()
-                    if ftype != _T_HELLO:
-                        raise P2PDaemonError("expected HELLO")
-                    conn.remote_id = PeerID(payload)
+                    if ep.startswith(RELAY_SCHEME):
+                        conn = await self._dial_via_relay(ep)
+                    else:
+                        host, port = split_endpoint(ep)
+                        reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
+                        conn = _Connection(self, reader, writer)
+                        await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+                        ftype, _, _, payload = await conn.recv_frame()
+                        if ftype != _T_HELLO:
+                            raise P2PDaemonError("expected HELLO")
+                        conn.remote_id = PeerID(payload)
                     if conn.remote_id != peer_id:
                         logger.warning(f"peer at {ep} identifies as {conn.remote_id}, expected {peer_id}")
                     self._endpoint_book[conn.remote_id] = ep
@@ -343,6 +503,8 @@ class P2P:
                     if task is not None:
                         task.cancel()
                     conn.serving_streams.pop(call_id, None)
+                elif ftype == _T_RELAY_OPEN:
+                    asyncio.create_task(self._relay_dial_back(call_id))
                 elif ftype == _T_PING:
                     pass
         except (asyncio.IncompleteReadError, ConnectionError, OSError):

@@ -255,3 +255,39 @@ def test_switch_moe_training():
     finally:
         server.shutdown()
         dht.shutdown()
+
+
+def test_expert_server_behind_relay():
+    """An expert server with NO inbound sockets serves through a circuit relay:
+    handler loops register with a public relay peer and advertise relay://
+    endpoints in the DHT (the reference's libp2p relay / NAT traversal)."""
+    from hivemind_amd.p2p import P2P
+    from hivemind_amd.utils.asyncio_utils import EventLoopThread
+
+    # public relay peer on its own loop
+    relay_loop = EventLoopThread(name="relay")
+    relay_loop.start_and_wait()
+    import asyncio as _asyncio
+
+    relay = _asyncio.run_coroutine_threadsafe(P2P.create(), relay_loop.loop).result(15)
+
+    dht = DHT(start=True)  # the DHT node itself is public here; only the
+    # expert server's RPC plane goes through the relay
+    server = Server.create(
+        dht=dht, expert_uids=["nat.0", "nat.1"], expert_cls="ffn", hidden_dim=16,
+        optim_cls=None, device="cpu", num_connection_handlers=2,
+        relay_endpoint=relay.endpoint, start=True,
+    )
+    try:
+        experts = get_experts(dht, ["nat.0", "nat.1"])
+        assert all(e is not None for e in experts)
+        x = torch.randn(3, 16, requires_grad=True)
+        out = experts[1](x)
+        assert out.shape == (3, 16)
+        out.sum().backward()
+        assert x.grad is not None
+    finally:
+        server.shutdown()
+        dht.shutdown()
+        _asyncio.run_coroutine_threadsafe(relay.shutdown(), relay_loop.loop).result(10)
+        relay_loop.shutdown()

@@ -145,3 +145,58 @@ def test_servicer_stub():
         await server.shutdown()
 
     run(main())
+
+
+def test_circuit_relay():
+    """A peer with NO listener (NATed) registers with a public relay peer and
+    becomes reachable at relay://<relay>/<peer>: unary and streaming RPCs run
+    through the spliced circuit (the reference's libp2p relay capability)."""
+
+    async def main():
+        relay = await P2P.create()
+        # NATed peer: no inbound socket at all -- only reachable via the relay
+        nated = await P2P.create(listen=False, relay_endpoint=relay.endpoint)
+        assert nated.endpoint.startswith("relay://")
+        assert nated.endpoint.endswith(nated.peer_id.to_base58())
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return b"via-relay:" + payload
+
+        async def doubler(items, ctx):
+            async for item in items:
+                yield item * 2
+
+        nated.add_unary_handler("echo", echo)
+        nated.add_stream_handler("doubler", doubler)
+
+        caller = await P2P.create()
+        info = nated.peer_info
+        assert info.endpoints and info.endpoints[0].startswith("relay://")
+        result = await caller.call_unary(info, "echo", b"hello", timeout=10)
+        assert result == b"via-relay:hello"
+
+        async def inputs():
+            yield b"ab"
+            yield b"cd"
+
+        out = [item async for item in caller.call_stream(info, "doubler", inputs())]
+        assert out == [b"abab", b"cdcd"]
+
+        # a second caller reuses the same advertised endpoint
+        caller2 = await P2P.create(listen=False)
+        result = await caller2.call_unary(info, "echo", b"x", timeout=10)
+        assert result == b"via-relay:x"
+
+        # dialing an unregistered target through the relay fails cleanly
+        from hivemind_amd.p2p.transport import P2PDaemonError
+        from hivemind_amd.p2p.peer_id import PeerInfo
+        from hivemind_amd.utils.crypto import PrivateKey as _PK
+        bogus_id = PeerID.from_identity(_PK())
+        bogus = PeerInfo(bogus_id, (f"relay://{relay.endpoint}/{bogus_id.to_base58()}",))
+        with pytest.raises(P2PDaemonError):
+            await caller.call_unary(bogus, "echo", b"", timeout=10)
+
+        for p in (caller, caller2, nated, relay):
+            await p.shutdown()
+
+    run(main())
