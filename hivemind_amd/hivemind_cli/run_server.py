@@ -36,6 +36,8 @@ def main():
     parser.add_argument("--checkpoint_dir", type=Path, default=None)
     parser.add_argument("--load_experts", action="store_true")
     parser.add_argument("--stats_report_interval", type=float, default=60.0)
+    parser.add_argument("--relay_endpoint", type=str, default=None,
+                        help="host:port of a public relay peer; serve through it when behind NAT (no inbound sockets)")
     args = parser.parse_args()
 
     optim_cls = {"adam": torch.optim.Adam, "sgd": torch.optim.SGD, "none": None}[args.optimizer]
@@ -56,6 +58,7 @@ def main():
         checkpoint_dir=args.checkpoint_dir,
         load_experts_from_dir=args.load_experts,
         stats_report_interval=args.stats_report_interval,
+        relay_endpoint=args.relay_endpoint,
         start=True,
     )
     logger.info(
