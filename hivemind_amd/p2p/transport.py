@@ -175,6 +175,7 @@ class P2P:
         self._relay_served: Dict[PeerID, float] = {}
         self._relay_waiting: Dict[bytes, asyncio.Future] = {}
         self._relay_endpoint: Optional[str] = None
+        self._relay_splices: set = set()
 
     # ------------------------------------------------------------------ setup
 
@@ -393,7 +394,12 @@ class P2P:
                 await caller_conn.close()
             return
         await caller_conn.send_frame(_T_RELAY_OK, b"")
-        await self._relay_splice(caller_conn, accept_conn)
+        task = asyncio.current_task()
+        self._relay_splices.add(task)
+        try:
+            await self._relay_splice(caller_conn, accept_conn)
+        finally:
+            self._relay_splices.discard(task)
 
     async def connect_endpoint(self, endpoint: str) -> PeerInfo:
         """Dial a bare endpoint and learn the peer's identity (bootstrap helper)."""
@@ -666,6 +672,9 @@ class P2P:
 
     async def shutdown(self):
         self._alive = False
+        for task in list(self._relay_splices):
+            task.cancel()
+        self._relay_splices.clear()
         if self._server is not None:
             self._server.close()
             try:
