@@ -80,6 +80,8 @@ def main():
     parser.add_argument("--batches-per-client", type=int, default=None)
     parser.add_argument("--batch-size", type=int, default=None)
     parser.add_argument("--max-batch-size", type=int, default=None)
+    parser.add_argument("--expert-cls", type=str, default="ffn",
+                        help="expert class; 'nop' isolates pure transport cost")
     parser.add_argument("--num-handlers", type=int, default=8,
                         help="balanced connection-handler loops on the server (8 measured best on a 256-core MI355X node)")
     args = parser.parse_args()
@@ -107,7 +109,7 @@ def main():
     dht = DHT(start=True)
     uids = [f"bench_ffn.{i}" for i in range(cfg["num_experts"])]
     server = Server.create(
-        dht=dht, expert_uids=uids, expert_cls="ffn", hidden_dim=cfg["hidden_dim"],
+        dht=dht, expert_uids=uids, expert_cls=args.expert_cls, hidden_dim=cfg["hidden_dim"],
         optim_cls=(torch.optim.Adam if cfg["backprop"] else None),
         max_batch_size=cfg["max_batch_size"], device=device, start=True,
         num_connection_handlers=args.num_handlers,

@@ -177,6 +177,7 @@ class DecentralizedAverager(ServicerBase):
             request_timeout=request_timeout,
         )
         self._initial_group_bits = initial_group_bits
+        self._inflight_futures: set = set()
         self._running_groups: Dict[bytes, asyncio.Future] = {}
         self.last_data_plane: Optional[str] = None  # "rccl" | "rpc" after a round
         # dedicated communicator: concurrent averagers (grad + state) must never
@@ -200,9 +201,16 @@ class DecentralizedAverager(ServicerBase):
     def _loop(self):
         return self.dht.loop
 
+    def _keep_future(self, future):
+        """Strong-reference fire-and-forget futures until done: asyncio tasks
+        are weakly referenced and can be GC'd mid-await otherwise."""
+        self._inflight_futures.add(future)
+        future.add_done_callback(self._inflight_futures.discard)
+        return future
+
     def run_in_background(self, await_ready: bool = True, timeout: Optional[float] = 30.0):
         self._alive = True
-        asyncio.run_coroutine_threadsafe(self._startup(), self._loop)
+        self._keep_future(asyncio.run_coroutine_threadsafe(self._startup(), self._loop))
         if await_ready:
             self._ready.result(timeout)
 
@@ -308,7 +316,7 @@ class DecentralizedAverager(ServicerBase):
         if not require_trigger:
             control.allow_allreduce()
         logger.debug(f"{self.prefix}@{self.peer_id}: step() created control {id(control):#x} trig={control.triggered}")
-        asyncio.run_coroutine_threadsafe(self._step(control), self._loop)
+        self._keep_future(asyncio.run_coroutine_threadsafe(self._step(control), self._loop))
         return control.result(timeout) if wait else control
 
     async def _step(self, step: StepControl):

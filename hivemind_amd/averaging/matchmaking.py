@@ -374,6 +374,7 @@ class PotentialLeaders:
     """Tracks candidate leaders sorted by declared expiration (reference matchmaking.py:414-546)."""
 
     def __init__(self, peer_id: PeerID, min_matchmaking_time: float):
+        self._background_tasks: set = set()
         self.peer_id = peer_id
         self.min_matchmaking_time = min_matchmaking_time
         self.running = asyncio.Event()
@@ -414,9 +415,11 @@ class PotentialLeaders:
                     self.declared_group_key, self.declared_expiration_time = None, float("inf")
                     self.leader_queue.clear()
                     self.past_attempts.clear()
-                    asyncio.create_task(
+                    _undeclare = asyncio.create_task(
                         key_manager.declare_averager(prev_key, self.peer_id, prev_expiration, looking_for_group=False)
                     )
+                    self._background_tasks.add(_undeclare)  # keep alive: tasks are weakly referenced
+                    _undeclare.add_done_callback(self._background_tasks.discard)
 
     def suggest_leader(self, peer_id: PeerID):
         self._suggested.append(peer_id)
@@ -494,6 +497,8 @@ class PotentialLeaders:
             if self.declared_group_key is not None:
                 prev_key, prev_expiration = self.declared_group_key, self.declared_expiration_time
                 with contextlib.suppress(Exception):
-                    asyncio.create_task(
+                    _undeclare = asyncio.create_task(
                         key_manager.declare_averager(prev_key, self.peer_id, prev_expiration, looking_for_group=False)
                     )
+                    self._background_tasks.add(_undeclare)  # keep alive: tasks are weakly referenced
+                    _undeclare.add_done_callback(self._background_tasks.discard)

@@ -57,6 +57,7 @@ _T_RELAY_OPEN = 12  # relay -> target (over its registered conn): call_id = chan
 _T_RELAY_ACCEPT = 13  # target -> relay (on a fresh socket): call_id = channel
 
 RELAY_SCHEME = "relay://"
+UNIX_SCHEME = "unix:"
 RELAY_REGISTER_HANDLER = "__relay__.register"
 
 MAX_FRAME_SIZE = 256 * 1024 * 1024  # control plane sanity bound
@@ -176,6 +177,8 @@ class P2P:
         self._relay_waiting: Dict[bytes, asyncio.Future] = {}
         self._relay_endpoint: Optional[str] = None
         self._relay_splices: set = set()
+        self._uds_path: Optional[str] = None
+        self._uds_server: Optional[asyncio.AbstractServer] = None
 
     # ------------------------------------------------------------------ setup
 
@@ -198,14 +201,38 @@ class P2P:
             self._port = self._server.sockets[0].getsockname()[1]
             # any listening peer can serve as a circuit relay for NATed peers
             self.add_unary_handler(RELAY_REGISTER_HANDLER, self._rpc_relay_register)
+            # same-host fast path: a unix-domain socket advertised alongside the
+            # TCP endpoint; remote peers fail the unix dial instantly and fall
+            # back to TCP, local peers skip the loopback TCP stack
+            try:
+                import tempfile
+
+                self._uds_path = os.path.join(
+                    tempfile.gettempdir(), f"hivemind_p2p_{os.getpid()}_{self._port}.sock"
+                )
+                self._uds_server = await asyncio.start_unix_server(
+                    self._on_accept, self._uds_path, limit=STREAM_BUFFER_LIMIT
+                )
+            except Exception as e:
+                logger.debug(f"no unix-domain listener: {e!r}")
+                self._uds_path, self._uds_server = None, None
         if relay_endpoint is not None:
             await self.register_with_relay(relay_endpoint)
         return self
 
     @property
+    def tcp_endpoint(self) -> str:
+        return make_endpoint(self._listen_host, self._port)
+
+    @property
     def endpoint(self) -> str:
+        """Advertised address: "unix:<path>,host:port" when a same-host socket
+        exists (dialers try components in order, unix fails instantly off-host),
+        "relay://<relay>/<peer>" for NATed peers, else plain host:port."""
         if not self._listen and self._relay_endpoint is not None:
             return f"{RELAY_SCHEME}{self._relay_endpoint}/{self.peer_id.to_base58()}"
+        if self._uds_path is not None:
+            return f"{UNIX_SCHEME}{self._uds_path},{self.tcp_endpoint}"
         return make_endpoint(self._listen_host, self._port)
 
     @property
@@ -227,7 +254,8 @@ class P2P:
         become reachable at ``relay://<relay_endpoint>/<our peer id>``."""
         info = await self.connect_endpoint(relay_endpoint)
         await self.call_unary(info.peer_id, RELAY_REGISTER_HANDLER, b"")
-        self._relay_endpoint = relay_endpoint
+        tcp_parts = [p for p in relay_endpoint.split(",") if not p.startswith(UNIX_SCHEME)]
+        self._relay_endpoint = tcp_parts[0] if tcp_parts else relay_endpoint
 
     async def _rpc_relay_register(self, _payload: bytes, ctx: "RpcContext") -> bytes:
         import time as _time
@@ -403,8 +431,23 @@ class P2P:
 
     async def connect_endpoint(self, endpoint: str) -> PeerInfo:
         """Dial a bare endpoint and learn the peer's identity (bootstrap helper)."""
-        host, port = split_endpoint(endpoint)
-        reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
+        parts = [p for p in endpoint.split(",") if p]
+        last_exc: Optional[Exception] = None
+        for part in parts:
+            try:
+                return await self._connect_endpoint_single(part)
+            except Exception as e:
+                last_exc = e
+        raise P2PDaemonError(f"could not connect to any of {parts}: {last_exc!r}")
+
+    async def _connect_endpoint_single(self, endpoint: str) -> PeerInfo:
+        if endpoint.startswith(UNIX_SCHEME):
+            reader, writer = await asyncio.wait_for(
+                asyncio.open_unix_connection(endpoint[len(UNIX_SCHEME):], limit=STREAM_BUFFER_LIMIT), timeout=10
+            )
+        else:
+            host, port = split_endpoint(endpoint)
+            reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
         conn = _Connection(self, reader, writer)
         await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
         ftype, _, _, payload = await conn.recv_frame()
@@ -437,6 +480,8 @@ class P2P:
             endpoints.insert(0, endpoint)
         if peer_id in self._endpoint_book:
             endpoints.append(self._endpoint_book[peer_id])
+        # expand "unix:/path,host:port" advertisements into individual dials
+        endpoints = [part for ep in endpoints for part in ep.split(",") if part]
         last_exc: Optional[Exception] = None
         async with self._conn_lock:
             conn = self._connections.get(peer_id)
@@ -446,6 +491,16 @@ class P2P:
                 try:
                     if ep.startswith(RELAY_SCHEME):
                         conn = await self._dial_via_relay(ep)
+                    elif ep.startswith(UNIX_SCHEME):
+                        reader, writer = await asyncio.wait_for(
+                            asyncio.open_unix_connection(ep[len(UNIX_SCHEME):], limit=STREAM_BUFFER_LIMIT), timeout=10
+                        )
+                        conn = _Connection(self, reader, writer)
+                        await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+                        ftype, _, _, payload = await conn.recv_frame()
+                        if ftype != _T_HELLO:
+                            raise P2PDaemonError("expected HELLO")
+                        conn.remote_id = PeerID(payload)
                     else:
                         host, port = split_endpoint(ep)
                         reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
@@ -510,7 +565,9 @@ class P2P:
                         task.cancel()
                     conn.serving_streams.pop(call_id, None)
                 elif ftype == _T_RELAY_OPEN:
-                    asyncio.create_task(self._relay_dial_back(call_id))
+                    task = asyncio.create_task(self._relay_dial_back(call_id))
+                    self._relay_splices.add(task)  # strong ref until done (tasks are weakly held)
+                    task.add_done_callback(self._relay_splices.discard)
                 elif ftype == _T_PING:
                     pass
         except (asyncio.IncompleteReadError, ConnectionError, OSError):
@@ -680,6 +737,16 @@ class P2P:
             try:
                 await self._server.wait_closed()
             except Exception:
+                pass
+        if self._uds_server is not None:
+            self._uds_server.close()
+            try:
+                await self._uds_server.wait_closed()
+            except Exception:
+                pass
+            try:
+                os.unlink(self._uds_path)
+            except OSError:
                 pass
         for conn in list(self._connections.values()):
             await conn.close()

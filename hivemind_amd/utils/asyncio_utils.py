@@ -32,6 +32,7 @@ class EventLoopThread(threading.Thread):
 
     def __init__(self, name: str = "hivemind-loop"):
         super().__init__(name=name, daemon=True)
+        self._inflight: set = set()
         self.loop = create_event_loop()
         self._started = threading.Event()
 
@@ -51,7 +52,16 @@ class EventLoopThread(threading.Thread):
         return future.result(timeout)
 
     def run_coroutine_async(self, coro: Awaitable[T]) -> concurrent.futures.Future:
-        return asyncio.run_coroutine_threadsafe(coro, self.loop)
+        """Schedule a coroutine and return its future. The future is ALSO kept
+        in a strong-reference registry until it completes: asyncio tasks are
+        only weakly referenced by the loop, and a task whose entire await graph
+        is an unreferenced cycle gets garbage-collected MID-AWAIT (the
+        coroutine receives GeneratorExit) -- observed as a flaky DHT startup
+        hang when the caller discarded this future."""
+        future = asyncio.run_coroutine_threadsafe(coro, self.loop)
+        self._inflight.add(future)
+        future.add_done_callback(self._inflight.discard)
+        return future
 
     def shutdown(self, timeout: float = 5.0):
         async def _stop():
