@@ -90,6 +90,21 @@ class RpcContext:
         self.local_id = local_id
 
 
+def _tune_socket_buffers(writer: asyncio.StreamWriter, size: int = 4 * 1024 * 1024):
+    """Unix sockets default to ~200 KB buffers; multi-MB tensor frames then pay
+    many extra wakeups. Loopback TCP autotunes, AF_UNIX does not -- set both
+    directions explicitly (best effort)."""
+    try:
+        import socket as _socket
+
+        sock = writer.get_extra_info("socket")
+        if sock is not None and sock.family == _socket.AF_UNIX:
+            sock.setsockopt(_socket.SOL_SOCKET, _socket.SO_SNDBUF, size)
+            sock.setsockopt(_socket.SOL_SOCKET, _socket.SO_RCVBUF, size)
+    except Exception:
+        pass
+
+
 class _Connection:
     """One TCP connection to a peer; carries multiplexed calls in both directions."""
 
@@ -97,6 +112,7 @@ class _Connection:
         self.p2p = p2p
         self.reader = reader
         self.writer = writer
+        _tune_socket_buffers(writer)
         self.remote_id: Optional[PeerID] = None
         self.send_lock = asyncio.Lock()
         # caller-side state for calls we initiated over this connection
@@ -205,6 +221,8 @@ class P2P:
             # TCP endpoint; remote peers fail the unix dial instantly and fall
             # back to TCP, local peers skip the loopback TCP stack
             try:
+                if os.environ.get("HIVEMIND_NO_UDS"):
+                    raise RuntimeError("disabled via HIVEMIND_NO_UDS")
                 import tempfile
 
                 self._uds_path = os.path.join(
