@@ -310,7 +310,12 @@ class TrainingStateAverager(DecentralizedAverager):
         """Copy optimized params back into the model (reference state_averager.py:589-595)."""
         if self.offload_optimizer:
             for main_param, opt_param in zip(self.main_parameters, self.optimized_parameters):
-                main_param.detach().copy_(opt_param.detach().to(main_param.device, main_param.dtype), non_blocking=True)
+                # write through .data: with delayed (DPU) steps this runs while
+                # the NEXT microbatch's autograd graph may still reference the
+                # parameter -- .data bypasses the version counter, so backward
+                # proceeds with slightly stale weights instead of crashing
+                # (reference state_averager applies delayed updates the same way)
+                main_param.data.copy_(opt_param.detach().to(main_param.device, main_param.dtype), non_blocking=True)
 
     @torch.no_grad()
     def _load_local_tensors_into_averager_(self):
@@ -329,11 +334,13 @@ class TrainingStateAverager(DecentralizedAverager):
                 # new_local = local + (averaged - old_local): tolerates local
                 # progress made while the (delayed) round was running
                 for loc, avg, old in zip(local, averaged, self.delta_buffers):
-                    loc.add_(avg.to(loc.device, loc.dtype) - old.to(loc.device, loc.dtype))
+                    loc.data.add_(avg.to(loc.device, loc.dtype) - old.to(loc.device, loc.dtype))
                 self.delta_buffers = None
             else:
+                # .data: see _apply_optimizer_parameters_ -- delayed rounds may
+                # overlap an in-flight autograd graph
                 for loc, avg in zip(local, averaged):
-                    loc.copy_(avg.to(loc.device, loc.dtype), non_blocking=True)
+                    loc.data.copy_(avg.to(loc.device, loc.dtype), non_blocking=True)
 
     def _update_scheduler(self):
         """Advance the LR scheduler to the current epoch (epoch-based schedule)."""
