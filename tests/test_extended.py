@@ -341,3 +341,39 @@ def test_dht_schema_validator():
     assert got is not None and got.value == b"exp1"
     dht2.shutdown()
     dht1.shutdown()
+
+
+def test_dpu_delayed_apply_does_not_break_autograd():
+    """Delayed (background) optimizer steps land while the next microbatch's
+    graph is alive; writes must go through .data or backward crashes with
+    'variable needed for gradient computation has been modified' (regression:
+    examples/albert trainer with DPU on a 1-peer swarm)."""
+    from hivemind_amd import Optimizer
+    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM
+
+    torch.manual_seed(0)
+    config = AlbertConfig.tiny()
+    config.dtype = torch.float32
+    model = AlbertForMaskedLM(config)
+    dht = DHT(start=True)
+    opt = Optimizer(
+        dht=dht, run_id="dpu_crash_repro", target_batch_size=8, batch_size_per_step=2,
+        optimizer=lambda pg: torch.optim.SGD(pg, lr=0.05),
+        params=[{"params": list(model.parameters())}],
+        offload_optimizer=True, delay_optimizer_step=True, delay_grad_averaging=True,
+        matchmaking_time=0.5, averaging_timeout=20.0,
+        tracker_opts=dict(min_refresh_period=0.1, default_refresh_period=0.2),
+    )
+    try:
+        for step in range(24):
+            ids = torch.randint(0, config.vocab_size, (2, 32))
+            labels = ids.clone()
+            labels[torch.rand(labels.shape) > 0.3] = -100
+            loss, _ = model(ids, labels=labels)
+            loss.backward()  # raised RuntimeError before the .data fix
+            opt.step()
+            opt.zero_grad()
+        assert opt.local_epoch >= 2
+    finally:
+        opt.shutdown()
+        dht.shutdown()
