@@ -241,3 +241,34 @@ def test_load_state_priority():
         avg.shutdown()
     for d in dhts:
         d.shutdown()
+
+
+def test_load_state_skips_dead_donor():
+    """If the best-priority donor dies before serving, the download falls
+    through to the next donor instead of failing (reference averager
+    load_state_from_peers donor iteration)."""
+    dhts = make_dht_swarm(3)
+    good = DecentralizedAverager(
+        [torch.full((4,), 7.0)], dhts[0], start=True, prefix="dead_donor",
+        target_group_size=2, request_timeout=0.5, declare_state_period=0.5,
+    )
+    dead = DecentralizedAverager(
+        [torch.full((4,), 666.0)], dhts[1], start=True, prefix="dead_donor",
+        target_group_size=2, request_timeout=0.5, declare_state_period=0.5,
+    )
+    dead.state_sharing_priority = 100  # best donor on paper
+    good.state_sharing_priority = 1
+    client = DecentralizedAverager(
+        [torch.zeros(4)], dhts[2], start=True, prefix="dead_donor",
+        target_group_size=2, request_timeout=0.5, allow_state_sharing=False,
+    )
+    time.sleep(1.2)  # both donors declared
+    # kill the high-priority donor abruptly (its DHT record remains)
+    dead.shutdown()
+    result = client.load_state_from_peers(timeout=30)
+    assert result is not None, "download should fall through to the live donor"
+    assert torch.allclose(result[1][0], torch.full((4,), 7.0)), result[1]
+    client.shutdown()
+    good.shutdown()
+    for d in dhts:
+        d.shutdown()
