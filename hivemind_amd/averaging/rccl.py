@@ -99,12 +99,21 @@ class DistributedAllReduceRunner:
         return float(sum(g.item() for g in gathered))
 
     def run(self) -> None:
-        """Execute the bucketed weighted all-reduce, in place (blocking)."""
+        """Execute the bucketed weighted all-reduce, in place (blocking).
+
+        The whole round (weight gather + buckets) runs under the process-wide
+        collective lock: collectives from two averagers must never interleave
+        on this rank. NOTE (cross-rank ordering): rounds on different process
+        groups must be INITIATED in the same order on every rank -- the
+        Optimizer guarantees this by awaiting each averaging round before
+        starting the next; background-overlapped rounds (DPU) rely on the
+        pre-scheduled matchmaking making every rank trigger rounds in epoch
+        order."""
         device = self.tensors[0].device if self.tensors else torch.device("cpu")
-        total_weight = self._gather_total_weight(device)
-        scale = self.weight / total_weight if total_weight > 0 else 0.0
         use_side_stream = device.type == "cuda"
         with _COLLECTIVE_LOCK:
+            total_weight = self._gather_total_weight(device)
+            scale = self.weight / total_weight if total_weight > 0 else 0.0
             if use_side_stream:
                 stream = _get_side_stream(device)
                 stream.wait_stream(torch.cuda.current_stream(device))
@@ -123,10 +132,13 @@ class DistributedAllReduceRunner:
                     for tensor in bucket_tensors:
                         n = tensor.numel()
                         avg_part = averaged[offset : offset + n].view_as(tensor)
+                        # .data: with reuse_tensors the target IS a live model
+                        # parameter, and a backward pass may be in flight on
+                        # another thread (DPU) -- bypass the version counter
                         if self.averaging_alpha == 1.0:
-                            tensor.detach().copy_(avg_part)
+                            tensor.data.copy_(avg_part)
                         else:
-                            tensor.detach().add_(avg_part - tensor.detach(), alpha=self.averaging_alpha)
+                            tensor.data.add_(avg_part - tensor.data, alpha=self.averaging_alpha)
                         offset += n
             if use_side_stream:
                 torch.cuda.current_stream(device).wait_stream(stream)
