@@ -19,7 +19,15 @@ from .compression import (
     Uniform8BitQuantization,
 )
 from .dht import DHT
-from .optim import GradScaler, GradientAverager, Optimizer, ProgressTracker, TrainingStateAverager
+from .moe import (
+    ModuleBackend,
+    RemoteExpert,
+    RemoteMixtureOfExperts,
+    RemoteSwitchMixtureOfExperts,
+    Server,
+    register_expert_class,
+)
+from .optim import GradScaler, GradientAverager, Optimizer, ProgressTracker, TrainingAverager, TrainingStateAverager
 from .p2p import P2P, PeerID, PeerInfo
 from .utils import get_dht_time, get_logger, use_hivemind_log_handler
 
