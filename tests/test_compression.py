@@ -80,3 +80,49 @@ def test_quantization_preserves_average(compression_type):
     true_avg = torch.stack(tensors).mean(0)
     approx_avg = torch.stack(restored).mean(0)
     assert (true_avg - approx_avg).abs().mean() < 0.05
+
+
+def test_adaptive_compression_selection_and_roundtrip():
+    """Size / role / per-key adaptive codecs pick the right child codec and
+    round-trip through the wire format (reference test: adaptive.py:25-66)."""
+    from hivemind_amd.compression import (
+        Float16Compression,
+        NoCompression,
+        Uniform8BitQuantization,
+        deserialize_torch_tensor,
+    )
+    from hivemind_amd.compression.adaptive import (
+        PerTensorCompression,
+        RoleAdaptiveCompression,
+        SizeAdaptiveCompression,
+    )
+    from hivemind_amd.compression.base import CompressionInfo, CompressionType, TensorRole
+
+    small = torch.randn(10)
+    large = torch.randn(10_000)
+
+    size_adaptive = SizeAdaptiveCompression(
+        threshold=2**10, less=NoCompression(), greater_equal=Float16Compression()
+    )
+    w_small = size_adaptive.compress(small, CompressionInfo.from_tensor(small))
+    w_large = size_adaptive.compress(large, CompressionInfo.from_tensor(large))
+    assert w_small.compression == CompressionType.NONE
+    assert w_large.compression == CompressionType.FLOAT16
+    assert torch.allclose(deserialize_torch_tensor(w_small), small)
+    assert torch.allclose(deserialize_torch_tensor(w_large), large, atol=1e-2)
+
+    role_adaptive = RoleAdaptiveCompression(
+        gradient=Uniform8BitQuantization(), parameter=Float16Compression(), default=NoCompression()
+    )
+    g = role_adaptive.compress(large, CompressionInfo.from_tensor(large, role=TensorRole.GRADIENT))
+    p = role_adaptive.compress(large, CompressionInfo.from_tensor(large, role=TensorRole.PARAMETER))
+    u = role_adaptive.compress(large, CompressionInfo.from_tensor(large, role=TensorRole.UNSPECIFIED))
+    assert g.compression == CompressionType.UNIFORM_8BIT
+    assert p.compression == CompressionType.FLOAT16
+    assert u.compression == CompressionType.NONE
+    assert (deserialize_torch_tensor(g) - large).abs().mean() < 0.05  # int8 quantization error
+
+    per_tensor = PerTensorCompression({"a": NoCompression(), "b": Float16Compression()})
+    wa = per_tensor.compress(small, CompressionInfo.from_tensor(small, key="a"))
+    wb = per_tensor.compress(small, CompressionInfo.from_tensor(small, key="b"))
+    assert wa.compression == CompressionType.NONE and wb.compression == CompressionType.FLOAT16
