@@ -154,3 +154,42 @@ def test_optimizer_convergence_two_peers():
         assert acc > 0.75, f"accuracy too low: {acc}"
     for d in dhts:
         d.shutdown()
+
+
+def test_training_averager_legacy():
+    """Legacy TrainingAverager: step() averages params and optimizer statistics
+    across two peers (reference test for optim/training_averager.py)."""
+    from hivemind_amd.optim.training_averager import TrainingAverager
+
+    dhts = make_dht_swarm(2)
+    models = [nn.Linear(4, 1, bias=False) for _ in range(2)]
+    with torch.no_grad():
+        models[0].weight.fill_(1.0)
+        models[1].weight.fill_(3.0)
+    opts = [torch.optim.SGD(m.parameters(), lr=0.1, momentum=0.9) for m in models]
+    averagers = [
+        TrainingAverager(
+            opt, average_parameters=True, average_gradients=False,
+            average_opt_statistics=("momentum_buffer",),
+            dht=dht, start=True, prefix="legacy_ta", target_group_size=2,
+            min_matchmaking_time=1.0, request_timeout=0.5,
+        )
+        for opt, dht in zip(opts, dhts)
+    ]
+    results = [None, None]
+
+    def run(i):
+        results[i] = averagers[i].step(timeout=60)
+
+    threads = [threading.Thread(target=run, args=(i,)) for i in range(2)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(90)
+    assert all(r is not None for r in results), results
+    for m in models:
+        assert torch.allclose(m.weight.detach(), torch.full((1, 4), 2.0), atol=1e-4), m.weight
+    for avg in averagers:
+        avg.shutdown()
+    for d in dhts:
+        d.shutdown()
