@@ -291,3 +291,19 @@ def test_expert_server_behind_relay():
         dht.shutdown()
         _asyncio.run_coroutine_threadsafe(relay.shutdown(), relay_loop.loop).result(10)
         relay_loop.shutdown()
+
+
+def test_background_server_context():
+    """background_server context manager serves experts for the with-block
+    (reference server.py:308 background_server)."""
+    from hivemind_amd.moe import background_server
+
+    with background_server(
+        expert_uids=["bgctx.0"], expert_cls="ffn", hidden_dim=8, optim_cls=None, device="cpu"
+    ) as server_info:
+        dht = DHT(initial_peers=[f"{ep}" for ep in server_info.endpoints], start=True)
+        (expert,) = get_experts(dht, ["bgctx.0"])
+        assert expert is not None
+        out = expert(torch.randn(2, 8))
+        assert out.shape == (2, 8)
+        dht.shutdown()
