@@ -115,7 +115,13 @@ class Optimizer(torch.optim.Optimizer):
             params=params,
             scheduler=scheduler,
             offload_optimizer=offload_optimizer,
-            custom_gradients=reuse_grad_buffers,
+            # offloaded optimizers get their gradients from
+            # _load_averaged_gradients_into_optimizer_ (or the local-grad
+            # loader) on the OPTIMIZER's schedule; auto-loading live
+            # main-param grads inside the (possibly delayed) step both used
+            # the wrong (unaveraged) gradients and raced the training thread
+            # (reference optimizer.py:291 custom_gradients=offload_optimizer)
+            custom_gradients=offload_optimizer,
             performance_ema_alpha=performance_ema_alpha,
             compression=state_averaging_compression,
             state_compression=load_state_compression,
@@ -458,14 +464,18 @@ class Optimizer(torch.optim.Optimizer):
                 "with reuse_grad_buffers=True, gradients are reset automatically at the global step; "
                 "do not call zero_grad manually"
             )
-        for group in self.param_groups:
-            for param in group["params"]:
-                if param.grad is None:
-                    continue
-                if set_to_none:
-                    param.grad = None
-                else:
-                    param.grad.zero_()
+        # clear the MODEL's gradients. With offload_optimizer, self.param_groups
+        # exposes the inner (offloaded) optimizer whose .grad tensors belong to
+        # the background step thread -- nulling those here raced a delayed
+        # optimizer step and segfaulted inside torch SGD (observed ~1/3 of
+        # full-suite runs via test_dpu_delayed_apply_does_not_break_autograd)
+        for param in self.state_averager.main_parameters:
+            if param.grad is None:
+                continue
+            if set_to_none:
+                param.grad = None
+            else:
+                param.grad.zero_()
 
     def state_dict(self) -> dict:
         state_dict = self.state_averager.optimizer.state_dict()
