@@ -268,8 +268,12 @@ class TrainingStateAverager(DecentralizedAverager):
                 self.optimizer.zero_grad(set_to_none=False)
                 if self.offload_optimizer:
                     for param in self.main_parameters:
-                        if param.grad is not None:
-                            param.grad.zero_()
+                        # snapshot the ref: the training thread may concurrently
+                        # set param.grad = None (its own zero_grad); zeroing a
+                        # tensor whose last reference vanished mid-call segfaults
+                        grad = param.grad
+                        if grad is not None:
+                            grad.zero_()
             if averaging_round:
                 if self.delta_rule_averaging:
                     with torch.no_grad():
@@ -300,10 +304,11 @@ class TrainingStateAverager(DecentralizedAverager):
         """Copy model gradients into the (possibly offloaded) optimizer params."""
         if self.offload_optimizer:
             for main_param, opt_param in zip(self.main_parameters, self.optimized_parameters):
-                if main_param.grad is not None:
+                grad = main_param.grad  # ref snapshot (see _do zero_grad note)
+                if grad is not None:
                     if opt_param.grad is None:
                         opt_param.grad = torch.zeros_like(opt_param)
-                    opt_param.grad.copy_(main_param.grad.to(opt_param.device, opt_param.dtype), non_blocking=True)
+                    opt_param.grad.copy_(grad.to(opt_param.device, opt_param.dtype), non_blocking=True)
 
     @torch.no_grad()
     def _apply_optimizer_parameters_(self):
