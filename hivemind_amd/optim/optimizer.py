@@ -371,7 +371,13 @@ class Optimizer(torch.optim.Optimizer):
             delay_averaging=self.delay_state_averaging and not self.auxiliary,
             averaging_control=averaging_control,
             averaging_opts=dict(timeout=self.averaging_timeout) if should_average_state else None,
-            zero_grad=True,
+            # zero_grad stays False (reference optimizer.py:481-491): the
+            # TRAINING thread owns the live .grad tensors -- clearing them from
+            # the (possibly delayed) background step races the user's own
+            # zero_grad/backward and segfaulted in torch's .grad accessor.
+            # Accumulators are reset by grad_averager.step(reset_accumulators);
+            # the offloaded optimizer's grads are rewritten by the next round.
+            zero_grad=False,
         )
         self.scheduled_state = None
 
