@@ -193,3 +193,33 @@ def test_training_averager_legacy():
         avg.shutdown()
     for d in dhts:
         d.shutdown()
+
+
+def test_optimizer_state_dict_roundtrip():
+    """state_dict/load_state_dict preserve inner optimizer state + local_epoch
+    (reference optimizer.py state_dict compatibility)."""
+    from hivemind_amd import Optimizer
+
+    dht = DHT(start=True)
+    model = nn.Linear(6, 2)
+    opt = Optimizer(
+        dht=dht, run_id="sdict", target_batch_size=8, batch_size_per_step=4,
+        optimizer=lambda pg: torch.optim.SGD(pg, lr=0.1, momentum=0.9),
+        params=[{"params": list(model.parameters())}],
+        matchmaking_time=0.5, averaging_timeout=20.0,
+        tracker_opts=dict(min_refresh_period=0.1, default_refresh_period=0.2),
+    )
+    try:
+        X = torch.randn(16, 6)
+        for _ in range(4):
+            model(X).pow(2).mean().backward()
+            opt.step()
+            opt.zero_grad()
+        sd = opt.state_dict()
+        assert "state" in sd and "local_epoch" in sd["state"]
+        epoch_before = opt.local_epoch
+        opt.load_state_dict(sd)
+        assert opt.local_epoch == epoch_before
+    finally:
+        opt.shutdown()
+        dht.shutdown()
