@@ -144,6 +144,7 @@ def main():
         matchmaking_time=1.0 if world_size > 1 else 0.5,
         averaging_timeout=120.0,
         reuse_grad_buffers=False,
+        grad_rccl_wire_dtype=torch.bfloat16,  # halve xGMI bytes for grad all-reduce
         delay_optimizer_step=args.dpu,
         delay_grad_averaging=args.dpu,
         average_state_every=4,

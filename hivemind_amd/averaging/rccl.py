@@ -120,10 +120,17 @@ class DistributedAllReduceRunner:
                 ctx = torch.cuda.stream(stream)
             else:
                 ctx = _NullCtx()
+            # gloo lacks reduced-precision collectives; wire casting is an
+            # xGMI-bandwidth optimization for the nccl(=RCCL) backend only
+            try:
+                backend = dist.get_backend(self.process_group)
+            except Exception:
+                backend = None
+            wire_dtype = self.wire_dtype if str(backend) == "nccl" else None
             with ctx:
                 for bucket_tensors in self._iter_buckets():
                     flat = torch.cat([t.detach().reshape(-1) for t in bucket_tensors])
-                    work_dtype = self.wire_dtype or flat.dtype
+                    work_dtype = wire_dtype or flat.dtype
                     flat_scaled = flat.to(work_dtype)
                     flat_scaled.mul_(scale)
                     dist.all_reduce(flat_scaled, op=dist.ReduceOp.SUM, group=self.process_group)

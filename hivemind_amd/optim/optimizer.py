@@ -68,6 +68,7 @@ class Optimizer(torch.optim.Optimizer):
         delay_state_averaging: bool = True,
         average_state_every: int = 1,
         use_local_updates: bool = False,
+        grad_rccl_wire_dtype=None,
         client_mode: bool = False,
         auxiliary: bool = False,
         grad_compression: CompressionBase = NoCompression(),
@@ -130,9 +131,15 @@ class Optimizer(torch.optim.Optimizer):
             **(averager_opts or {}),
         )
         if not use_local_updates:
+            grad_opts = dict(averager_opts or {})
+            if grad_rccl_wire_dtype is not None:
+                # cast gradient buckets on the RCCL/xGMI wire only (the named
+                # baseline config compresses gradient averaging; state
+                # averaging stays full precision)
+                grad_opts.setdefault("allreduce_wire_dtype", grad_rccl_wire_dtype)
             self.grad_averager: Optional[GradientAverager] = self._make_gradient_averager(
                 grad_averager_factory, reuse_grad_buffers=reuse_grad_buffers, compression=grad_compression,
-                **(averager_opts or {}),
+                **grad_opts,
             )
         else:
             self.grad_averager = None
