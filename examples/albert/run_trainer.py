@@ -102,6 +102,15 @@ def main():
     while opt.local_epoch < args.max_epochs:
         ids, labels = make_batch()
         loss, _ = model(ids, labels=labels)
+        if not torch.isfinite(loss):
+            # reference run_trainer: restore the last backup on NaN/inf loss
+            logger.warning(f"non-finite loss {loss.item()}; restoring from backup")
+            opt.zero_grad()
+            if os.path.exists(args.state_path):
+                backup = torch.load(args.state_path, weights_only=False)
+                model.load_state_dict(backup["model"])
+                opt.load_state_dict(backup["optimizer"])
+            continue
         loss.backward()
         opt.step()
         opt.zero_grad()

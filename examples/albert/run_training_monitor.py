@@ -33,6 +33,20 @@ def main():
     dht = DHT(initial_peers=args.initial_peers, start=True)
     logger.info(f"monitor connected; watching run '{args.run_id}'")
     progress_key = f"{args.run_id}_progress"
+
+    state_client = None
+    if args.save_checkpoint_to:
+        # a non-sharing client averager that can download the swarm's training
+        # state from the best donor (reference monitor checkpointing)
+        import torch
+
+        from hivemind_amd.averaging import DecentralizedAverager
+
+        state_client = DecentralizedAverager(
+            [torch.zeros(1)], dht, start=True, prefix=f"{args.run_id}_state_averager",
+            client_mode=True, allow_state_sharing=False, target_group_size=2,
+        )
+
     while True:
         record = dht.get(progress_key, latest=True)
         if record is not None and isinstance(record.value, dict):
@@ -52,6 +66,17 @@ def main():
             )
         else:
             logger.info("no training progress published yet")
+        if state_client is not None:
+            try:
+                import torch
+
+                state = state_client.load_state_from_peers(timeout=30)
+                if state is not None:
+                    metadata, tensors = state
+                    torch.save({"metadata": metadata, "tensors": tensors}, args.save_checkpoint_to)
+                    logger.info(f"saved swarm checkpoint ({len(tensors)} tensors) to {args.save_checkpoint_to}")
+            except Exception as e:
+                logger.warning(f"checkpoint download failed: {e!r}")
         time.sleep(args.refresh_period)
 
 

@@ -101,6 +101,10 @@ class DownloadData(RpcMessage):
 
 
 class DecentralizedAverager(ServicerBase):
+    # all averager subclasses (Training/Gradient/PowerSGD) speak one wire
+    # protocol: a plain client averager can download state from any of them
+    _servicer_name = "DecentralizedAverager"
+
     """Averages a fixed-schema list of tensors with dynamically matched groups of peers."""
 
     _matchmaking: Matchmaking

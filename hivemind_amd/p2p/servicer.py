@@ -97,6 +97,7 @@ class ServicerBase:
 
     _rpc_handlers = None
     _stub_class: Optional[Type[StubBase]] = None
+    _servicer_name: Optional[str] = None  # override: share one wire protocol across subclasses
 
     @classmethod
     def _collect_rpc_handlers(cls):
@@ -131,7 +132,8 @@ class ServicerBase:
     @classmethod
     def _handler_name(cls, method_name: str, namespace: Optional[str]) -> str:
         ns = f"{namespace}::" if namespace else ""
-        return f"{ns}{cls.__name__}.{method_name}"
+        name = cls._servicer_name or cls.__name__
+        return f"{ns}{name}.{method_name}"
 
     @classmethod
     def _make_stub_class(cls) -> Type[StubBase]:
