@@ -267,13 +267,28 @@ class P2P:
 
     # ----------------------------------------------------------------- relay
 
-    async def register_with_relay(self, relay_endpoint: str) -> None:
+    async def register_with_relay(self, relay_endpoint: str, keepalive: float = 30.0) -> None:
         """NATed side: keep an outbound connection to a public relay peer and
-        become reachable at ``relay://<relay_endpoint>/<our peer id>``."""
+        become reachable at ``relay://<relay_endpoint>/<our peer id>``. A
+        background keepalive re-registers periodically so a relay restart or a
+        dropped connection (or an expiring NAT mapping) self-heals."""
         info = await self.connect_endpoint(relay_endpoint)
         await self.call_unary(info.peer_id, RELAY_REGISTER_HANDLER, b"")
         tcp_parts = [p for p in relay_endpoint.split(",") if not p.startswith(UNIX_SCHEME)]
         self._relay_endpoint = tcp_parts[0] if tcp_parts else relay_endpoint
+
+        async def _keepalive():
+            while self._alive:
+                await asyncio.sleep(keepalive)
+                try:
+                    peer = await self.connect_endpoint(relay_endpoint)
+                    await self.call_unary(peer.peer_id, RELAY_REGISTER_HANDLER, b"", timeout=10)
+                except Exception as e:
+                    logger.debug(f"relay keepalive failed (will retry): {e!r}")
+
+        task = asyncio.create_task(_keepalive())
+        self._relay_splices.add(task)  # anchored + cancelled on shutdown
+        task.add_done_callback(self._relay_splices.discard)
 
     async def _rpc_relay_register(self, _payload: bytes, ctx: "RpcContext") -> bytes:
         import time as _time

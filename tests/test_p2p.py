@@ -200,3 +200,36 @@ def test_circuit_relay():
             await p.shutdown()
 
     run(main())
+
+
+def test_relay_reregisters_after_relay_connection_drop():
+    """The NATed peer's keepalive re-registers after its relay connection dies
+    (relay restart / NAT timeout self-healing)."""
+
+    async def main():
+        relay = await P2P.create()
+        nated = await P2P.create(listen=False)
+        await nated.register_with_relay(relay.endpoint, keepalive=0.3)
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return b"ok:" + payload
+
+        nated.add_unary_handler("echo", echo)
+        caller = await P2P.create()
+        assert await caller.call_unary(nated.peer_info, "echo", b"1", timeout=10) == b"ok:1"
+
+        # sever the nated<->relay control connection (simulates relay-side drop)
+        conn = relay._connections.get(nated.peer_id)
+        assert conn is not None
+        await conn.close()
+        nated_conn = nated._connections.get(relay.peer_id)
+        if nated_conn is not None:
+            await nated_conn.close()
+        await asyncio.sleep(1.0)  # keepalive re-registers
+
+        caller2 = await P2P.create(listen=False)
+        assert await caller2.call_unary(nated.peer_info, "echo", b"2", timeout=10) == b"ok:2"
+        for p in (caller, caller2, nated, relay):
+            await p.shutdown()
+
+    run(main())
