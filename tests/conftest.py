@@ -14,25 +14,36 @@ def pytest_configure(config):
     config.addinivalue_line("markers", "forked: legacy marker accepted for compatibility")
 
 
+_EXIT_STATUS = {"status": 0}
+
+
 def pytest_sessionfinish(session, exitstatus):
+    _EXIT_STATUS["status"] = int(exitstatus)
+
+
+def pytest_unconfigure(config):
     """Exit without CPython finalization: daemon event-loop threads (DHT loops
     shared across tests) can touch the interpreter during Py_Finalize and
-    SIGABRT AFTER all tests passed (~1/3 of full-suite runs), turning a green
-    run into rc=134. Stop the loops we can find, flush the report, then
-    _exit with pytest's own status."""
+    SIGABRT/SEGV AFTER all tests passed (~1/3 of full-suite runs), turning a
+    green run into rc=134/139. This runs after the terminal summary is
+    printed; stop the loops we can find, flush, then _exit with pytest's own
+    status."""
     import threading
 
-    from hivemind_amd.utils.asyncio_utils import EventLoopThread
+    try:
+        from hivemind_amd.utils.asyncio_utils import EventLoopThread
 
-    for thread in threading.enumerate():
-        if isinstance(thread, EventLoopThread):
-            try:
-                thread.shutdown(timeout=1.0)
-            except Exception:
-                pass
+        for thread in threading.enumerate():
+            if isinstance(thread, EventLoopThread):
+                try:
+                    thread.shutdown(timeout=1.0)
+                except Exception:
+                    pass
+    except Exception:
+        pass
     sys.stdout.flush()
     sys.stderr.flush()
-    os._exit(int(exitstatus))
+    os._exit(_EXIT_STATUS["status"])
 
 
 @pytest.fixture(autouse=True)

@@ -98,7 +98,7 @@ def _run_training_peer(dht, X, y, target_batch_size, batch_per_step, results, id
         run_id="conv_test",
         target_batch_size=target_batch_size,
         batch_size_per_step=batch_per_step,
-        optimizer=lambda pg: torch.optim.SGD(pg, lr=0.3),
+        optimizer=lambda pg: torch.optim.SGD(pg, lr=0.5),  # grads are per-sample averaged
         params=[{"params": list(model.parameters())}],
         matchmaking_time=1.0,
         averaging_timeout=30.0,
