@@ -377,3 +377,85 @@ def test_dpu_delayed_apply_does_not_break_autograd():
     finally:
         opt.shutdown()
         dht.shutdown()
+
+
+def test_composite_validator_merging():
+    """CompositeValidator merges same-type validators: one signature per record,
+    merged schemas validate keys from either source (reference
+    test_dht_validation.py:61)."""
+    import dataclasses
+
+    from typing import Dict as _Dict
+
+    import pydantic
+
+    from hivemind_amd.dht.crypto import SignatureValidator
+    from hivemind_amd.dht.routing import DHTID
+    from hivemind_amd.dht.schema import SchemaValidator
+    from hivemind_amd.dht.validation import CompositeValidator, DHTRecord
+    from hivemind_amd.utils.crypto import PrivateKey
+    from hivemind_amd.utils.serializer import MSGPackSerializer
+    from hivemind_amd.utils.timed_storage import get_dht_time
+
+    class SchemaA(pydantic.BaseModel):
+        field_a: bytes
+
+    class SchemaB(pydantic.BaseModel):
+        field_b: _Dict[bytes, int]
+
+    sig = SignatureValidator(PrivateKey())
+    composite = CompositeValidator([SchemaValidator(SchemaA, allow_extra_keys=False), sig])
+    composite.extend([SchemaValidator(SchemaB, allow_extra_keys=False), SignatureValidator(PrivateKey())])
+    # merged: one schema validator (2 schemas) + one signature validator
+    kinds = [type(v).__name__ for v in composite._validators]
+    assert kinds.count("SchemaValidator") == 1 and kinds.count("SignatureValidator") == 1
+
+    record = DHTRecord(
+        key=DHTID.generate(source="field_b").to_bytes(),
+        subkey=MSGPackSerializer.dumps(b"peer" + sig.local_public_key),
+        value=MSGPackSerializer.dumps(777),
+        expiration_time=get_dht_time() + 10,
+    )
+    signed = dataclasses.replace(record, value=composite.sign_value(record))
+    assert signed.value.count(b"[signature:") == 1  # merged validators sign once
+    assert composite.validate(signed)
+    assert composite.strip_value(signed) == record.value
+
+    unknown = DHTRecord(
+        key=DHTID.generate(source="unknown_key").to_bytes(),
+        subkey=b"",
+        value=MSGPackSerializer.dumps(777),
+        expiration_time=get_dht_time() + 10,
+    )
+    signed_unknown = dataclasses.replace(unknown, value=composite.sign_value(unknown))
+    assert not composite.validate(signed_unknown)  # no schema covers unknown_key
+
+
+def test_expert_backend_scheduler_state_restored():
+    """Checkpoint restore rewinds the LR scheduler's step count (reference
+    test_expert_backend.py:67 test_restore_update_count)."""
+    from hivemind_amd.moe.server.checkpoints import load_experts, store_experts
+    from hivemind_amd.moe.server.layers import name_to_block
+    from hivemind_amd.moe.server.layers.lr_schedule import get_linear_schedule_with_warmup
+    from hivemind_amd.moe.server.module_backend import ModuleBackend
+    from hivemind_amd.utils.tensor_descr import BatchTensorDescriptor
+
+    expert = name_to_block["ffn"](8)
+    optimizer = torch.optim.SGD(expert.parameters(), lr=1.0)
+    scheduler = get_linear_schedule_with_warmup(optimizer, num_warmup_steps=100, num_training_steps=1000)
+    backend = ModuleBackend(
+        name="restore.0", module=expert, optimizer=optimizer, scheduler=scheduler,
+        args_schema=(BatchTensorDescriptor(8),),
+    )
+    backends = {"restore.0": backend}
+    x, g = torch.randn(2, 8), torch.randn(2, 8)
+    with tempfile.TemporaryDirectory() as tmpdir:
+        for _ in range(3):
+            backend.backward(x, g)
+        store_experts(backends, Path(tmpdir))
+        lr_at_save = optimizer.param_groups[0]["lr"]
+        for _ in range(4):
+            backend.backward(x, g)
+        assert optimizer.param_groups[0]["lr"] != lr_at_save
+        load_experts(backends, Path(tmpdir))
+        assert optimizer.param_groups[0]["lr"] == pytest.approx(lr_at_save)
