@@ -25,7 +25,8 @@ def main():
     parser.add_argument("--num_params", type=int, default=1_000_000)
     parser.add_argument("--num_rounds", type=int, default=3)
     parser.add_argument("--compression", choices=["none", "fp16", "blockwise8"], default="none")
-    parser.add_argument("--matchmaking_time", type=float, default=2.0)
+    parser.add_argument("--matchmaking_time", type=float, default=5.0,
+                        help="groups assemble EARLY once full; this window only gives laggards slack")
     args = parser.parse_args()
     target_group_size = args.target_group_size or args.num_peers
 
@@ -60,11 +61,16 @@ def main():
     round_times, successes = [], 0
     for round_idx in range(args.num_rounds):
         t0 = time.perf_counter()
-        controls = [avg.step(wait=False, timeout=120) for avg in averagers]
+        # a peer that misses a round by milliseconds (leader assembled at its
+        # expiration just before the last join arrived) has nobody left to
+        # group with until the others step again -- bound that straggler cost
+        # instead of letting it run a 120 s deadline (same dynamics as the
+        # reference's skipped test_overcrowded)
+        controls = [avg.step(wait=False, timeout=30) for avg in averagers]
         ok = 0
         for c in controls:
             try:
-                if c.result(150) is not None:
+                if c.result(60) is not None:
                     ok += 1
             except Exception:
                 pass
