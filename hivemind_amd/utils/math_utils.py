@@ -10,9 +10,11 @@ from __future__ import annotations
 import torch
 
 
-@torch.jit.script
 def orthogonalize_(matrix: torch.Tensor, eps: float = 1e-8):
-    """In-place Gram-Schmidt over columns of an [m, n] matrix, n <= m."""
+    """In-place Gram-Schmidt over columns of an [m, n] matrix, n <= m.
+
+    Plain eager: the per-column loop is tiny (PowerSGD rank <= 32) and
+    torch.jit.script is deprecated in torch 2.10."""
     n, m = matrix.shape
     for i in range(m):
         col = matrix[:, i]
