@@ -40,6 +40,8 @@ def main():
                         choices=["albert-base", "albert-large", "tiny", "llama-8b", "llama-1b", "llama-tiny"])
     parser.add_argument("--dpu", action="store_true",
                         help="delayed parameter updates: overlap averaging + optimizer step with compute")
+    parser.add_argument("--powersgd-rank", type=int, default=0,
+                        help="if > 0, average gradients with rank-r PowerSGD + error feedback (baseline config 3)")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -145,6 +147,11 @@ def main():
         averaging_timeout=120.0,
         reuse_grad_buffers=False,
         grad_rccl_wire_dtype=torch.bfloat16,  # halve xGMI bytes for grad all-reduce
+        grad_averager_factory=(
+            (lambda **kw: __import__("hivemind_amd.optim.power_sgd_averager", fromlist=["PowerSGDGradientAverager"])
+             .PowerSGDGradientAverager(averager_rank=args.powersgd_rank, **kw))
+            if args.powersgd_rank > 0 else None
+        ),
         delay_optimizer_step=args.dpu,
         delay_grad_averaging=args.dpu,
         average_state_every=4,
@@ -222,6 +229,7 @@ def main():
             "config": {
                 "model": args.model,
                 "dpu": args.dpu,
+                "powersgd_rank": args.powersgd_rank or None,
                 "global_batch": args.target_batch_size,
                 "per_gpu_batch": args.batch,
                 "seq_len": args.seq_len,
