@@ -124,3 +124,45 @@ def test_load_balancing():
     assert sum(load_balance_peers(30, [None, None, None])) == 30
     assert hagenbach_bishoff(10, [1, 1]) == [5, 5]
     assert sum(hagenbach_bishoff(11, [1.0, 2.0, 3.0])) == 11
+
+
+def test_partitioning_asynchronous_liveness():
+    """Compression-heavy partitioning must not starve the event loop: prefetch
+    runs in an executor, so concurrent coroutines keep getting scheduled
+    (reference test_allreduce.py:80-115)."""
+    import time as _time
+
+    from hivemind_amd.compression import Quantile8BitQuantization, deserialize_torch_tensor
+    from hivemind_amd.utils.asyncio_utils import aenumerate
+
+    tensors = [torch.randn(1024, 1024), torch.randn(512, 2048), torch.randn(8_000, 1024)]
+    peer_fractions = [0.4, 0.3, 0.3]
+
+    async def main():
+        partition = TensorPartContainer(tensors, peer_fractions, compression=Quantile8BitQuantization())
+        read_started, read_finished = asyncio.Event(), asyncio.Event()
+
+        async def write_tensors():
+            for peer_index in range(len(peer_fractions)):
+                async for part_index, part in aenumerate(partition.iterate_input_parts_for(peer_index)):
+                    partition.register_processed_part(peer_index, part_index, deserialize_torch_tensor(part))
+            assert read_started.is_set(), "reading should start before writing finishes"
+
+        async def read_tensors():
+            async for _ in partition.iterate_output_tensors():
+                read_started.set()
+            read_finished.set()
+
+        async def wait_synchronously():
+            waited = 0.0
+            while not read_finished.is_set():
+                await asyncio.sleep(0.01)
+                waited += 0.01
+            return waited
+
+        t0 = _time.perf_counter()
+        *_, waited = await asyncio.gather(write_tensors(), read_tensors(), wait_synchronously())
+        wall = _time.perf_counter() - t0
+        assert waited > wall / 4, f"event loop ran only {100 * waited / wall:.1f}% of the time"
+
+    asyncio.run(main())
