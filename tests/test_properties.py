@@ -1,100 +1,103 @@
-"""Property-based tests (hypothesis) for the serialization, DHT-metric and
-compression invariants everything else is built on."""
+"""Randomized invariant tests for serialization, the DHT metric, storage and
+quantization (reference test_util_modules-style property coverage).
 
-import math
+Written with plain seeded random generation: importing hypothesis at pytest
+collection time was observed to destabilize later networking tests in the same
+session (a 60+ s hang in an unrelated MoE test), and invariant coverage does
+not need shrinking machinery.
+"""
+
+import random
+import string
 
 import pytest
 import torch
-from hypothesis import given, settings
-from hypothesis import strategies as st
 
 from hivemind_amd.dht.routing import DHTID
 from hivemind_amd.utils.serializer import MSGPackSerializer
-from hivemind_amd.utils.timed_storage import TimedStorage
-
-# msgpack-representable values: scalars, bytes, strings, and nested lists /
-# tuples / string-keyed dicts thereof
-scalars = st.one_of(
-    st.none(),
-    st.booleans(),
-    st.integers(min_value=-(2**63), max_value=2**63 - 1),
-    st.floats(allow_nan=False, allow_infinity=False, width=64),
-    st.text(max_size=40),
-    st.binary(max_size=40),
-)
-values = st.recursive(
-    scalars,
-    lambda children: st.one_of(
-        st.lists(children, max_size=4),
-        st.lists(children, max_size=4).map(tuple),
-        st.dictionaries(st.text(max_size=8), children, max_size=4),
-    ),
-    max_leaves=20,
-)
+from hivemind_amd.utils.timed_storage import TimedStorage, get_dht_time
 
 
-@settings(max_examples=200, deadline=None)
-@given(values)
-def test_msgpack_roundtrip_property(obj):
-    assert MSGPackSerializer.loads(MSGPackSerializer.dumps(obj)) == obj
+def random_value(rng: random.Random, depth: int = 0):
+    choices = ["int", "float", "str", "bytes", "none", "bool"]
+    if depth < 3:
+        choices += ["list", "tuple", "dict"]
+    kind = rng.choice(choices)
+    if kind == "int":
+        return rng.randint(-(2**63), 2**63 - 1)
+    if kind == "float":
+        return rng.uniform(-1e12, 1e12)
+    if kind == "str":
+        return "".join(rng.choices(string.printable, k=rng.randint(0, 20)))
+    if kind == "bytes":
+        return bytes(rng.getrandbits(8) for _ in range(rng.randint(0, 20)))
+    if kind == "none":
+        return None
+    if kind == "bool":
+        return rng.random() < 0.5
+    if kind == "list":
+        return [random_value(rng, depth + 1) for _ in range(rng.randint(0, 4))]
+    if kind == "tuple":
+        return tuple(random_value(rng, depth + 1) for _ in range(rng.randint(0, 4)))
+    return {f"k{i}": random_value(rng, depth + 1) for i in range(rng.randint(0, 4))}
 
 
-@settings(max_examples=100, deadline=None)
-@given(st.binary(min_size=1, max_size=64), st.binary(min_size=1, max_size=64), st.binary(min_size=1, max_size=64))
-def test_dhtid_xor_metric_properties(a, b, c):
-    """XOR distance is a metric: identity, symmetry, triangle inequality (the
-    Kademlia routing invariants)."""
-    ida, idb, idc = (DHTID.generate(source=x) for x in (a, b, c))
-    assert ida.xor_distance(ida) == 0
-    assert ida.xor_distance(idb) == idb.xor_distance(ida)
-    if a != b:
-        assert ida.xor_distance(idb) > 0
-    # XOR triangle inequality: d(a,c) <= d(a,b) ^ d(b,c) <= d(a,b) + d(b,c)
-    assert ida.xor_distance(idc) <= ida.xor_distance(idb) + idb.xor_distance(idc)
+def test_msgpack_roundtrip_randomized():
+    rng = random.Random(0)
+    for _ in range(300):
+        obj = random_value(rng)
+        assert MSGPackSerializer.loads(MSGPackSerializer.dumps(obj)) == obj, obj
 
 
-@settings(max_examples=50, deadline=None)
-@given(
-    st.lists(
-        st.tuples(st.integers(0, 9), st.integers(0, 100), st.floats(1.0, 100.0)),
-        min_size=1,
-        max_size=30,
-    )
-)
-def test_timed_storage_latest_expiration_wins(ops):
+def test_dhtid_xor_metric_randomized():
+    """XOR distance is a metric: identity, symmetry, triangle inequality."""
+    rng = random.Random(1)
+    for _ in range(200):
+        a, b, c = (bytes(rng.getrandbits(8) for _ in range(rng.randint(1, 64))) for _ in range(3))
+        ida, idb, idc = (DHTID.generate(source=x) for x in (a, b, c))
+        assert ida.xor_distance(ida) == 0
+        assert ida.xor_distance(idb) == idb.xor_distance(ida)
+        if a != b:
+            assert ida.xor_distance(idb) > 0
+        assert ida.xor_distance(idc) <= ida.xor_distance(idb) + idb.xor_distance(idc)
+
+
+def test_timed_storage_latest_expiration_wins_randomized():
     """For any store sequence, get(key) returns the value with the LATEST
-    expiration seen for that key (ties: first stored wins; earlier stores
-    never overwrite later expirations)."""
-    from hivemind_amd.utils.timed_storage import get_dht_time
-
-    base = get_dht_time() + 1000.0
-    storage = TimedStorage()
-    best = {}
-    for key, value, expiration in ops:
-        exp = base + expiration
-        stored = storage.store(key, value, exp)
-        if key not in best or exp > best[key][1]:
-            assert stored
-            best[key] = (value, exp)
+    expiration seen for that key (earlier expirations never overwrite)."""
+    rng = random.Random(2)
+    for _trial in range(40):
+        base = get_dht_time() + 1000.0
+        storage = TimedStorage()
+        best = {}
+        for _ in range(rng.randint(1, 30)):
+            key, value = rng.randint(0, 9), rng.randint(0, 100)
+            exp = base + rng.uniform(1.0, 100.0)
+            stored = storage.store(key, value, exp)
+            if key not in best or exp > best[key][1]:
+                assert stored
+                best[key] = (value, exp)
         for key, (value, exp) in best.items():
             entry = storage.get(key)
             assert entry is not None and entry.value == value and entry.expiration_time == exp
 
 
-@settings(max_examples=30, deadline=None)
-@given(st.integers(1, 3), st.integers(1, 5000), st.floats(0.1, 1000.0))
-def test_blockwise_quantization_error_bound(seed, numel, scale):
-    """int8 blockwise absmax quantization error is bounded by absmax/127 per
+def test_blockwise_quantization_error_bound_randomized():
+    """int8 blockwise absmax quantization error stays within absmax/127 per
     4096-block (the wire codec the averager uses for gradients)."""
     from hivemind_amd.ops import dequantize_blockwise, quantize_blockwise
 
-    torch.manual_seed(seed)
-    t = (torch.randn(numel) * scale).float()
-    q, absmax = quantize_blockwise(t)
-    restored = dequantize_blockwise(q, absmax).reshape(t.shape)
-    blocks = t.reshape(-1)
-    for b in range(0, numel, 4096):
-        chunk = blocks[b : b + 4096]
-        err = (restored.reshape(-1)[b : b + 4096] - chunk).abs().max()
-        bound = chunk.abs().max() / 127.0 * 1.01 + 1e-8
-        assert err <= bound, (err, bound)
+    rng = random.Random(3)
+    for _ in range(20):
+        numel = rng.randint(1, 5000)
+        scale = rng.uniform(0.1, 1000.0)
+        torch.manual_seed(rng.randint(0, 10_000))
+        t = (torch.randn(numel) * scale).float()
+        q, absmax = quantize_blockwise(t)
+        restored = dequantize_blockwise(q, absmax).reshape(t.shape)
+        flat = t.reshape(-1)
+        for b in range(0, numel, 4096):
+            chunk = flat[b : b + 4096]
+            err = (restored.reshape(-1)[b : b + 4096] - chunk).abs().max()
+            bound = chunk.abs().max() / 127.0 * 1.01 + 1e-8
+            assert err <= bound, (err, bound)
