@@ -223,7 +223,15 @@ class AllReduceRunner:
         sender_index = None
         try:
             first_message = await asyncio.wait_for(anext_impl(stream.__aiter__() if hasattr(stream, "__aiter__") else stream), self.sender_timeout)
-            sender_peer_id = PeerID(first_message.peer_id)
+            # identify the sender from the AUTHENTICATED connection, not the
+            # self-declared message field (reference allreduce.py:263 uses
+            # context.remote_id; ADVICE round 1: the message field would let a
+            # group member impersonate another sender or double-contribute)
+            sender_peer_id = context.remote_id
+            if first_message.peer_id and PeerID(first_message.peer_id) != sender_peer_id:
+                logger.debug(
+                    f"{self}: sender {sender_peer_id} declared a different peer_id in-band; ignoring the in-band value"
+                )
             if sender_peer_id not in self.sender_peer_ids:
                 yield AveragingData(code=DataCode.ERROR, group_id=self.group_id)
                 return

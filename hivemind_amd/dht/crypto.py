@@ -63,9 +63,15 @@ class SignatureValidator(RecordValidatorBase):
         if signature_match is None:
             logger.debug("protected record has no signature")
             return False
-        signature = base64.b64decode(signature_match.group(1))
-        stripped = dataclasses.replace(record, value=self.SIGNATURE_RE.sub(b"", record.value))
-        return public_key.verify(self._serialize_record(stripped), signature)
+        try:
+            signature = base64.b64decode(signature_match.group(1))
+            stripped = dataclasses.replace(record, value=self.SIGNATURE_RE.sub(b"", record.value))
+            return public_key.verify(self._serialize_record(stripped), signature)
+        except Exception:
+            # a malformed/hostile envelope must reject this record only, not
+            # blow up the whole batched rpc_store/rpc_find
+            logger.debug("malformed signature envelope on protected record")
+            return False
 
     def sign_value(self, record: DHTRecord) -> bytes:
         if self._local_public_key not in record.key and (not record.subkey or self._local_public_key not in record.subkey):

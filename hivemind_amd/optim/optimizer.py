@@ -329,6 +329,13 @@ class Optimizer(torch.optim.Optimizer):
         if self.use_gradient_averaging and not self.auxiliary:
             if swarm_not_empty:
                 began_averaging_gradients = self._begin_averaging_gradients()
+                if not began_averaging_gradients:
+                    # failed to start the round: fall back to local accumulators
+                    # (reference optimizer.py:452-454) so the optimizer steps on
+                    # this round's gradients, not the averager's stale buffers,
+                    # and the accumulators can't double-count into next round
+                    self.grad_averager.load_accumulators_into_averager_()
+                    self.grad_averager.reset_accumulated_grads_()
             else:
                 # single-peer swarm: skip matchmaking, use local accumulators as-is
                 self.grad_averager.load_accumulators_into_averager_()
@@ -425,6 +432,13 @@ class Optimizer(torch.optim.Optimizer):
             except Exception as e:
                 logger.log(self.status_loglevel, f"gradient averaging failed: {e!r}; using local gradients")
         self.scheduled_grads = None
+        if maybe_step_control is not None and not averaged:
+            # the round was triggered but failed: re-load the accumulators into
+            # the averager so the optimizer sees deterministic local gradients
+            # instead of a partially-averaged buffer (reference
+            # optimizer.py:608-610 + _load_local_gradients_into_optimizer)
+            self.grad_averager.load_accumulators_into_averager_()
+            self.grad_averager.reset_accumulated_grads_()
         self._load_averaged_gradients_into_optimizer_()
 
     def _load_averaged_gradients_into_optimizer_(self):
@@ -442,10 +456,14 @@ class Optimizer(torch.optim.Optimizer):
     # --------------------------------------------------------------- re-sync
 
     def _should_load_state_from_peers(self) -> bool:
-        """True if the swarm advanced 2+ epochs past us (reference optimizer.py:655-673)."""
+        """True if the swarm advanced past us (reference optimizer.py:655-673):
+        strict check (any epoch difference) on the first call after an epoch
+        update, lenient check (2+ epochs behind) otherwise -- so a peer that is
+        exactly one epoch behind still catches up once per epoch instead of
+        training stale forever."""
         if self._should_check_synchronization_on_update and self.tracker.fetched_global_progress_this_epoch.is_set():
             self._should_check_synchronization_on_update = False
-            return self.local_epoch < self.tracker.global_epoch - 1
+            return self.local_epoch != self.tracker.global_epoch
         return self.local_epoch < self.tracker.global_epoch - 1
 
     def is_synchronized_with_peers(self) -> bool:

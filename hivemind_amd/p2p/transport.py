@@ -26,7 +26,7 @@ import os
 import struct
 from typing import AsyncIterator, Awaitable, Callable, Dict, List, Optional, Tuple, Union
 
-from ..utils.crypto import PrivateKey
+from ..utils.crypto import PrivateKey, PublicKey
 from ..utils.logging import get_logger
 from ..utils.networking import LOCALHOST, make_endpoint, split_endpoint
 from ..utils.serializer import MSGPackSerializer
@@ -55,6 +55,11 @@ _T_RELAY_CONNECT = 10  # caller -> relay: payload = target peer id
 _T_RELAY_OK = 11  # relay -> caller: target dialed back, pipe is live
 _T_RELAY_OPEN = 12  # relay -> target (over its registered conn): call_id = channel
 _T_RELAY_ACCEPT = 13  # target -> relay (on a fresh socket): call_id = channel
+_T_AUTH = 14  # handshake: Ed25519 signature over the remote HELLO's nonce
+
+# Domain separator for handshake signatures; a signature produced here can
+# never be replayed as a DHT record signature (dht/crypto.py) or vice versa.
+_AUTH_TAG = b"hivemind-amd-handshake-v1:"
 
 RELAY_SCHEME = "relay://"
 UNIX_SCHEME = "unix:"
@@ -307,12 +312,9 @@ class P2P:
             conn = _Connection(self, reader, writer)
             await conn.recv_frame()  # absorb the relay's eager HELLO
             await conn.send_frame(_T_RELAY_ACCEPT, channel)
-            # from here the socket is spliced to the caller: ordinary handshake
-            await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-            ftype, _, _, payload = await conn.recv_frame()
-            if ftype != _T_HELLO:
-                raise P2PDaemonError("relayed caller did not say HELLO")
-            conn.remote_id = PeerID(payload)
+            # from here the socket is spliced to the caller: ordinary
+            # authenticated handshake (the relay is only a byte pipe)
+            await self._handshake_outbound(conn)
             self._register_connection(conn)
             conn.reader_task = asyncio.create_task(self._connection_loop(conn))
         except Exception as e:
@@ -338,12 +340,10 @@ class P2P:
                 raise P2PDaemonError(payload.decode(errors="replace"))
             if ftype != _T_RELAY_OK:
                 raise P2PDaemonError(f"unexpected relay response type {ftype}")
-            # circuit is live: ordinary handshake with the target through the pipe
-            await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-            ftype, _, _, payload = await conn.recv_frame()
-            if ftype != _T_HELLO:
-                raise P2PDaemonError("relayed target did not say HELLO")
-            conn.remote_id = PeerID(payload)
+            # circuit is live: ordinary authenticated handshake with the target
+            await self._handshake_outbound(conn)
+            if conn.remote_id != target:
+                raise P2PDaemonError(f"relayed peer proved identity {conn.remote_id}, expected {target}")
             return conn
         except Exception:
             try:
@@ -408,10 +408,58 @@ class P2P:
 
     # ----------------------------------------------------------- connections
 
+    def _hello_payload(self) -> Tuple[bytes, bytes]:
+        """(payload, nonce): HELLO announces PeerID + Ed25519 pubkey + a fresh
+        nonce the remote must sign to prove it holds the key the PeerID hashes.
+
+        The reference gets authenticated identity from libp2p's secure channel
+        (hivemind/p2p/p2p_daemon_bindings/datastructures.py:66-88: PeerID =
+        multihash(pubkey) verified by TLS); this transport proves it with an
+        explicit challenge-response."""
+        nonce = os.urandom(16)
+        payload = MSGPackSerializer.dumps(
+            [self.peer_id.to_bytes(), self._identity.get_public_key().to_bytes(), nonce]
+        )
+        return payload, nonce
+
+    async def _finish_auth(self, conn: _Connection, remote_hello: bytes, my_nonce: bytes) -> PeerID:
+        """Verify the remote HELLO and exchange AUTH signatures. Returns the
+        PROVEN remote PeerID; raises (and the caller closes the socket) if the
+        claimed identity does not hash from the presented key or the signature
+        over our nonce fails."""
+        try:
+            claimed_id_bytes, pubkey_bytes, remote_nonce = MSGPackSerializer.loads(remote_hello)
+            pubkey = PublicKey.from_bytes(pubkey_bytes)
+        except Exception as e:
+            raise P2PDaemonError(f"malformed HELLO: {e!r}") from e
+        claimed_id = PeerID(claimed_id_bytes)
+        if PeerID.from_public_key(pubkey) != claimed_id:
+            raise P2PDaemonError(f"peer claims id {claimed_id} but its public key hashes differently")
+        my_pub = self._identity.get_public_key().to_bytes()
+        signature = self._identity.sign(_AUTH_TAG + remote_nonce + my_pub)
+        await conn.send_frame(_T_AUTH, b"", "", signature)
+        ftype, _, _, remote_sig = await conn.recv_frame()
+        if ftype != _T_AUTH:
+            raise P2PDaemonError(f"expected AUTH frame, got type {ftype}")
+        if not pubkey.verify(_AUTH_TAG + my_nonce + pubkey_bytes, remote_sig):
+            raise P2PDaemonError(f"peer {claimed_id} failed the handshake signature check")
+        return claimed_id
+
+    async def _handshake_outbound(self, conn: _Connection) -> None:
+        """Client side of the handshake: send HELLO, read the server's HELLO,
+        run the mutual AUTH exchange. Sets conn.remote_id to the proven id."""
+        hello, nonce = self._hello_payload()
+        await conn.send_frame(_T_HELLO, b"", "", hello)
+        ftype, _, _, payload = await conn.recv_frame()
+        if ftype != _T_HELLO:
+            raise P2PDaemonError("expected HELLO")
+        conn.remote_id = await self._finish_auth(conn, payload, nonce)
+
     async def _on_accept(self, reader: asyncio.StreamReader, writer: asyncio.StreamWriter):
         conn = _Connection(self, reader, writer)
         try:
-            await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
+            hello, nonce = self._hello_payload()
+            await conn.send_frame(_T_HELLO, b"", "", hello)
             ftype, call_id, _, payload = await conn.recv_frame()
             if ftype == _T_RELAY_CONNECT:
                 await self._serve_relay_connect(conn, PeerID(payload))
@@ -425,8 +473,9 @@ class P2P:
                 return
             if ftype != _T_HELLO:
                 raise P2PDaemonError("expected HELLO")
-            conn.remote_id = PeerID(payload)
-        except Exception:
+            conn.remote_id = await self._finish_auth(conn, payload, nonce)
+        except Exception as e:
+            logger.debug(f"inbound handshake failed: {e!r}")
             writer.close()
             return
         self._register_connection(conn)
@@ -482,11 +531,7 @@ class P2P:
             host, port = split_endpoint(endpoint)
             reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
         conn = _Connection(self, reader, writer)
-        await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-        ftype, _, _, payload = await conn.recv_frame()
-        if ftype != _T_HELLO:
-            raise P2PDaemonError("expected HELLO")
-        conn.remote_id = PeerID(payload)
+        await self._handshake_outbound(conn)
         self._endpoint_book[conn.remote_id] = endpoint
         existing = self._connections.get(conn.remote_id)
         if existing is not None and not existing.closed.is_set():
@@ -529,22 +574,17 @@ class P2P:
                             asyncio.open_unix_connection(ep[len(UNIX_SCHEME):], limit=STREAM_BUFFER_LIMIT), timeout=10
                         )
                         conn = _Connection(self, reader, writer)
-                        await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-                        ftype, _, _, payload = await conn.recv_frame()
-                        if ftype != _T_HELLO:
-                            raise P2PDaemonError("expected HELLO")
-                        conn.remote_id = PeerID(payload)
+                        await self._handshake_outbound(conn)
                     else:
                         host, port = split_endpoint(ep)
                         reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
                         conn = _Connection(self, reader, writer)
-                        await conn.send_frame(_T_HELLO, b"", "", self.peer_id.to_bytes())
-                        ftype, _, _, payload = await conn.recv_frame()
-                        if ftype != _T_HELLO:
-                            raise P2PDaemonError("expected HELLO")
-                        conn.remote_id = PeerID(payload)
+                        await self._handshake_outbound(conn)
                     if conn.remote_id != peer_id:
-                        logger.warning(f"peer at {ep} identifies as {conn.remote_id}, expected {peer_id}")
+                        # authenticated identity != who we meant to dial: refuse
+                        # (stale endpoint book / DHT poisoning / MITM attempt)
+                        await conn.close()
+                        raise P2PDaemonError(f"peer at {ep} proved identity {conn.remote_id}, expected {peer_id}")
                     self._endpoint_book[conn.remote_id] = ep
                     self._register_connection(conn)
                     conn.reader_task = asyncio.create_task(self._connection_loop(conn))

@@ -233,3 +233,59 @@ def test_relay_reregisters_after_relay_connection_drop():
             await p.shutdown()
 
     run(main())
+
+
+def test_handshake_rejects_impersonator():
+    """A peer that claims a PeerID it does not hold the key for must be
+    rejected during the handshake (unauthenticated HELLO let
+    any peer claim any identity). Reference gets this from libp2p TLS
+    (p2p_daemon_bindings/datastructures.py:66-88)."""
+
+    async def main():
+        server = await P2P.create()
+        honest = await P2P.create()
+        impostor = await P2P.create()
+        # the impostor claims the honest peer's identity in its HELLO but
+        # cannot sign with the honest peer's key
+        impostor.peer_id = honest.peer_id
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return payload
+
+        server.add_unary_handler("echo", echo)
+        with pytest.raises(Exception):
+            await impostor.call_unary(server.peer_info, "echo", b"x", timeout=5)
+        # the server must not have registered a connection under the forged id
+        assert honest.peer_id not in server._connections
+        # an honest client still works
+        assert await honest.call_unary(server.peer_info, "echo", b"ok", timeout=5) == b"ok"
+        await honest.shutdown()
+        await impostor.shutdown()
+        await server.shutdown()
+
+    run(main())
+
+
+def test_dial_by_id_rejects_wrong_peer():
+    """Dialing PeerInfo(id=X, endpoint) where the endpoint's peer proves a
+    different identity must fail (mismatched dials
+    were only warned about)."""
+    from hivemind_amd.p2p.peer_id import PeerInfo
+    from hivemind_amd.p2p.transport import P2PDaemonError
+
+    async def main():
+        server = await P2P.create()
+        client = await P2P.create()
+        other_id = PeerID.from_identity(PrivateKey())
+        forged = PeerInfo(other_id, server.peer_info.endpoints)
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return payload
+
+        server.add_unary_handler("echo", echo)
+        with pytest.raises(Exception):
+            await client.call_unary(forged, "echo", b"x", timeout=5)
+        await client.shutdown()
+        await server.shutdown()
+
+    run(main())
