@@ -15,6 +15,7 @@ Synthetic data (random token ids of the benchmark shape), random-init weights.
 
 import argparse
 import json
+import math
 import os
 import sys
 import time
@@ -49,6 +50,19 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     n_gpus = world_size if world_size > 1 else args.gpus
     use_gpu = torch.cuda.is_available()
+
+    # The metric is named "hivemind.Optimizer": its cost is the global step
+    # (gradient averaging over xGMI + fused optimizer + epoch transition), so
+    # the timed window must contain complete global steps at every N whatever
+    # step count the driver chose (20x128 samples never
+    # reached target_batch_size 4096 and the bench timed only fwd+bwd).
+    # Epochs fire every ceil(target / (batch * N)) steps; raise the per-GPU
+    # batch until >=2 epoch transitions fit in the timed window.
+    min_batch = math.ceil(args.target_batch_size / max(1, n_gpus * max(args.steps // 2, 1)))
+    if args.batch < min_batch:
+        log(f"raising per-GPU batch {args.batch} -> {min_batch} so {args.steps} "
+            f"timed steps at {n_gpus} GPUs span >=2 global steps (target {args.target_batch_size})")
+        args.batch = min_batch
 
     import torch.distributed as dist
 
@@ -193,11 +207,27 @@ def main():
     sync()
 
     log(f"timing: {args.steps} steps")
+    epoch_before = opt.local_epoch
+    step_times = []
+    epoch_steps = []  # wall time of steps in which an epoch transition landed
     t0 = time.perf_counter()
     for i in range(args.steps):
+        e0, s0 = opt.local_epoch, time.perf_counter()
         one_step()
+        dt = time.perf_counter() - s0
+        step_times.append(dt)
+        if opt.local_epoch != e0:
+            epoch_steps.append(dt)
     sync()
     elapsed = time.perf_counter() - t0
+    epochs_completed = opt.local_epoch - epoch_before
+
+    # every rank must have amortized >=1 full global step, else the headline
+    # metric was not measured -- fail loudly rather than report fwd+bwd only
+    if epochs_completed < 1:
+        log(f"FATAL: timed window contained {epochs_completed} epoch transitions; "
+            f"the metric requires >=1 full global step (averaging + optimizer)")
+        raise RuntimeError("bench timed window contained no global step")
 
     # take the max elapsed across ranks (slowest peer defines the swarm rate);
     # runs on the gloo subgroup so it cannot interleave with RCCL averaging
@@ -226,6 +256,11 @@ def main():
             "vs_baseline": round(samples_per_sec / (BASELINE_SAMPLES_PER_SEC_PER_PEER * n_gpus), 2),
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
+            "epochs_in_timed_window": epochs_completed,
+            "global_step_ms": round(1000.0 * sum(epoch_steps) / len(epoch_steps), 2) if epoch_steps else None,
+            "plain_step_ms": round(
+                1000.0 * (sum(step_times) - sum(epoch_steps)) / max(1, len(step_times) - len(epoch_steps)), 2
+            ),
             "config": {
                 "model": args.model,
                 "dpu": args.dpu,
