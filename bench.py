@@ -43,6 +43,10 @@ def main():
                         help="delayed parameter updates: overlap averaging + optimizer step with compute")
     parser.add_argument("--powersgd-rank", type=int, default=0,
                         help="if > 0, average gradients with rank-r PowerSGD + error feedback (baseline config 3)")
+    parser.add_argument("--grad-compression", type=str, default="bf16",
+                        choices=["none", "bf16", "int8"],
+                        help="gradient wire format on the RCCL/xGMI plane: bf16 cast (default) or "
+                             "blockwise-int8 quantized butterfly (baseline config 2)")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -160,7 +164,8 @@ def main():
         matchmaking_time=1.0 if world_size > 1 else 0.5,
         averaging_timeout=120.0,
         reuse_grad_buffers=False,
-        grad_rccl_wire_dtype=torch.bfloat16,  # halve xGMI bytes for grad all-reduce
+        grad_rccl_wire_dtype=torch.bfloat16 if args.grad_compression == "bf16" else None,
+        grad_rccl_compression="blockwise_int8" if args.grad_compression == "int8" else None,
         grad_averager_factory=(
             (lambda **kw: __import__("hivemind_amd.optim.power_sgd_averager", fromlist=["PowerSGDGradientAverager"])
              .PowerSGDGradientAverager(averager_rank=args.powersgd_rank, **kw))
@@ -265,6 +270,7 @@ def main():
                 "model": args.model,
                 "dpu": args.dpu,
                 "powersgd_rank": args.powersgd_rank or None,
+                "grad_compression": args.grad_compression,
                 "global_batch": args.target_batch_size,
                 "per_gpu_batch": args.batch,
                 "seq_len": args.seq_len,

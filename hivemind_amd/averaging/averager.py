@@ -61,7 +61,14 @@ from .key_manager import GroupKeyManager
 from .load_balancing import load_balance_peers
 from .matchmaking import JoinRequest, Matchmaking, MatchmakingException, MessageFromLeader
 from .partition import DEFAULT_PART_SIZE_BYTES
-from .rccl import DistributedAllReduceRunner, distributed_world_info, group_matches_world
+from .rccl import (
+    DistributedAllReduceRunner,
+    distributed_world_info,
+    get_process_group_for_ranks,
+    group_matches_world,
+    group_world_ranks,
+    release_collective_ticket,
+)
 
 logger = get_logger(__name__)
 
@@ -79,6 +86,7 @@ class GroupMetadata:
     user_gathered: dict
     dist_infos: list
     weights: list
+    rccl_ticket: Optional[int] = None  # launch-order slot for the RCCL plane
 
 
 class _null_actx:
@@ -137,6 +145,7 @@ class DecentralizedAverager(ServicerBase):
         allow_state_sharing: Optional[bool] = None,
         declare_state_period: float = 30.0,
         allreduce_wire_dtype: Optional[torch.dtype] = None,
+        allreduce_codec: Optional[str] = None,
         use_rccl_when_available: bool = True,
         shutdown_timeout: float = 5.0,
     ):
@@ -159,6 +168,7 @@ class DecentralizedAverager(ServicerBase):
         self.reducer_timeout = reducer_timeout if reducer_timeout is not None else (next_chunk_timeout or 15.0)
         self.compression, self.state_compression = compression, state_compression
         self.allreduce_wire_dtype = allreduce_wire_dtype
+        self.allreduce_codec = allreduce_codec
         self.use_rccl_when_available = use_rccl_when_available
         self.shutdown_timeout = shutdown_timeout
         self.declare_state_period = declare_state_period
@@ -184,14 +194,13 @@ class DecentralizedAverager(ServicerBase):
         self._inflight_futures: set = set()
         self._running_groups: Dict[bytes, asyncio.Future] = {}
         self.last_data_plane: Optional[str] = None  # "rccl" | "rpc" after a round
-        # dedicated communicator: concurrent averagers (grad + state) must never
-        # interleave collectives on one communicator -- cross-rank ordering of
-        # threads is not deterministic (SURVEY.md §7 "hard parts"). new_group()
-        # is collective, so every rank must construct its averagers in the same
-        # order (true for Optimizer and for symmetric training scripts).
-        import torch.distributed as dist
-
-        self._dist_process_group = dist.new_group() if (dist.is_available() and dist.is_initialized()) else None
+        # Communicators are created LAZILY per matched rank-set via
+        # rccl.get_process_group_for_ranks (use_local_synchronization=True):
+        # only group members rendezvous, so ranks that never average (or only
+        # average in other groups) cannot hang everyone in a collective
+        # new_group(). Cross-averager interleaving on a shared communicator is
+        # prevented by the CollectiveSequencer tickets + process collective
+        # lock (rccl.py).
         self._state_updated = asyncio.Event()
         self._declare_state_task: Optional[asyncio.Task] = None
         self._ready = concurrent.futures.Future()
@@ -299,6 +308,7 @@ class DecentralizedAverager(ServicerBase):
         allow_retries: bool = True,
         require_trigger: bool = False,
         wait: bool = True,
+        rccl_ticket: Optional[int] = None,
     ) -> Union[Optional[Dict[PeerID, Any]], StepControl]:
         """Look for a group and average tensors with it (reference averager.py:367-419).
 
@@ -317,6 +327,7 @@ class DecentralizedAverager(ServicerBase):
             weight=weight,
             gather_binary=gather_binary,
         )
+        control.rccl_ticket = rccl_ticket
         if not require_trigger:
             control.allow_allreduce()
         logger.debug(f"{self.prefix}@{self.peer_id}: step() created control {id(control):#x} trig={control.triggered}")
@@ -351,7 +362,8 @@ class DecentralizedAverager(ServicerBase):
                     logger.debug(f"{self.prefix}@{self.peer_id}: trigger received")
                     step.stage = AveragingStage.RUNNING_ALLREDUCE
                     gathered = await asyncio.wait_for(
-                        self._aggregate_with_group(group_info, step.weight), timeout=self.allreduce_timeout
+                        self._aggregate_with_group(group_info, step.weight, ticket=step.rccl_ticket),
+                        timeout=self.allreduce_timeout,
                     )
                     step.set_result(gathered)
                 except (
@@ -371,6 +383,11 @@ class DecentralizedAverager(ServicerBase):
             raise
         except BaseException as e:
             step.set_exception(e if isinstance(e, Exception) else AveragingError(repr(e)))
+        finally:
+            # free the launch-order slot whether the round ran, failed, or was
+            # abandoned -- otherwise every later RCCL round on this rank stalls
+            release_collective_ticket(step.rccl_ticket)
+            step.rccl_ticket = None
 
     def _parse_group_metadata(self, group_info: GroupInfo, weight: float) -> "GroupMetadata":
         """Unpack per-peer gathered metadata: [bandwidth, mode, user_gather, dist_info, weight]."""
@@ -392,12 +409,13 @@ class DecentralizedAverager(ServicerBase):
             weights=weights,
         )
 
-    def _can_use_rccl(self, meta: "GroupMetadata") -> bool:
-        return (
-            self.use_rccl_when_available
-            and group_matches_world(meta.dist_infos)
-            and all(m == AveragingMode.NODE for m in meta.modes)
-        )
+    def _rccl_group_ranks(self, meta: "GroupMetadata") -> Optional[list]:
+        """World ranks of the matched group if the whole group lives in our
+        torch.distributed world (any subset -- Moshpit subgroups of the node
+        included), else None -> RPC butterfly."""
+        if not self.use_rccl_when_available or not all(m == AveragingMode.NODE for m in meta.modes):
+            return None
+        return group_world_ranks(meta.dist_infos)
 
     async def _average_tensors_with_group(
         self,
@@ -413,17 +431,22 @@ class DecentralizedAverager(ServicerBase):
         world, else the RPC butterfly. Subclasses (PowerSGD) call this several
         times per round with distinct group_id suffixes."""
         lock_ctx = enter_asynchronously(self.lock_averaged_tensors) if take_lock else _null_actx()
-        if self._can_use_rccl(meta):
+        rccl_ranks = self._rccl_group_ranks(meta)
+        if rccl_ranks is not None:
             self.last_data_plane = "rccl"
+            # full node and Moshpit subgroups alike: cached member-only communicator
+            process_group = get_process_group_for_ranks(rccl_ranks)
             async with lock_ctx:
                 await asyncio.get_event_loop().run_in_executor(
                     None,
                     DistributedAllReduceRunner(
                         tensors,
                         weight,
-                        process_group=self._dist_process_group,
+                        process_group=process_group,
                         wire_dtype=self.allreduce_wire_dtype,
+                        codec=self.allreduce_codec,
                         averaging_alpha=self.averaging_alpha,
+                        ticket=meta.rccl_ticket,
                     ).run,
                 )
             return
@@ -440,11 +463,14 @@ class DecentralizedAverager(ServicerBase):
                 tensors, group_info, group_id=group_id, peer_fractions=peer_fractions, weight=weight, modes=meta.modes
             )
 
-    async def _aggregate_with_group(self, group_info: GroupInfo, weight: float) -> Dict[PeerID, Any]:
+    async def _aggregate_with_group(
+        self, group_info: GroupInfo, weight: float, ticket: Optional[int] = None
+    ) -> Dict[PeerID, Any]:
         """Run the data plane for one assembled group (reference averager.py:514-562)."""
         logger.debug(f"{self.prefix}@{self.peer_id}: entering aggregation, group={group_info.group_id.hex()[:8]}")
         try:
             meta = self._parse_group_metadata(group_info, weight)
+            meta.rccl_ticket = ticket
             await self._aggregate_tensors_with_group(group_info, weight, meta)
             self._state_updated.set()
             return meta.user_gathered

@@ -39,6 +39,9 @@ class StepControl:
         self._began_allreduce = False
         self._trigger = concurrent.futures.Future()  # set by allow_allreduce()
         self._result: concurrent.futures.Future = concurrent.futures.Future()
+        # launch-order ticket for the RCCL data plane (rccl.CollectiveSequencer);
+        # issued by the optimizer at scheduling time, released when the round ends
+        self.rccl_ticket = None
 
     # ------------------------------------------------------------ attributes
 

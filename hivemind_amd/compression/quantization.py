@@ -90,6 +90,14 @@ class Uniform8BitQuantization(Quantization):
         std_unbiased = centered.norm() / math.sqrt(max(centered.numel() - 1, 1))
         scale = self.RANGE_IN_SIGMAS * std_unbiased / self.n_bins
         scale = torch.clamp_min(scale, torch.finfo(torch.float32).eps)
+        if tensor.is_cuda:
+            # on-device arithmetic equivalent of torch.quantize_per_tensor
+            # (which is CPU-oriented); bucket means reduce on the GPU and only
+            # the uint8 codes + 256-entry codebook cross PCIe
+            quantized = torch.clamp(torch.round(centered / scale) + offset, 0, self.n_bins - 1).to(torch.uint8)
+            lookup = average_buckets(centered, quantized, self.n_bins)
+            codebook = (lookup + shift).cpu().numpy().astype(self.codebook_dtype)
+            return quantized.cpu().numpy().astype(self.indices_dtype), codebook
         quantized = torch.quantize_per_tensor(centered, float(scale), offset, torch.quint8).int_repr()
         lookup = average_buckets(centered, quantized, self.n_bins)
         codebook = (lookup + shift).numpy().astype(self.codebook_dtype)
@@ -104,6 +112,8 @@ class Quantile8BitQuantization(Quantization):
 
     def quantize(self, tensor: torch.Tensor, allow_inplace: bool = False) -> Tuple[np.ndarray, np.ndarray]:
         tensor = tensor.detach().flatten()
+        if tensor.is_cuda:
+            tensor = tensor.cpu()  # np.quantile-based; one transfer, then CPU math
         codebook = quantile_qq_approximation(tensor.numpy(), self.n_bins)
         borders = (codebook[:-1] + codebook[1:]) / 2
         quantized = np.digitize(tensor.numpy(), borders).astype(self.indices_dtype)
@@ -118,11 +128,13 @@ class Quantile8BitQuantization(Quantization):
 
 
 def average_buckets(tensor: torch.Tensor, quant_weight: torch.Tensor, n_bins: int) -> torch.Tensor:
-    """Per-bucket mean of original values (reference quantization.py:88-94)."""
-    bin_sums = torch.zeros(n_bins, dtype=torch.float32).scatter_add_(
-        0, quant_weight.flatten().long(), tensor.flatten().to(torch.float32)
+    """Per-bucket mean of original values (reference quantization.py:88-94).
+    Runs on whichever device the tensor lives on."""
+    codes = quant_weight.flatten().long()
+    bin_sums = torch.zeros(n_bins, dtype=torch.float32, device=tensor.device).scatter_add_(
+        0, codes, tensor.flatten().to(torch.float32)
     )
-    bin_counts = torch.clamp_min_(torch.bincount(quant_weight.flatten().long(), minlength=n_bins), 1)
+    bin_counts = torch.clamp_min_(torch.bincount(codes, minlength=n_bins), 1)
     return bin_sums / bin_counts
 
 
@@ -154,6 +166,19 @@ class BlockwiseQuantization(Quantization):
     blocksize = BLOCKSIZE
 
     def quantize(self, tensor: torch.Tensor, allow_inplace: bool = False) -> Tuple[np.ndarray, np.ndarray]:
+        if tensor.is_cuda:
+            # the hand-written CDNA4 kernel (ops/hip/elementwise.hip
+            # quantize_blockwise_int8) computes the same per-4096-block
+            # absmax/127 codes; pad to the wire's block-aligned layout
+            from ..ops import quantize_blockwise as hip_quantize_blockwise
+
+            q, absmax = hip_quantize_blockwise(tensor.detach().contiguous())
+            num_blocks = absmax.numel()
+            if q.numel() < num_blocks * self.blocksize:
+                padded_q = torch.zeros(num_blocks * self.blocksize, dtype=torch.int8, device=q.device)
+                padded_q[: q.numel()] = q
+                q = padded_q
+            return q.cpu().numpy().view(np.uint8), absmax.cpu().numpy().astype(np.float32)
         flat = tensor.detach().to(torch.float32).flatten()
         n = flat.numel()
         num_blocks = (n + self.blocksize - 1) // self.blocksize

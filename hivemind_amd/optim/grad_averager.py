@@ -132,6 +132,15 @@ class GradientAverager(DecentralizedAverager):
         control.weight = float(self.local_samples_accumulated) if weight is None else weight
         if reset_accumulators:
             self.reset_accumulated_grads_()
+        if control.rccl_ticket is None:
+            # reserve the RCCL launch slot at trigger time: step() runs in the
+            # optimizer's foreground epoch logic, whose order is identical on
+            # every rank (see rccl.CollectiveSequencer)
+            from ..averaging.rccl import issue_collective_ticket, release_collective_ticket
+
+            control.rccl_ticket = issue_collective_ticket()
+            if control.done():
+                release_collective_ticket(control.rccl_ticket)
         control.allow_allreduce()
         logger.debug(f"grad_averager: triggered control {id(control):#x}")
         return control.result(timeout) if wait else control

@@ -69,6 +69,7 @@ class Optimizer(torch.optim.Optimizer):
         average_state_every: int = 1,
         use_local_updates: bool = False,
         grad_rccl_wire_dtype=None,
+        grad_rccl_compression: Optional[str] = None,
         client_mode: bool = False,
         auxiliary: bool = False,
         grad_compression: CompressionBase = NoCompression(),
@@ -137,6 +138,10 @@ class Optimizer(torch.optim.Optimizer):
                 # baseline config compresses gradient averaging; state
                 # averaging stays full precision)
                 grad_opts.setdefault("allreduce_wire_dtype", grad_rccl_wire_dtype)
+            if grad_rccl_compression is not None:
+                # "blockwise_int8": quantized direct-send butterfly over xGMI
+                # (BASELINE config 2 -- blockwise-quantized gradient averaging)
+                grad_opts.setdefault("allreduce_codec", grad_rccl_compression)
             self.grad_averager: Optional[GradientAverager] = self._make_gradient_averager(
                 grad_averager_factory, reuse_grad_buffers=reuse_grad_buffers, compression=grad_compression,
                 **grad_opts,

@@ -219,6 +219,22 @@ class TrainingStateAverager(DecentralizedAverager):
         if increment_epoch:
             self.local_epoch += 1
 
+        if averaging_round:
+            # reserve the RCCL launch slot HERE, in the foreground: every rank
+            # schedules its rounds in the same program order, so tickets give a
+            # consistent cross-rank launch order even when the round itself
+            # runs on the background executor
+            from ..averaging.rccl import issue_collective_ticket, release_collective_ticket
+
+            ticket = issue_collective_ticket()
+            if averaging_control is not None:
+                averaging_control.rccl_ticket = ticket
+                if averaging_control.done():  # round already over: don't stall later rounds
+                    release_collective_ticket(ticket)
+            else:
+                averaging_opts = dict(averaging_opts or {})
+                averaging_opts["rccl_ticket"] = ticket
+
         if optimizer_step or zero_grad or averaging_round:
             task = lambda: self._do(
                 wait_for_trigger, optimizer_step, zero_grad, averaging_round, averaging_control, grad_scaler,
