@@ -27,7 +27,7 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..ops import fused_bias_gelu, fused_cross_entropy, fused_layernorm
+from ..ops import flash_attention, flash_attention_usable, fused_bias_gelu, fused_cross_entropy, fused_layernorm
 
 
 @dataclass
@@ -94,7 +94,10 @@ class AlbertLayer(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.num_heads, self.head_dim)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # [B, heads, S, hd]
-        attn = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_mask)
+        if attn_mask is None and flash_attention_usable(q, k):
+            attn = flash_attention(q, k, v)  # hand-written CDNA4 kernels
+        else:
+            attn = F.scaled_dot_product_attention(q, k, v, attn_mask=attn_mask)
         attn = attn.transpose(1, 2).reshape(B, S, H)
         x = self.attn_norm(self.attn_out(attn), residual=x)
         up = F.linear(x, self.ffn_up_weight)  # rocBLAS GEMM, no bias

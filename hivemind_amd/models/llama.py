@@ -92,7 +92,18 @@ class LlamaDecoderLayer(nn.Module):
         q = fused_rope(q, cos_table, sin_table)
         k = fused_rope(k, cos_table, sin_table)
         q, k, v = (t.transpose(1, 2) for t in (q, k, v))  # [B, heads, S, hd]
-        attn = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
+        from ..ops import flash_attention, flash_attention_usable
+
+        if flash_attention_usable(q, k):
+            # CDNA4 flash kernels; backward needs H == Hkv, so expand grouped
+            # kv heads (autograd sum-reduces dk/dv back through the repeat)
+            groups = self.num_heads // self.num_kv_heads
+            if groups > 1:
+                k = k.repeat_interleave(groups, dim=1)
+                v = v.repeat_interleave(groups, dim=1)
+            attn = flash_attention(q, k, v, causal=True)
+        else:
+            attn = F.scaled_dot_product_attention(q, k, v, is_causal=True, enable_gqa=True)
         attn = attn.transpose(1, 2).reshape(B, S, H)
         x = residual + self.o_proj(attn)
         residual = x

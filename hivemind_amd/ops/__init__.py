@@ -314,6 +314,55 @@ def build_rope_tables(seq_len: int, head_dim: int, base: float = 500000.0, devic
 
 
 # ---------------------------------------------------------------------------
+# flash attention: hand-written CDNA4 kernels replacing AOTriton
+# ---------------------------------------------------------------------------
+
+
+class _FlashAttention(torch.autograd.Function):
+    """bf16 flash attention on the hand-written CDNA4 kernels
+    (ops/hip/flash_attention.hip). Forward saves O + logsumexp; backward runs
+    the delta pre-pass and the two-kernel (dkdv, dq) split."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        o, lse = hip_ops().flash_attn_fwd(q, k, v, causal, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.causal, ctx.scale = causal, scale
+        return o
+
+    @staticmethod
+    def backward(ctx, d_out):
+        q, k, v, o, lse = ctx.saved_tensors
+        dq, dk, dv = hip_ops().flash_attn_bwd(d_out, q, k, v, o, lse, ctx.causal, ctx.scale)
+        return dq, dk, dv, None, None
+
+
+def flash_attention_usable(q: torch.Tensor, k: torch.Tensor) -> bool:
+    """True if the CDNA4 flash kernels handle these shapes: CUDA bf16
+    [B, H, S, D] with D in {64, 128} and S % 64 == 0."""
+    return (
+        q.is_cuda
+        and hip_available()
+        and q.dtype == torch.bfloat16
+        and q.shape[-1] in (64, 128)
+        and q.shape[-2] % 64 == 0
+        and k.shape[-2] == q.shape[-2]
+    )
+
+
+def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+                    causal: bool = False, scale: Optional[float] = None) -> torch.Tensor:
+    """[B, H, S, D] attention. CUDA tensors run the hand-written CDNA4 flash
+    kernels; CPU tensors use the torch reference (the numerics oracle).
+    Backward requires H == Hkv -- expand grouped kv heads first."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if q.is_cuda:
+        return _FlashAttention.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal, float(scale))
+    return torch.nn.functional.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale)
+
+
+# ---------------------------------------------------------------------------
 # averaging primitives
 # ---------------------------------------------------------------------------
 

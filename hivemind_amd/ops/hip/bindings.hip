@@ -12,6 +12,15 @@
 extern "C" void launch_gemm_bt_bf16(const void*, const void*, const void*, void*, void*, int, int, int, int, void*);
 extern "C" void launch_mfma_probe(const void*, const void*, void*, void*);
 extern "C" void launch_stage_probe(const void*, void*, int, void*);
+extern "C" void launch_flash_fwd(const void*, const void*, const void*, void*, void*,
+                                 int, int, int, int, int, float, int, void*);
+extern "C" void launch_flash_delta(const void*, const void*, void*, long long, int, void*);
+extern "C" void launch_flash_bwd_dkdv(const void*, const void*, const void*, const void*,
+                                      const void*, const void*, void*, void*,
+                                      int, int, int, int, float, int, void*);
+extern "C" void launch_flash_bwd_dq(const void*, const void*, const void*, const void*,
+                                    const void*, const void*, void*,
+                                    int, int, int, int, float, int, void*);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -459,6 +468,52 @@ torch::Tensor stage_probe(torch::Tensor A) {
   return out;
 }
 
+// ------------------------------------------------------------- attention
+
+static void check_flash_shapes(const torch::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() && t.scalar_type() == torch::kBFloat16,
+              name, " must be a contiguous CUDA bf16 tensor");
+  TORCH_CHECK(t.dim() == 4, name, " must be [B, H, S, D]");
+}
+
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                                          bool causal, double scale) {
+  check_flash_shapes(q, "q"); check_flash_shapes(k, "k"); check_flash_shapes(v, "v");
+  int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  int Hkv = k.size(1);
+  TORCH_CHECK(D == 64 || D == 128, "flash attention requires head_dim 64 or 128, got ", D);
+  TORCH_CHECK(S % 64 == 0, "flash attention requires seq_len % 64 == 0, got ", S);
+  TORCH_CHECK(k.size(2) == S && v.size(2) == S, "kv seq_len mismatch");
+  TORCH_CHECK(H % Hkv == 0, "H must be a multiple of Hkv");
+  auto o = torch::empty_like(q);
+  auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
+  launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), lse.data_ptr(),
+                   B, H, Hkv, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
+  return {o, lse};
+}
+
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor d_out, torch::Tensor q, torch::Tensor k,
+                                          torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+                                          bool causal, double scale) {
+  check_flash_shapes(q, "q"); check_flash_shapes(d_out, "d_out");
+  int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
+  TORCH_CHECK(k.size(1) == H, "flash attention backward requires H == Hkv (expand kv first)");
+  auto d_out_c = d_out.contiguous();
+  auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
+  launch_flash_delta(d_out_c.data_ptr(), o.data_ptr(), delta.data_ptr(),
+                     (long long)B * H * S, D, (void*)current_stream());
+  auto dq = torch::empty_like(q);
+  auto dk = torch::empty_like(k);
+  auto dv = torch::empty_like(v);
+  launch_flash_bwd_dkdv(d_out_c.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                        lse.data_ptr(), delta.data_ptr(), dk.data_ptr(), dv.data_ptr(),
+                        B, H, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
+  launch_flash_bwd_dq(d_out_c.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+                      lse.data_ptr(), delta.data_ptr(), dq.data_ptr(),
+                      B, H, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
+  return {dq, dk, dv};
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("stage_probe", &stage_probe, "gload_lds staging probe");
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 operand layout probe");
@@ -482,4 +537,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bias_gelu_bwd_xb", &bias_gelu_bwd_xb, "backward of bias+gelu (recomputes x+bias)");
   m.def("layernorm_fwd", &layernorm_fwd, "fused (residual+)layernorm forward");
   m.def("layernorm_bwd", &layernorm_bwd, "layernorm backward");
+  m.def("flash_attn_fwd", &flash_attn_fwd, "CDNA4 flash attention forward -> (O, logsumexp)");
+  m.def("flash_attn_bwd", &flash_attn_bwd, "CDNA4 flash attention backward -> (dQ, dK, dV)");
 }

@@ -1,0 +1,549 @@
+// Hand-written CDNA4 flash attention (forward + backward), bf16, D in {64,128}.
+//
+// Replaces AOTriton's sdpa kernels on the ALBERT/Llama hot path -- round-1
+// profiling showed bwd_kernel_dk_dv + bwd_kernel_dq at 18.6% of step time
+// (profiles/albert_kernels.md), the largest single block. Reference hot spot
+// K10 (/root/reference/hivemind/moe/server/layers/common.py:33-80
+// is the reference's only attention code -- single-device, no kernels).
+//
+// Design (cdna_hip_programming.md §B "fused attention prefill" + §5/§6):
+// * tiles of 64 q-rows x 64 kv-rows; one workgroup = 4 waves, each wave owns
+//   16 rows of the output tile; grid = (S/64, B*H) >> 256 CUs.
+// * v_mfma_f32_16x16x32_bf16 with the layouts verified in round 1
+//   (mfma_gemm_impl.hip probe): A/B operand lane l holds row (l&15),
+//   k-segment (l>>4)*8; C/D holds col (l&15), row (l>>4)*4+reg.
+// * all LDS tiles are reg-staged with the guide's XOR swizzle
+//   byte ^= ((row&7)<<4) -- rows of 128/256 B would otherwise put all 16
+//   lanes of a fragment read in one bank (32-way conflict, Guideline 4).
+// * online softmax entirely in registers: row max/sum via 16-lane
+//   __shfl_xor butterflies (wave-parallel softmax, common-mistake #6);
+//   row statistics end up replicated across each 16-lane group, so the
+//   O-rescale factor needs no broadcast at all.
+// * scale (1/sqrt(D)) is baked into the Q (fwd, bwd-dq) / K (bwd-dkdv)
+//   fragments at load time; the softmax-jacobian scale is baked into the
+//   written dS tile, so no per-element multiplies survive in inner loops.
+// * backward follows the two-pass split AOTriton/flash2 use: dkdv
+//   (parallel over kv tiles, recomputes P^T from the saved logsumexp) and
+//   dq (parallel over q tiles), after a delta = rowsum(dO*O) pre-pass.
+//
+// Constraints (enforced by the binding): S % 64 == 0, D in {64, 128},
+// contiguous [B, H, S, D] bf16 tensors; GQA (Hkv < H) supported in forward,
+// backward requires H == Hkv (ALBERT; Llama uses repeat_kv before sdpa).
+
+#include <hip/hip_runtime.h>
+#include <cstdint>
+
+typedef unsigned short ushort_t;
+typedef __attribute__((ext_vector_type(8))) short bf16x8;
+typedef __attribute__((ext_vector_type(4))) float f32x4;
+
+__device__ __forceinline__ float bf2f(ushort_t u) {
+  return __uint_as_float(((unsigned int)u) << 16);
+}
+__device__ __forceinline__ ushort_t f2bf(float f) {
+  unsigned int w = __float_as_uint(f);
+  unsigned int bias = 0x7FFF + ((w >> 16) & 1);
+  return (ushort_t)((w + bias) >> 16);
+}
+
+constexpr int TILE = 64;     // q-rows and kv-rows per workgroup tile
+constexpr int NTHREADS = 256;  // 4 waves
+
+// ---- swizzled LDS addressing -------------------------------------------
+// Tiles are [rows][COLS] bf16 with COLS in {64, 128}. Byte offset XOR
+// ((row&7)<<4) spreads the 16 rows of a fragment read over 8 distinct 16B
+// slots: 2 lanes/bank, which is free (MI355X_MICROARCH m136).
+
+__device__ __forceinline__ int swz_off(int row, int col, int cols) {
+  return ((row * cols + col) * 2) ^ ((row & 7) << 4);
+}
+
+__device__ __forceinline__ bf16x8 lds_read8(const ushort_t* lds, int row, int col, int cols) {
+  return *reinterpret_cast<const bf16x8*>(reinterpret_cast<const char*>(lds) + swz_off(row, col, cols));
+}
+
+__device__ __forceinline__ void lds_write8(ushort_t* lds, int row, int col, int cols, bf16x8 v) {
+  *reinterpret_cast<bf16x8*>(reinterpret_cast<char*>(lds) + swz_off(row, col, cols)) = v;
+}
+
+__device__ __forceinline__ void lds_write1(ushort_t* lds, int row, int col, int cols, ushort_t v) {
+  *reinterpret_cast<ushort_t*>(reinterpret_cast<char*>(lds) + swz_off(row, col, cols)) = v;
+}
+
+// stage a [TILE][D] bf16 tile from global (row-major, row stride D) into
+// swizzled LDS; 16B per thread per step, coalesced.
+template <int D>
+__device__ __forceinline__ void stage_tile(ushort_t* lds, const ushort_t* src) {
+  constexpr int SEGS = D / 8;               // 16B segments per row
+  constexpr int TOTAL = TILE * SEGS;
+  for (int idx = threadIdx.x; idx < TOTAL; idx += NTHREADS) {
+    int row = idx / SEGS, seg = idx % SEGS;
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(src + (long long)row * D + seg * 8);
+    lds_write8(lds, row, seg * 8, D, v);
+  }
+}
+
+// stage a [TILE][D] tile TRANSPOSED into [D][TILE] swizzled LDS
+// (dst[d][s] = src[s][d]); vectorized 16B global reads, scalar LDS writes.
+template <int D>
+__device__ __forceinline__ void stage_tile_t(ushort_t* lds, const ushort_t* src) {
+  constexpr int SEGS = D / 8;
+  constexpr int TOTAL = TILE * SEGS;
+  for (int idx = threadIdx.x; idx < TOTAL; idx += NTHREADS) {
+    int s = idx / SEGS, seg = idx % SEGS;
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(src + (long long)s * D + seg * 8);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds_write1(lds, seg * 8 + j, s, TILE, (ushort_t)v[j]);
+  }
+}
+
+// row-group butterfly: combine over the 16 lanes that share (lane>>4)
+__device__ __forceinline__ float rowmax16(float v) {
+#pragma unroll
+  for (int m = 1; m < 16; m <<= 1) v = fmaxf(v, __shfl_xor(v, m, 64));
+  return v;
+}
+__device__ __forceinline__ float rowsum16(float v) {
+#pragma unroll
+  for (int m = 1; m < 16; m <<= 1) v += __shfl_xor(v, m, 64);
+  return v;
+}
+
+// ======================================================================
+// forward
+// ======================================================================
+// grid (S/64, B*H); saves O and logsumexp (LSE = m + log(l), natural units
+// of the scaled scores).
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
+    const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
+    const ushort_t* __restrict__ V, ushort_t* __restrict__ O,
+    float* __restrict__ LSE, int B, int H, int Hkv, int S, float scale) {
+  constexpr int KSTEPS = D / 32;   // MFMA K-steps over the head dim
+  constexpr int DFRAGS = D / 16;   // output d-blocks per wave
+  __shared__ ushort_t k_l[TILE * D];
+  __shared__ ushort_t vt_l[D * TILE];
+  __shared__ ushort_t p_l[4][16 * TILE];  // per-wave P tile [16 q][64 s]
+
+  const int q_tile = blockIdx.x;
+  const long long bh = blockIdx.y;
+  const long long b = bh / H, h = bh % H;
+  const long long q_base = (bh * S) * D;
+  const long long kv_base = ((b * Hkv + h / (H / Hkv)) * S) * D;
+
+  const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+  const int q0 = q_tile * TILE;            // tile's first q row
+  const int wq = wave * 16;                // wave's q rows within the tile
+  const int fr = lane & 15, fq = lane >> 4;
+
+  // Q fragments in registers for the whole kernel, pre-scaled (Q-hoist)
+  bf16x8 q_frag[KSTEPS];
+  {
+    const ushort_t* qrow = Q + q_base + (long long)(q0 + wq + fr) * D;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      bf16x8 raw = *reinterpret_cast<const bf16x8*>(qrow + ks * 32 + fq * 8);
+      bf16x8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = (short)f2bf(bf2f((ushort_t)raw[j]) * scale);
+      q_frag[ks] = out;
+    }
+  }
+
+  f32x4 o_acc[DFRAGS];
+#pragma unroll
+  for (int i = 0; i < DFRAGS; ++i) o_acc[i] = {0.f, 0.f, 0.f, 0.f};
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+
+  const int s_end = CAUSAL ? (q0 + TILE) : S;
+  for (int s0 = 0; s0 < s_end; s0 += TILE) {
+    __syncthreads();
+    stage_tile<D>(k_l, K + kv_base + (long long)s0 * D);
+    stage_tile_t<D>(vt_l, V + kv_base + (long long)s0 * D);
+    __syncthreads();
+
+    // S-tile: wave computes [16 q][64 s] as 4 col-block fragments
+    f32x4 sacc[4];
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      sacc[ns] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bf16x8 kf = lds_read8(k_l, ns * 16 + fr, ks * 32 + fq * 8, D);
+        sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, sacc[ns], 0, 0, 0);
+      }
+    }
+    if (CAUSAL && s0 + TILE > q0) {
+#pragma unroll
+      for (int ns = 0; ns < 4; ++ns)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          int qg = q0 + wq + fq * 4 + r, sg = s0 + ns * 16 + fr;
+          if (sg > qg) sacc[ns][r] = -1e30f;
+        }
+    }
+
+    // online softmax: row stats replicated across each 16-lane group
+    float p[4][4];  // [ns][reg]
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float pm = fmaxf(fmaxf(sacc[0][r], sacc[1][r]), fmaxf(sacc[2][r], sacc[3][r]));
+      pm = rowmax16(pm);
+      float m_new = fmaxf(m_run[r], pm);
+      float corr = __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+      float psum = 0.f;
+#pragma unroll
+      for (int ns = 0; ns < 4; ++ns) {
+        float e = (sacc[ns][r] <= -1e29f) ? 0.f : __expf(sacc[ns][r] - m_new);
+        p[ns][r] = e;
+        psum += e;
+      }
+      l_run[r] = l_run[r] * corr + rowsum16(psum);
+#pragma unroll
+      for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr;
+    }
+
+    // P -> wave-private LDS (bf16), then O += P @ V^T
+    ushort_t* pw = p_l[wave];
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lds_write1(pw, fq * 4 + r, ns * 16 + fr, TILE, f2bf(p[ns][r]));
+    // wave-private region: in-wave ds ordering suffices, no barrier
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd) {
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {  // 64 s = 2 MFMA k-steps
+        bf16x8 pa = lds_read8(pw, fr, ks * 32 + fq * 8, TILE);
+        bf16x8 vb = lds_read8(vt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        o_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[nd], 0, 0, 0);
+      }
+    }
+  }
+
+  // epilogue: O /= l, save LSE
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qg = q0 + wq + fq * 4 + r;
+    float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd)
+      O[q_base + (long long)qg * D + nd * 16 + fr] = f2bf(o_acc[nd][r] * inv_l);
+    if (fr == 0) LSE[bh * S + qg] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
+  }
+}
+
+// ======================================================================
+// delta pre-pass: delta[b,h,s] = sum_d dO * O   (one wave per row)
+// ======================================================================
+
+template <int D>
+__global__ __launch_bounds__(NTHREADS) void flash_delta_kernel(
+    const ushort_t* __restrict__ dO, const ushort_t* __restrict__ O,
+    float* __restrict__ delta, long long rows) {
+  long long row0 = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  const int lane = threadIdx.x & 63;
+  for (long long row = row0; row < rows; row += (long long)gridDim.x * 4) {
+    float acc = 0.f;
+#pragma unroll
+    for (int d = lane; d < D; d += 64)
+      acc += bf2f(dO[row * D + d]) * bf2f(O[row * D + d]);
+#pragma unroll
+    for (int m = 1; m < 64; m <<= 1) acc += __shfl_xor(acc, m, 64);
+    if (lane == 0) delta[row] = acc;
+  }
+}
+
+// ======================================================================
+// backward pass 1: dK, dV   (grid over kv tiles; recomputes P^T)
+// ======================================================================
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
+    const ushort_t* __restrict__ dO, const ushort_t* __restrict__ Q,
+    const ushort_t* __restrict__ K, const ushort_t* __restrict__ V,
+    const float* __restrict__ LSE, const float* __restrict__ delta,
+    ushort_t* __restrict__ dK, ushort_t* __restrict__ dV,
+    int B, int H, int S, float scale) {
+  constexpr int KSTEPS = D / 32;
+  constexpr int DFRAGS = D / 16;
+  __shared__ ushort_t q_l[TILE * D];    // Q tile, natural orientation
+  __shared__ ushort_t qt_l[D * TILE];   // Q tile, transposed
+  __shared__ ushort_t do_l[TILE * D];
+  __shared__ ushort_t dot_l[D * TILE];
+  __shared__ ushort_t t_l[TILE * TILE];  // P^T then dS^T (wave-private rows)
+  __shared__ float lse_l[TILE], dlt_l[TILE];
+
+  const int kv_tile = blockIdx.x;
+  const long long bh = blockIdx.y;
+  const long long base = (bh * S) * D;
+
+  const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+  const int s0 = kv_tile * TILE;
+  const int ws = wave * 16;  // wave's kv rows within the tile
+  const int fr = lane & 15, fq = lane >> 4;
+
+  // K (pre-scaled) and V fragments in registers for the whole kernel
+  bf16x8 k_frag[KSTEPS], v_frag[KSTEPS];
+  {
+    const ushort_t* krow = K + base + (long long)(s0 + ws + fr) * D;
+    const ushort_t* vrow = V + base + (long long)(s0 + ws + fr) * D;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      bf16x8 raw = *reinterpret_cast<const bf16x8*>(krow + ks * 32 + fq * 8);
+      bf16x8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = (short)f2bf(bf2f((ushort_t)raw[j]) * scale);
+      k_frag[ks] = out;
+      v_frag[ks] = *reinterpret_cast<const bf16x8*>(vrow + ks * 32 + fq * 8);
+    }
+  }
+
+  f32x4 dv_acc[DFRAGS], dk_acc[DFRAGS];
+#pragma unroll
+  for (int i = 0; i < DFRAGS; ++i) { dv_acc[i] = {0.f, 0.f, 0.f, 0.f}; dk_acc[i] = {0.f, 0.f, 0.f, 0.f}; }
+
+  const int q_start = CAUSAL ? s0 : 0;
+  for (int qq0 = q_start; qq0 < S; qq0 += TILE) {
+    __syncthreads();
+    stage_tile<D>(q_l, Q + base + (long long)qq0 * D);
+    stage_tile_t<D>(qt_l, Q + base + (long long)qq0 * D);
+    stage_tile<D>(do_l, dO + base + (long long)qq0 * D);
+    stage_tile_t<D>(dot_l, dO + base + (long long)qq0 * D);
+    for (int i = tid; i < TILE; i += NTHREADS) {
+      lse_l[i] = LSE[bh * S + qq0 + i];
+      dlt_l[i] = delta[bh * S + qq0 + i];
+    }
+    __syncthreads();
+
+    // S^T[s][q] (pre-scaled via k_frag) and dP^T[s][q]
+    f32x4 st[4], dpt[4];
+#pragma unroll
+    for (int nq = 0; nq < 4; ++nq) {
+      st[nq] = {0.f, 0.f, 0.f, 0.f};
+      dpt[nq] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bf16x8 qf = lds_read8(q_l, nq * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 dof = lds_read8(do_l, nq * 16 + fr, ks * 32 + fq * 8, D);
+        st[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[ks], qf, st[nq], 0, 0, 0);
+        dpt[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[ks], dof, dpt[nq], 0, 0, 0);
+      }
+    }
+
+    // P^T = exp(S^T - LSE[q]); write to wave-private T rows
+    float pt[4][4];
+#pragma unroll
+    for (int nq = 0; nq < 4; ++nq) {
+      float lse_q = lse_l[nq * 16 + fr];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int sg = s0 + ws + fq * 4 + r, qg = qq0 + nq * 16 + fr;
+        float e = (CAUSAL && sg > qg) ? 0.f : __expf(st[nq][r] - lse_q);
+        pt[nq][r] = e;
+        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, f2bf(e));
+      }
+    }
+    // dV += P^T @ dO   (a: own T rows over q; b: dO^T rows over q)
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 pa = lds_read8(t_l, ws + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 db = lds_read8(dot_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        dv_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db, dv_acc[nd], 0, 0, 0);
+      }
+
+    // dS^T = scale * P^T * (dP^T - delta[q]); overwrite own T rows
+#pragma unroll
+    for (int nq = 0; nq < 4; ++nq) {
+      float dlt_q = dlt_l[nq * 16 + fr];
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        float ds = scale * pt[nq][r] * (dpt[nq][r] - dlt_q);
+        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, f2bf(ds));
+      }
+    }
+    // dK += dS^T @ Q   (b: Q^T rows over q)
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 da = lds_read8(t_l, ws + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 qb = lds_read8(qt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        dk_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, qb, dk_acc[nd], 0, 0, 0);
+      }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int sg = s0 + ws + fq * 4 + r;
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd) {
+      dK[base + (long long)sg * D + nd * 16 + fr] = f2bf(dk_acc[nd][r]);
+      dV[base + (long long)sg * D + nd * 16 + fr] = f2bf(dv_acc[nd][r]);
+    }
+  }
+}
+
+// ======================================================================
+// backward pass 2: dQ   (grid over q tiles)
+// ======================================================================
+
+template <int D, bool CAUSAL>
+__global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
+    const ushort_t* __restrict__ dO, const ushort_t* __restrict__ Q,
+    const ushort_t* __restrict__ K, const ushort_t* __restrict__ V,
+    const float* __restrict__ LSE, const float* __restrict__ delta,
+    ushort_t* __restrict__ dQ, int B, int H, int S, float scale) {
+  constexpr int KSTEPS = D / 32;
+  constexpr int DFRAGS = D / 16;
+  __shared__ ushort_t k_l[TILE * D];
+  __shared__ ushort_t kt_l[D * TILE];
+  __shared__ ushort_t v_l[TILE * D];
+  __shared__ ushort_t t_l[TILE * TILE];
+  __shared__ float lse_l[TILE], dlt_l[TILE];
+
+  const int q_tile = blockIdx.x;
+  const long long bh = blockIdx.y;
+  const long long base = (bh * S) * D;
+
+  const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
+  const int q0 = q_tile * TILE;
+  const int wq = wave * 16;
+  const int fr = lane & 15, fq = lane >> 4;
+
+  bf16x8 q_frag[KSTEPS], do_frag[KSTEPS];
+  {
+    const ushort_t* qrow = Q + base + (long long)(q0 + wq + fr) * D;
+    const ushort_t* dorow = dO + base + (long long)(q0 + wq + fr) * D;
+#pragma unroll
+    for (int ks = 0; ks < KSTEPS; ++ks) {
+      bf16x8 raw = *reinterpret_cast<const bf16x8*>(qrow + ks * 32 + fq * 8);
+      bf16x8 out;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) out[j] = (short)f2bf(bf2f((ushort_t)raw[j]) * scale);
+      q_frag[ks] = out;
+      do_frag[ks] = *reinterpret_cast<const bf16x8*>(dorow + ks * 32 + fq * 8);
+    }
+  }
+  for (int i = tid; i < TILE; i += NTHREADS) {
+    lse_l[i] = LSE[bh * S + q0 + i];
+    dlt_l[i] = delta[bh * S + q0 + i];
+  }
+
+  f32x4 dq_acc[DFRAGS];
+#pragma unroll
+  for (int i = 0; i < DFRAGS; ++i) dq_acc[i] = {0.f, 0.f, 0.f, 0.f};
+
+  const int s_end = CAUSAL ? (q0 + TILE) : S;
+  for (int s0 = 0; s0 < s_end; s0 += TILE) {
+    __syncthreads();
+    stage_tile<D>(k_l, K + base + (long long)s0 * D);
+    stage_tile_t<D>(kt_l, K + base + (long long)s0 * D);
+    stage_tile<D>(v_l, V + base + (long long)s0 * D);
+    __syncthreads();
+
+    f32x4 sacc[4], dp[4];
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns) {
+      sacc[ns] = {0.f, 0.f, 0.f, 0.f};
+      dp[ns] = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int ks = 0; ks < KSTEPS; ++ks) {
+        bf16x8 kf = lds_read8(k_l, ns * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 vf = lds_read8(v_l, ns * 16 + fr, ks * 32 + fq * 8, D);
+        sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, sacc[ns], 0, 0, 0);
+        dp[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[ks], vf, dp[ns], 0, 0, 0);
+      }
+    }
+
+    // dS = scale * P * (dP - delta[q]); write to wave-private T rows
+#pragma unroll
+    for (int ns = 0; ns < 4; ++ns)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int qrow = wq + fq * 4 + r;
+        int qg = q0 + qrow, sg = s0 + ns * 16 + fr;
+        float p = (CAUSAL && sg > qg) ? 0.f : __expf(sacc[ns][r] - lse_l[qrow]);
+        float ds = scale * p * (dp[ns][r] - dlt_l[qrow]);
+        lds_write1(t_l, qrow, ns * 16 + fr, TILE, f2bf(ds));
+      }
+    // dQ += dS @ K   (b: K^T rows over s)
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        bf16x8 da = lds_read8(t_l, wq + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 kb = lds_read8(kt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        dq_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, kb, dq_acc[nd], 0, 0, 0);
+      }
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int qg = q0 + wq + fq * 4 + r;
+#pragma unroll
+    for (int nd = 0; nd < DFRAGS; ++nd)
+      dQ[base + (long long)qg * D + nd * 16 + fr] = f2bf(dq_acc[nd][r]);
+  }
+}
+
+// ======================================================================
+// launchers
+// ======================================================================
+
+#define DISPATCH_FWD(DV, CV) \
+  hipLaunchKernelGGL((flash_fwd_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
+                     (hipStream_t)stream, (const ushort_t*)Q, (const ushort_t*)K, (const ushort_t*)V, \
+                     (ushort_t*)O, (float*)LSE, B, H, Hkv, S, scale)
+
+extern "C" void launch_flash_fwd(const void* Q, const void* K, const void* V, void* O, void* LSE,
+                                 int B, int H, int Hkv, int S, int D, float scale, int causal,
+                                 void* stream) {
+  if (D == 64) { if (causal) DISPATCH_FWD(64, true); else DISPATCH_FWD(64, false); }
+  else         { if (causal) DISPATCH_FWD(128, true); else DISPATCH_FWD(128, false); }
+}
+
+extern "C" void launch_flash_delta(const void* dO, const void* O, void* delta,
+                                   long long rows, int D, void* stream) {
+  long long blocks = (rows + 3) / 4;
+  if (blocks > 2048) blocks = 2048;
+  if (D == 64)
+    hipLaunchKernelGGL((flash_delta_kernel<64>), dim3((int)blocks), dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)O, (float*)delta, rows);
+  else
+    hipLaunchKernelGGL((flash_delta_kernel<128>), dim3((int)blocks), dim3(NTHREADS), 0,
+                       (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)O, (float*)delta, rows);
+}
+
+#define DISPATCH_DKDV(DV, CV) \
+  hipLaunchKernelGGL((flash_bwd_dkdv_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
+                     (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)Q, (const ushort_t*)K, \
+                     (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
+                     (ushort_t*)dK, (ushort_t*)dV, B, H, S, scale)
+
+extern "C" void launch_flash_bwd_dkdv(const void* dO, const void* Q, const void* K, const void* V,
+                                      const void* LSE, const void* delta, void* dK, void* dV,
+                                      int B, int H, int S, int D, float scale, int causal, void* stream) {
+  if (D == 64) { if (causal) DISPATCH_DKDV(64, true); else DISPATCH_DKDV(64, false); }
+  else         { if (causal) DISPATCH_DKDV(128, true); else DISPATCH_DKDV(128, false); }
+}
+
+#define DISPATCH_DQ(DV, CV) \
+  hipLaunchKernelGGL((flash_bwd_dq_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
+                     (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)Q, (const ushort_t*)K, \
+                     (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
+                     (ushort_t*)dQ, B, H, S, scale)
+
+extern "C" void launch_flash_bwd_dq(const void* dO, const void* Q, const void* K, const void* V,
+                                    const void* LSE, const void* delta, void* dQ,
+                                    int B, int H, int S, int D, float scale, int causal, void* stream) {
+  if (D == 64) { if (causal) DISPATCH_DQ(64, true); else DISPATCH_DQ(64, false); }
+  else         { if (causal) DISPATCH_DQ(128, true); else DISPATCH_DQ(128, false); }
+}

@@ -18,7 +18,11 @@ this_dir = os.path.dirname(os.path.abspath(__file__))
 
 ext = CUDAExtension(
     name="hivemind_amd._hip_ops",
-    sources=["hivemind_amd/ops/hip/bindings.hip", "hivemind_amd/ops/hip/mfma_gemm_impl.hip"],
+    sources=[
+        "hivemind_amd/ops/hip/bindings.hip",
+        "hivemind_amd/ops/hip/mfma_gemm_impl.hip",
+        "hivemind_amd/ops/hip/flash_attention.hip",
+    ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],
         "nvcc": ["-O3", "-std=c++17"],
