@@ -16,10 +16,10 @@ extern "C" void launch_flash_fwd(const void*, const void*, const void*, void*, v
                                  int, int, int, int, int, float, int, void*);
 extern "C" void launch_flash_delta(const void*, const void*, void*, long long, int, void*);
 extern "C" void launch_flash_bwd_dkdv(const void*, const void*, const void*, const void*,
-                                      const void*, const void*, void*, void*,
-                                      int, int, int, int, float, int, void*);
+                                      const void*, const void*, const void*, const void*,
+                                      void*, void*, int, int, int, int, float, int, void*);
 extern "C" void launch_flash_bwd_dq(const void*, const void*, const void*, const void*,
-                                    const void*, const void*, void*,
+                                    const void*, const void*, const void*, void*,
                                     int, int, int, int, float, int, void*);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
@@ -487,7 +487,10 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torc
   TORCH_CHECK(H % Hkv == 0, "H must be a multiple of Hkv");
   auto o = torch::empty_like(q);
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
-  launch_flash_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), o.data_ptr(), lse.data_ptr(),
+  // one [B,H,S,D]->[B,H,D,S] copy so the PV operand stages with vector
+  // LDS writes (in-kernel scalar transposes dominated the tile cost)
+  auto vt = v.transpose(-1, -2).contiguous();
+  launch_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(), lse.data_ptr(),
                    B, H, Hkv, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
   return {o, lse};
 }
@@ -495,20 +498,27 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torc
 std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor d_out, torch::Tensor q, torch::Tensor k,
                                           torch::Tensor v, torch::Tensor o, torch::Tensor lse,
                                           bool causal, double scale) {
-  check_flash_shapes(q, "q"); check_flash_shapes(d_out, "d_out");
+  check_flash_shapes(q, "q");
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(k.size(1) == H, "flash attention backward requires H == Hkv (expand kv first)");
   auto d_out_c = d_out.contiguous();
+  check_flash_shapes(d_out_c, "d_out");
   auto delta = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   launch_flash_delta(d_out_c.data_ptr(), o.data_ptr(), delta.data_ptr(),
                      (long long)B * H * S, D, (void*)current_stream());
+  // pre-transposed operand copies: one pass each instead of per-tile scalar
+  // LDS transposes inside every workgroup's kv/q loop
+  auto dot = d_out_c.transpose(-1, -2).contiguous();
+  auto qt = q.transpose(-1, -2).contiguous();
+  auto kt = k.transpose(-1, -2).contiguous();
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
-  launch_flash_bwd_dkdv(d_out_c.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+  launch_flash_bwd_dkdv(d_out_c.data_ptr(), dot.data_ptr(), q.data_ptr(), qt.data_ptr(),
+                        k.data_ptr(), v.data_ptr(),
                         lse.data_ptr(), delta.data_ptr(), dk.data_ptr(), dv.data_ptr(),
                         B, H, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
-  launch_flash_bwd_dq(d_out_c.data_ptr(), q.data_ptr(), k.data_ptr(), v.data_ptr(),
+  launch_flash_bwd_dq(d_out_c.data_ptr(), q.data_ptr(), k.data_ptr(), kt.data_ptr(), v.data_ptr(),
                       lse.data_ptr(), delta.data_ptr(), dq.data_ptr(),
                       B, H, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
   return {dq, dk, dv};

@@ -70,30 +70,19 @@ __device__ __forceinline__ void lds_write1(ushort_t* lds, int row, int col, int 
   *reinterpret_cast<ushort_t*>(reinterpret_cast<char*>(lds) + swz_off(row, col, cols)) = v;
 }
 
-// stage a [TILE][D] bf16 tile from global (row-major, row stride D) into
-// swizzled LDS; 16B per thread per step, coalesced.
-template <int D>
-__device__ __forceinline__ void stage_tile(ushort_t* lds, const ushort_t* src) {
-  constexpr int SEGS = D / 8;               // 16B segments per row
-  constexpr int TOTAL = TILE * SEGS;
-  for (int idx = threadIdx.x; idx < TOTAL; idx += NTHREADS) {
+// stage a [rows][COLS] bf16 tile from global (row-major, row stride ld) into
+// swizzled LDS; 16B per thread per step, fully vectorized both sides.
+// Transposed operand tiles come from PRE-TRANSPOSED global tensors ([B,H,D,S],
+// one torch transpose per call) so no scalar LDS writes survive -- the
+// in-kernel scalar transpose was ~2x the MFMA issue cost per tile.
+template <int COLS>
+__device__ __forceinline__ void stage_tile(ushort_t* lds, const ushort_t* src, long long ld, int rows) {
+  constexpr int SEGS = COLS / 8;  // 16B segments per row
+  const int total = rows * SEGS;
+  for (int idx = threadIdx.x; idx < total; idx += NTHREADS) {
     int row = idx / SEGS, seg = idx % SEGS;
-    bf16x8 v = *reinterpret_cast<const bf16x8*>(src + (long long)row * D + seg * 8);
-    lds_write8(lds, row, seg * 8, D, v);
-  }
-}
-
-// stage a [TILE][D] tile TRANSPOSED into [D][TILE] swizzled LDS
-// (dst[d][s] = src[s][d]); vectorized 16B global reads, scalar LDS writes.
-template <int D>
-__device__ __forceinline__ void stage_tile_t(ushort_t* lds, const ushort_t* src) {
-  constexpr int SEGS = D / 8;
-  constexpr int TOTAL = TILE * SEGS;
-  for (int idx = threadIdx.x; idx < TOTAL; idx += NTHREADS) {
-    int s = idx / SEGS, seg = idx % SEGS;
-    bf16x8 v = *reinterpret_cast<const bf16x8*>(src + (long long)s * D + seg * 8);
-#pragma unroll
-    for (int j = 0; j < 8; ++j) lds_write1(lds, seg * 8 + j, s, TILE, (ushort_t)v[j]);
+    bf16x8 v = *reinterpret_cast<const bf16x8*>(src + (long long)row * ld + seg * 8);
+    lds_write8(lds, row, seg * 8, COLS, v);
   }
 }
 
@@ -118,7 +107,8 @@ __device__ __forceinline__ float rowsum16(float v) {
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
-    const ushort_t* __restrict__ V, ushort_t* __restrict__ O,
+    const ushort_t* __restrict__ VT,  // [B,Hkv,D,S] pre-transposed
+    ushort_t* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int Hkv, int S, float scale) {
   constexpr int KSTEPS = D / 32;   // MFMA K-steps over the head dim
   constexpr int DFRAGS = D / 16;   // output d-blocks per wave
@@ -161,8 +151,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
   const int s_end = CAUSAL ? (q0 + TILE) : S;
   for (int s0 = 0; s0 < s_end; s0 += TILE) {
     __syncthreads();
-    stage_tile<D>(k_l, K + kv_base + (long long)s0 * D);
-    stage_tile_t<D>(vt_l, V + kv_base + (long long)s0 * D);
+    stage_tile<D>(k_l, K + kv_base + (long long)s0 * D, D, TILE);
+    stage_tile<TILE>(vt_l, VT + kv_base + s0, S, D);
     __syncthreads();
 
     // S-tile: wave computes [16 q][64 s] as 4 col-block fragments
@@ -186,7 +176,9 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
         }
     }
 
-    // online softmax: row stats replicated across each 16-lane group
+    // online softmax: row max replicated across each 16-lane group; the row
+    // SUM stays a per-lane partial (it scales linearly with the rescale
+    // factor) and is butterfly-reduced once in the epilogue
     float p[4][4];  // [ns][reg]
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
@@ -202,7 +194,7 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
         p[ns][r] = e;
         psum += e;
       }
-      l_run[r] = l_run[r] * corr + rowsum16(psum);
+      l_run[r] = l_run[r] * corr + psum;  // per-lane partial
 #pragma unroll
       for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr;
     }
@@ -215,18 +207,22 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
       for (int r = 0; r < 4; ++r)
         lds_write1(pw, fq * 4 + r, ns * 16 + fr, TILE, f2bf(p[ns][r]));
     // wave-private region: in-wave ds ordering suffices, no barrier
+    bf16x8 pa[2];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) pa[ks] = lds_read8(pw, fr, ks * 32 + fq * 8, TILE);
 #pragma unroll
     for (int nd = 0; nd < DFRAGS; ++nd) {
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {  // 64 s = 2 MFMA k-steps
-        bf16x8 pa = lds_read8(pw, fr, ks * 32 + fq * 8, TILE);
         bf16x8 vb = lds_read8(vt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
-        o_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, vb, o_acc[nd], 0, 0, 0);
+        o_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa[ks], vb, o_acc[nd], 0, 0, 0);
       }
     }
   }
 
-  // epilogue: O /= l, save LSE
+  // epilogue: reduce the per-lane partial row sums, then O /= l, save LSE
+#pragma unroll
+  for (int r = 0; r < 4; ++r) l_run[r] = rowsum16(l_run[r]);
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int qg = q0 + wq + fq * 4 + r;
@@ -265,7 +261,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_delta_kernel(
 
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
-    const ushort_t* __restrict__ dO, const ushort_t* __restrict__ Q,
+    const ushort_t* __restrict__ dO, const ushort_t* __restrict__ dOT,
+    const ushort_t* __restrict__ Q, const ushort_t* __restrict__ QT,
     const ushort_t* __restrict__ K, const ushort_t* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ delta,
     ushort_t* __restrict__ dK, ushort_t* __restrict__ dV,
@@ -311,10 +308,10 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
   const int q_start = CAUSAL ? s0 : 0;
   for (int qq0 = q_start; qq0 < S; qq0 += TILE) {
     __syncthreads();
-    stage_tile<D>(q_l, Q + base + (long long)qq0 * D);
-    stage_tile_t<D>(qt_l, Q + base + (long long)qq0 * D);
-    stage_tile<D>(do_l, dO + base + (long long)qq0 * D);
-    stage_tile_t<D>(dot_l, dO + base + (long long)qq0 * D);
+    stage_tile<D>(q_l, Q + base + (long long)qq0 * D, D, TILE);
+    stage_tile<TILE>(qt_l, QT + base + qq0, S, D);
+    stage_tile<D>(do_l, dO + base + (long long)qq0 * D, D, TILE);
+    stage_tile<TILE>(dot_l, dOT + base + qq0, S, D);
     for (int i = tid; i < TILE; i += NTHREADS) {
       lse_l[i] = LSE[bh * S + qq0 + i];
       dlt_l[i] = delta[bh * S + qq0 + i];
@@ -350,13 +347,15 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
       }
     }
     // dV += P^T @ dO   (a: own T rows over q; b: dO^T rows over q)
+    bf16x8 ta[2];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) ta[ks] = lds_read8(t_l, ws + fr, ks * 32 + fq * 8, TILE);
 #pragma unroll
     for (int nd = 0; nd < DFRAGS; ++nd)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 pa = lds_read8(t_l, ws + fr, ks * 32 + fq * 8, TILE);
         bf16x8 db = lds_read8(dot_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
-        dv_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa, db, dv_acc[nd], 0, 0, 0);
+        dv_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ta[ks], db, dv_acc[nd], 0, 0, 0);
       }
 
     // dS^T = scale * P^T * (dP^T - delta[q]); overwrite own T rows
@@ -371,12 +370,13 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     }
     // dK += dS^T @ Q   (b: Q^T rows over q)
 #pragma unroll
+    for (int ks = 0; ks < 2; ++ks) ta[ks] = lds_read8(t_l, ws + fr, ks * 32 + fq * 8, TILE);
+#pragma unroll
     for (int nd = 0; nd < DFRAGS; ++nd)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 da = lds_read8(t_l, ws + fr, ks * 32 + fq * 8, TILE);
         bf16x8 qb = lds_read8(qt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
-        dk_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, qb, dk_acc[nd], 0, 0, 0);
+        dk_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ta[ks], qb, dk_acc[nd], 0, 0, 0);
       }
   }
 
@@ -398,7 +398,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
     const ushort_t* __restrict__ dO, const ushort_t* __restrict__ Q,
-    const ushort_t* __restrict__ K, const ushort_t* __restrict__ V,
+    const ushort_t* __restrict__ K, const ushort_t* __restrict__ KT,
+    const ushort_t* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ delta,
     ushort_t* __restrict__ dQ, int B, int H, int S, float scale) {
   constexpr int KSTEPS = D / 32;
@@ -444,9 +445,9 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
   const int s_end = CAUSAL ? (q0 + TILE) : S;
   for (int s0 = 0; s0 < s_end; s0 += TILE) {
     __syncthreads();
-    stage_tile<D>(k_l, K + base + (long long)s0 * D);
-    stage_tile_t<D>(kt_l, K + base + (long long)s0 * D);
-    stage_tile<D>(v_l, V + base + (long long)s0 * D);
+    stage_tile<D>(k_l, K + base + (long long)s0 * D, D, TILE);
+    stage_tile<TILE>(kt_l, KT + base + s0, S, D);
+    stage_tile<D>(v_l, V + base + (long long)s0 * D, D, TILE);
     __syncthreads();
 
     f32x4 sacc[4], dp[4];
@@ -475,13 +476,15 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
         lds_write1(t_l, qrow, ns * 16 + fr, TILE, f2bf(ds));
       }
     // dQ += dS @ K   (b: K^T rows over s)
+    bf16x8 da[2];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) da[ks] = lds_read8(t_l, wq + fr, ks * 32 + fq * 8, TILE);
 #pragma unroll
     for (int nd = 0; nd < DFRAGS; ++nd)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 da = lds_read8(t_l, wq + fr, ks * 32 + fq * 8, TILE);
         bf16x8 kb = lds_read8(kt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
-        dq_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da, kb, dq_acc[nd], 0, 0, 0);
+        dq_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da[ks], kb, dq_acc[nd], 0, 0, 0);
       }
   }
 
@@ -500,10 +503,10 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 
 #define DISPATCH_FWD(DV, CV) \
   hipLaunchKernelGGL((flash_fwd_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
-                     (hipStream_t)stream, (const ushort_t*)Q, (const ushort_t*)K, (const ushort_t*)V, \
+                     (hipStream_t)stream, (const ushort_t*)Q, (const ushort_t*)K, (const ushort_t*)VT, \
                      (ushort_t*)O, (float*)LSE, B, H, Hkv, S, scale)
 
-extern "C" void launch_flash_fwd(const void* Q, const void* K, const void* V, void* O, void* LSE,
+extern "C" void launch_flash_fwd(const void* Q, const void* K, const void* VT, void* O, void* LSE,
                                  int B, int H, int Hkv, int S, int D, float scale, int causal,
                                  void* stream) {
   if (D == 64) { if (causal) DISPATCH_FWD(64, true); else DISPATCH_FWD(64, false); }
@@ -524,11 +527,13 @@ extern "C" void launch_flash_delta(const void* dO, const void* O, void* delta,
 
 #define DISPATCH_DKDV(DV, CV) \
   hipLaunchKernelGGL((flash_bwd_dkdv_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
-                     (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)Q, (const ushort_t*)K, \
+                     (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)dOT, \
+                     (const ushort_t*)Q, (const ushort_t*)QT, (const ushort_t*)K, \
                      (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
                      (ushort_t*)dK, (ushort_t*)dV, B, H, S, scale)
 
-extern "C" void launch_flash_bwd_dkdv(const void* dO, const void* Q, const void* K, const void* V,
+extern "C" void launch_flash_bwd_dkdv(const void* dO, const void* dOT, const void* Q, const void* QT,
+                                      const void* K, const void* V,
                                       const void* LSE, const void* delta, void* dK, void* dV,
                                       int B, int H, int S, int D, float scale, int causal, void* stream) {
   if (D == 64) { if (causal) DISPATCH_DKDV(64, true); else DISPATCH_DKDV(64, false); }
@@ -538,10 +543,11 @@ extern "C" void launch_flash_bwd_dkdv(const void* dO, const void* Q, const void*
 #define DISPATCH_DQ(DV, CV) \
   hipLaunchKernelGGL((flash_bwd_dq_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
                      (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)Q, (const ushort_t*)K, \
-                     (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
+                     (const ushort_t*)KT, (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
                      (ushort_t*)dQ, B, H, S, scale)
 
-extern "C" void launch_flash_bwd_dq(const void* dO, const void* Q, const void* K, const void* V,
+extern "C" void launch_flash_bwd_dq(const void* dO, const void* Q, const void* K, const void* KT,
+                                    const void* V,
                                     const void* LSE, const void* delta, void* dQ,
                                     int B, int H, int S, int D, float scale, int causal, void* stream) {
   if (D == 64) { if (causal) DISPATCH_DQ(64, true); else DISPATCH_DQ(64, false); }
