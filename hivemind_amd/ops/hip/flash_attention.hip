@@ -31,6 +31,7 @@
 // backward requires H == Hkv (ALBERT; Llama uses repeat_kv before sdpa).
 
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 #include <cstdint>
 
 typedef unsigned short ushort_t;
@@ -40,10 +41,16 @@ typedef __attribute__((ext_vector_type(4))) float f32x4;
 __device__ __forceinline__ float bf2f(ushort_t u) {
   return __uint_as_float(((unsigned int)u) << 16);
 }
+// pair conversion: one v_cvt_pk_bf16_f32 for two values (a scalar manual
+// round-to-even costs 3-4 VALU ops; fwd PMC showed VALU:MFMA = 21:1, and
+// conversions were a top contributor)
+__device__ __forceinline__ void f2bf2(float a, float b, ushort_t& ua, ushort_t& ub) {
+  __hip_bfloat162 h = __float22bfloat162_rn(float2{a, b});
+  ua = __bfloat16_as_ushort(h.x);
+  ub = __bfloat16_as_ushort(h.y);
+}
 __device__ __forceinline__ ushort_t f2bf(float f) {
-  unsigned int w = __float_as_uint(f);
-  unsigned int bias = 0x7FFF + ((w >> 16) & 1);
-  return (ushort_t)((w + bias) >> 16);
+  return __bfloat16_as_ushort(__float2bfloat16(f));
 }
 
 constexpr int TILE = 64;     // q-rows and kv-rows per workgroup tile
@@ -110,10 +117,15 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     const ushort_t* __restrict__ VT,  // [B,Hkv,D,S] pre-transposed
     ushort_t* __restrict__ O,
     float* __restrict__ LSE, int B, int H, int Hkv, int S, float scale) {
-  constexpr int KSTEPS = D / 32;   // MFMA K-steps over the head dim
-  constexpr int DFRAGS = D / 16;   // output d-blocks per wave
-  __shared__ ushort_t k_l[TILE * D];
-  __shared__ ushort_t vt_l[D * TILE];
+  constexpr int KSTEPS = D / 32;       // MFMA K-steps over the head dim
+  constexpr int DFRAGS = D / 16;       // output d-blocks per wave
+  constexpr int KITERS = TILE * (D / 8) / NTHREADS;  // staging 16B segs/thread
+  constexpr int VITERS = D * (TILE / 8) / NTHREADS;
+  // double-buffered K/V tiles: ONE barrier per kv tile -- next tile's global
+  // loads issue before the current tile's MFMA work (T14 async-stage split),
+  // LDS writes land after it, and the single barrier publishes them
+  __shared__ ushort_t k_l[2][TILE * D];
+  __shared__ ushort_t vt_l[2][D * TILE];
   __shared__ ushort_t p_l[4][16 * TILE];  // per-wave P tile [16 q][64 s]
 
   const int q_tile = blockIdx.x;
@@ -136,7 +148,12 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(qrow + ks * 32 + fq * 8);
       bf16x8 out;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) out[j] = (short)f2bf(bf2f((ushort_t)raw[j]) * scale);
+      for (int j = 0; j < 8; j += 2) {
+        ushort_t a, b;
+        f2bf2(bf2f((ushort_t)raw[j]) * scale, bf2f((ushort_t)raw[j + 1]) * scale, a, b);
+        out[j] = (short)a;
+        out[j + 1] = (short)b;
+      }
       q_frag[ks] = out;
     }
   }
@@ -148,12 +165,45 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
 
+  // per-thread staging geometry (constant): 16B segment (row, seg) pairs
+  const int tid_ = tid;
+  bf16x8 k_st[KITERS], v_st[VITERS];
+  const ushort_t* k_src = K + kv_base;
+  const ushort_t* vt_src = VT + kv_base;
+
+  auto load_tile_regs = [&](int s0) {
+#pragma unroll
+    for (int i = 0; i < KITERS; ++i) {
+      int idx = tid_ + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
+      k_st[i] = *reinterpret_cast<const bf16x8*>(k_src + (long long)(s0 + row) * D + seg * 8);
+    }
+#pragma unroll
+    for (int i = 0; i < VITERS; ++i) {
+      int idx = tid_ + i * NTHREADS, row = idx / 8, seg = idx % 8;
+      v_st[i] = *reinterpret_cast<const bf16x8*>(vt_src + (long long)row * S + s0 + seg * 8);
+    }
+  };
+  auto write_tile_lds = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < KITERS; ++i) {
+      int idx = tid_ + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
+      lds_write8(k_l[buf], row, seg * 8, D, k_st[i]);
+    }
+#pragma unroll
+    for (int i = 0; i < VITERS; ++i) {
+      int idx = tid_ + i * NTHREADS, row = idx / 8, seg = idx % 8;
+      lds_write8(vt_l[buf], row, seg * 8, TILE, v_st[i]);
+    }
+  };
+
   const int s_end = CAUSAL ? (q0 + TILE) : S;
-  for (int s0 = 0; s0 < s_end; s0 += TILE) {
-    __syncthreads();
-    stage_tile<D>(k_l, K + kv_base + (long long)s0 * D, D, TILE);
-    stage_tile<TILE>(vt_l, VT + kv_base + s0, S, D);
-    __syncthreads();
+  load_tile_regs(0);
+  write_tile_lds(0);
+  __syncthreads();
+
+  for (int s0 = 0, buf = 0; s0 < s_end; s0 += TILE, buf ^= 1) {
+    const bool has_next = s0 + TILE < s_end;
+    if (has_next) load_tile_regs(s0 + TILE);  // issue early; lands after MFMA
 
     // S-tile: wave computes [16 q][64 s] as 4 col-block fragments
     f32x4 sacc[4];
@@ -162,7 +212,7 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
       sacc[ns] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
-        bf16x8 kf = lds_read8(k_l, ns * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 kf = lds_read8(k_l[buf], ns * 16 + fr, ks * 32 + fq * 8, D);
         sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, sacc[ns], 0, 0, 0);
       }
     }
@@ -199,13 +249,17 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
       for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr;
     }
 
-    // P -> wave-private LDS (bf16), then O += P @ V^T
+    // P -> wave-private LDS (bf16, paired conversions), then O += P @ V^T
     ushort_t* pw = p_l[wave];
 #pragma unroll
-    for (int ns = 0; ns < 4; ++ns)
+    for (int ns = 0; ns < 4; ns += 2)
 #pragma unroll
-      for (int r = 0; r < 4; ++r)
-        lds_write1(pw, fq * 4 + r, ns * 16 + fr, TILE, f2bf(p[ns][r]));
+      for (int r = 0; r < 4; ++r) {
+        ushort_t a, b;
+        f2bf2(p[ns][r], p[ns + 1][r], a, b);
+        lds_write1(pw, fq * 4 + r, ns * 16 + fr, TILE, a);
+        lds_write1(pw, fq * 4 + r, (ns + 1) * 16 + fr, TILE, b);
+      }
     // wave-private region: in-wave ds ordering suffices, no barrier
     bf16x8 pa[2];
 #pragma unroll
@@ -214,9 +268,14 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     for (int nd = 0; nd < DFRAGS; ++nd) {
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {  // 64 s = 2 MFMA k-steps
-        bf16x8 vb = lds_read8(vt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 vb = lds_read8(vt_l[buf], nd * 16 + fr, ks * 32 + fq * 8, TILE);
         o_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pa[ks], vb, o_acc[nd], 0, 0, 0);
       }
+    }
+
+    if (has_next) {
+      write_tile_lds(buf ^ 1);  // publish next tile (loads issued pre-MFMA)
+      __syncthreads();
     }
   }
 
@@ -228,8 +287,12 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     int qg = q0 + wq + fq * 4 + r;
     float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
 #pragma unroll
-    for (int nd = 0; nd < DFRAGS; ++nd)
-      O[q_base + (long long)qg * D + nd * 16 + fr] = f2bf(o_acc[nd][r] * inv_l);
+    for (int nd = 0; nd < DFRAGS; nd += 2) {
+      ushort_t a, b;
+      f2bf2(o_acc[nd][r] * inv_l, o_acc[nd + 1][r] * inv_l, a, b);
+      O[q_base + (long long)qg * D + nd * 16 + fr] = a;
+      O[q_base + (long long)qg * D + (nd + 1) * 16 + fr] = b;
+    }
     if (fr == 0) LSE[bh * S + qg] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
   }
 }
@@ -242,16 +305,23 @@ template <int D>
 __global__ __launch_bounds__(NTHREADS) void flash_delta_kernel(
     const ushort_t* __restrict__ dO, const ushort_t* __restrict__ O,
     float* __restrict__ delta, long long rows) {
-  long long row0 = (long long)blockIdx.x * 4 + (threadIdx.x >> 6);
+  // one wave handles 64/(D/8) rows per iteration with 16B loads (scalar bf16
+  // loads were ~3x slower -- Guideline 13)
+  constexpr int SEGS = D / 8;            // 16B segments per row
+  constexpr int ROWS_PER_WAVE = 64 / SEGS;
   const int lane = threadIdx.x & 63;
-  for (long long row = row0; row < rows; row += (long long)gridDim.x * 4) {
+  const int sub_row = lane / SEGS, seg = lane % SEGS;
+  long long row0 = ((long long)blockIdx.x * 4 + (threadIdx.x >> 6)) * ROWS_PER_WAVE + sub_row;
+  for (long long row = row0; row < rows; row += (long long)gridDim.x * 4 * ROWS_PER_WAVE) {
+    bf16x8 a = *reinterpret_cast<const bf16x8*>(dO + row * D + seg * 8);
+    bf16x8 b = *reinterpret_cast<const bf16x8*>(O + row * D + seg * 8);
     float acc = 0.f;
 #pragma unroll
-    for (int d = lane; d < D; d += 64)
-      acc += bf2f(dO[row * D + d]) * bf2f(O[row * D + d]);
+    for (int j = 0; j < 8; ++j) acc += bf2f((ushort_t)a[j]) * bf2f((ushort_t)b[j]);
+    // reduce across the SEGS lanes that share a row
 #pragma unroll
-    for (int m = 1; m < 64; m <<= 1) acc += __shfl_xor(acc, m, 64);
-    if (lane == 0) delta[row] = acc;
+    for (int m = 1; m < SEGS; m <<= 1) acc += __shfl_xor(acc, m, 64);
+    if (seg == 0) delta[row] = acc;
   }
 }
 
@@ -269,12 +339,17 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     int B, int H, int S, float scale) {
   constexpr int KSTEPS = D / 32;
   constexpr int DFRAGS = D / 16;
-  __shared__ ushort_t q_l[TILE * D];    // Q tile, natural orientation
-  __shared__ ushort_t qt_l[D * TILE];   // Q tile, transposed
-  __shared__ ushort_t do_l[TILE * D];
-  __shared__ ushort_t dot_l[D * TILE];
+  // double-buffer at D=64 (single barrier/iter); D=128's four 16 KB tiles
+  // would leave one block/CU, so it keeps the two-barrier layout
+  constexpr int NBUF = (D == 64) ? 2 : 1;
+  constexpr int TITERS = TILE * (D / 8) / NTHREADS;
+  constexpr int TTITERS = D * (TILE / 8) / NTHREADS;
+  __shared__ ushort_t q_l[NBUF][TILE * D];    // Q tile, natural orientation
+  __shared__ ushort_t qt_l[NBUF][D * TILE];   // Q tile, transposed
+  __shared__ ushort_t do_l[NBUF][TILE * D];
+  __shared__ ushort_t dot_l[NBUF][D * TILE];
   __shared__ ushort_t t_l[TILE * TILE];  // P^T then dS^T (wave-private rows)
-  __shared__ float lse_l[TILE], dlt_l[TILE];
+  __shared__ float lse_l[NBUF][TILE], dlt_l[NBUF][TILE];
 
   const int kv_tile = blockIdx.x;
   const long long bh = blockIdx.y;
@@ -295,7 +370,12 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(krow + ks * 32 + fq * 8);
       bf16x8 out;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) out[j] = (short)f2bf(bf2f((ushort_t)raw[j]) * scale);
+      for (int j = 0; j < 8; j += 2) {
+        ushort_t a, b;
+        f2bf2(bf2f((ushort_t)raw[j]) * scale, bf2f((ushort_t)raw[j + 1]) * scale, a, b);
+        out[j] = (short)a;
+        out[j + 1] = (short)b;
+      }
       k_frag[ks] = out;
       v_frag[ks] = *reinterpret_cast<const bf16x8*>(vrow + ks * 32 + fq * 8);
     }
@@ -305,18 +385,53 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
 #pragma unroll
   for (int i = 0; i < DFRAGS; ++i) { dv_acc[i] = {0.f, 0.f, 0.f, 0.f}; dk_acc[i] = {0.f, 0.f, 0.f, 0.f}; }
 
-  const int q_start = CAUSAL ? s0 : 0;
-  for (int qq0 = q_start; qq0 < S; qq0 += TILE) {
-    __syncthreads();
-    stage_tile<D>(q_l, Q + base + (long long)qq0 * D, D, TILE);
-    stage_tile<TILE>(qt_l, QT + base + qq0, S, D);
-    stage_tile<D>(do_l, dO + base + (long long)qq0 * D, D, TILE);
-    stage_tile<TILE>(dot_l, dOT + base + qq0, S, D);
-    for (int i = tid; i < TILE; i += NTHREADS) {
-      lse_l[i] = LSE[bh * S + qq0 + i];
-      dlt_l[i] = delta[bh * S + qq0 + i];
+  bf16x8 q_st[TITERS], qt_st[TTITERS], do_st[TITERS], dot_st[TTITERS];
+  float lse_st = 0.f, dlt_st = 0.f;
+  auto load_regs = [&](int qq0) {
+#pragma unroll
+    for (int i = 0; i < TITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
+      q_st[i] = *reinterpret_cast<const bf16x8*>(Q + base + (long long)(qq0 + row) * D + seg * 8);
+      do_st[i] = *reinterpret_cast<const bf16x8*>(dO + base + (long long)(qq0 + row) * D + seg * 8);
     }
-    __syncthreads();
+#pragma unroll
+    for (int i = 0; i < TTITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / 8, seg = idx % 8;
+      qt_st[i] = *reinterpret_cast<const bf16x8*>(QT + base + (long long)row * S + qq0 + seg * 8);
+      dot_st[i] = *reinterpret_cast<const bf16x8*>(dOT + base + (long long)row * S + qq0 + seg * 8);
+    }
+    if (tid < TILE) {
+      lse_st = LSE[bh * S + qq0 + tid];
+      dlt_st = delta[bh * S + qq0 + tid];
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < TITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
+      lds_write8(q_l[buf], row, seg * 8, D, q_st[i]);
+      lds_write8(do_l[buf], row, seg * 8, D, do_st[i]);
+    }
+#pragma unroll
+    for (int i = 0; i < TTITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / 8, seg = idx % 8;
+      lds_write8(qt_l[buf], row, seg * 8, TILE, qt_st[i]);
+      lds_write8(dot_l[buf], row, seg * 8, TILE, dot_st[i]);
+    }
+    if (tid < TILE) {
+      lse_l[buf][tid] = lse_st;
+      dlt_l[buf][tid] = dlt_st;
+    }
+  };
+
+  const int q_start = CAUSAL ? s0 : 0;
+  load_regs(q_start);
+  write_lds(0);
+  __syncthreads();
+
+  for (int qq0 = q_start, buf = 0; qq0 < S; qq0 += TILE, buf ^= (NBUF - 1)) {
+    const bool has_next = qq0 + TILE < S;
+    if (NBUF == 2 && has_next) load_regs(qq0 + TILE);
 
     // S^T[s][q] (pre-scaled via k_frag) and dP^T[s][q]
     f32x4 st[4], dpt[4];
@@ -326,26 +441,33 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
       dpt[nq] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
-        bf16x8 qf = lds_read8(q_l, nq * 16 + fr, ks * 32 + fq * 8, D);
-        bf16x8 dof = lds_read8(do_l, nq * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 qf = lds_read8(q_l[buf], nq * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 dof = lds_read8(do_l[buf], nq * 16 + fr, ks * 32 + fq * 8, D);
         st[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[ks], qf, st[nq], 0, 0, 0);
         dpt[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[ks], dof, dpt[nq], 0, 0, 0);
       }
     }
 
-    // P^T = exp(S^T - LSE[q]); write to wave-private T rows
+    // P^T = exp(S^T - LSE[q]); write to wave-private T rows (paired cvt)
     float pt[4][4];
 #pragma unroll
     for (int nq = 0; nq < 4; ++nq) {
-      float lse_q = lse_l[nq * 16 + fr];
+      float lse_q = lse_l[buf][nq * 16 + fr];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int sg = s0 + ws + fq * 4 + r, qg = qq0 + nq * 16 + fr;
-        float e = (CAUSAL && sg > qg) ? 0.f : __expf(st[nq][r] - lse_q);
-        pt[nq][r] = e;
-        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, f2bf(e));
+        pt[nq][r] = (CAUSAL && sg > qg) ? 0.f : __expf(st[nq][r] - lse_q);
       }
     }
+#pragma unroll
+    for (int nq = 0; nq < 4; nq += 2)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        ushort_t a, b;
+        f2bf2(pt[nq][r], pt[nq + 1][r], a, b);
+        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, a);
+        lds_write1(t_l, ws + fq * 4 + r, (nq + 1) * 16 + fr, TILE, b);
+      }
     // dV += P^T @ dO   (a: own T rows over q; b: dO^T rows over q)
     bf16x8 ta[2];
 #pragma unroll
@@ -354,18 +476,21 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     for (int nd = 0; nd < DFRAGS; ++nd)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 db = lds_read8(dot_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 db = lds_read8(dot_l[buf], nd * 16 + fr, ks * 32 + fq * 8, TILE);
         dv_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ta[ks], db, dv_acc[nd], 0, 0, 0);
       }
 
     // dS^T = scale * P^T * (dP^T - delta[q]); overwrite own T rows
 #pragma unroll
-    for (int nq = 0; nq < 4; ++nq) {
-      float dlt_q = dlt_l[nq * 16 + fr];
+    for (int nq = 0; nq < 4; nq += 2) {
+      float d0 = dlt_l[buf][nq * 16 + fr], d1 = dlt_l[buf][(nq + 1) * 16 + fr];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        float ds = scale * pt[nq][r] * (dpt[nq][r] - dlt_q);
-        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, f2bf(ds));
+        ushort_t a, b;
+        f2bf2(scale * pt[nq][r] * (dpt[nq][r] - d0),
+              scale * pt[nq + 1][r] * (dpt[nq + 1][r] - d1), a, b);
+        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, a);
+        lds_write1(t_l, ws + fq * 4 + r, (nq + 1) * 16 + fr, TILE, b);
       }
     }
     // dK += dS^T @ Q   (b: Q^T rows over q)
@@ -375,9 +500,21 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     for (int nd = 0; nd < DFRAGS; ++nd)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 qb = lds_read8(qt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 qb = lds_read8(qt_l[buf], nd * 16 + fr, ks * 32 + fq * 8, TILE);
         dk_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ta[ks], qb, dk_acc[nd], 0, 0, 0);
       }
+
+    if (NBUF == 2) {
+      if (has_next) {
+        write_lds(buf ^ 1);
+        __syncthreads();
+      }
+    } else if (has_next) {
+      __syncthreads();  // everyone done reading before overwrite
+      load_regs(qq0 + TILE);
+      write_lds(0);
+      __syncthreads();
+    }
   }
 
 #pragma unroll
@@ -385,8 +522,10 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     int sg = s0 + ws + fq * 4 + r;
 #pragma unroll
     for (int nd = 0; nd < DFRAGS; ++nd) {
-      dK[base + (long long)sg * D + nd * 16 + fr] = f2bf(dk_acc[nd][r]);
-      dV[base + (long long)sg * D + nd * 16 + fr] = f2bf(dv_acc[nd][r]);
+      ushort_t a, b;
+      f2bf2(dk_acc[nd][r], dv_acc[nd][r], a, b);
+      dK[base + (long long)sg * D + nd * 16 + fr] = a;
+      dV[base + (long long)sg * D + nd * 16 + fr] = b;
     }
   }
 }
@@ -404,9 +543,12 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
     ushort_t* __restrict__ dQ, int B, int H, int S, float scale) {
   constexpr int KSTEPS = D / 32;
   constexpr int DFRAGS = D / 16;
-  __shared__ ushort_t k_l[TILE * D];
-  __shared__ ushort_t kt_l[D * TILE];
-  __shared__ ushort_t v_l[TILE * D];
+  constexpr int NBUF = (D == 64) ? 2 : 1;
+  constexpr int TITERS = TILE * (D / 8) / NTHREADS;
+  constexpr int TTITERS = D * (TILE / 8) / NTHREADS;
+  __shared__ ushort_t k_l[NBUF][TILE * D];
+  __shared__ ushort_t kt_l[NBUF][D * TILE];
+  __shared__ ushort_t v_l[NBUF][TILE * D];
   __shared__ ushort_t t_l[TILE * TILE];
   __shared__ float lse_l[TILE], dlt_l[TILE];
 
@@ -428,7 +570,12 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(qrow + ks * 32 + fq * 8);
       bf16x8 out;
 #pragma unroll
-      for (int j = 0; j < 8; ++j) out[j] = (short)f2bf(bf2f((ushort_t)raw[j]) * scale);
+      for (int j = 0; j < 8; j += 2) {
+        ushort_t a, b;
+        f2bf2(bf2f((ushort_t)raw[j]) * scale, bf2f((ushort_t)raw[j + 1]) * scale, a, b);
+        out[j] = (short)a;
+        out[j + 1] = (short)b;
+      }
       q_frag[ks] = out;
       do_frag[ks] = *reinterpret_cast<const bf16x8*>(dorow + ks * 32 + fq * 8);
     }
@@ -442,13 +589,42 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 #pragma unroll
   for (int i = 0; i < DFRAGS; ++i) dq_acc[i] = {0.f, 0.f, 0.f, 0.f};
 
+  bf16x8 k_st[TITERS], v_st[TITERS], kt_st[TTITERS];
+  auto load_regs = [&](int s0) {
+#pragma unroll
+    for (int i = 0; i < TITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
+      k_st[i] = *reinterpret_cast<const bf16x8*>(K + base + (long long)(s0 + row) * D + seg * 8);
+      v_st[i] = *reinterpret_cast<const bf16x8*>(V + base + (long long)(s0 + row) * D + seg * 8);
+    }
+#pragma unroll
+    for (int i = 0; i < TTITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / 8, seg = idx % 8;
+      kt_st[i] = *reinterpret_cast<const bf16x8*>(KT + base + (long long)row * S + s0 + seg * 8);
+    }
+  };
+  auto write_lds = [&](int buf) {
+#pragma unroll
+    for (int i = 0; i < TITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
+      lds_write8(k_l[buf], row, seg * 8, D, k_st[i]);
+      lds_write8(v_l[buf], row, seg * 8, D, v_st[i]);
+    }
+#pragma unroll
+    for (int i = 0; i < TTITERS; ++i) {
+      int idx = tid + i * NTHREADS, row = idx / 8, seg = idx % 8;
+      lds_write8(kt_l[buf], row, seg * 8, TILE, kt_st[i]);
+    }
+  };
+
   const int s_end = CAUSAL ? (q0 + TILE) : S;
-  for (int s0 = 0; s0 < s_end; s0 += TILE) {
-    __syncthreads();
-    stage_tile<D>(k_l, K + base + (long long)s0 * D, D, TILE);
-    stage_tile<TILE>(kt_l, KT + base + s0, S, D);
-    stage_tile<D>(v_l, V + base + (long long)s0 * D, D, TILE);
-    __syncthreads();
+  load_regs(0);
+  write_lds(0);
+  __syncthreads();
+
+  for (int s0 = 0, buf = 0; s0 < s_end; s0 += TILE, buf ^= (NBUF - 1)) {
+    const bool has_next = s0 + TILE < s_end;
+    if (NBUF == 2 && has_next) load_regs(s0 + TILE);
 
     f32x4 sacc[4], dp[4];
 #pragma unroll
@@ -457,8 +633,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
       dp[ns] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
-        bf16x8 kf = lds_read8(k_l, ns * 16 + fr, ks * 32 + fq * 8, D);
-        bf16x8 vf = lds_read8(v_l, ns * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 kf = lds_read8(k_l[buf], ns * 16 + fr, ks * 32 + fq * 8, D);
+        bf16x8 vf = lds_read8(v_l[buf], ns * 16 + fr, ks * 32 + fq * 8, D);
         sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, sacc[ns], 0, 0, 0);
         dp[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[ks], vf, dp[ns], 0, 0, 0);
       }
@@ -466,14 +642,18 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 
     // dS = scale * P * (dP - delta[q]); write to wave-private T rows
 #pragma unroll
-    for (int ns = 0; ns < 4; ++ns)
+    for (int ns = 0; ns < 4; ns += 2)
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         int qrow = wq + fq * 4 + r;
-        int qg = q0 + qrow, sg = s0 + ns * 16 + fr;
-        float p = (CAUSAL && sg > qg) ? 0.f : __expf(sacc[ns][r] - lse_l[qrow]);
-        float ds = scale * p * (dp[ns][r] - dlt_l[qrow]);
-        lds_write1(t_l, qrow, ns * 16 + fr, TILE, f2bf(ds));
+        int qg = q0 + qrow;
+        float p0 = (CAUSAL && s0 + ns * 16 + fr > qg) ? 0.f : __expf(sacc[ns][r] - lse_l[qrow]);
+        float p1 = (CAUSAL && s0 + (ns + 1) * 16 + fr > qg) ? 0.f : __expf(sacc[ns + 1][r] - lse_l[qrow]);
+        ushort_t a, b;
+        f2bf2(scale * p0 * (dp[ns][r] - dlt_l[qrow]),
+              scale * p1 * (dp[ns + 1][r] - dlt_l[qrow]), a, b);
+        lds_write1(t_l, qrow, ns * 16 + fr, TILE, a);
+        lds_write1(t_l, qrow, (ns + 1) * 16 + fr, TILE, b);
       }
     // dQ += dS @ K   (b: K^T rows over s)
     bf16x8 da[2];
@@ -483,17 +663,33 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
     for (int nd = 0; nd < DFRAGS; ++nd)
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
-        bf16x8 kb = lds_read8(kt_l, nd * 16 + fr, ks * 32 + fq * 8, TILE);
+        bf16x8 kb = lds_read8(kt_l[buf], nd * 16 + fr, ks * 32 + fq * 8, TILE);
         dq_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(da[ks], kb, dq_acc[nd], 0, 0, 0);
       }
+
+    if (NBUF == 2) {
+      if (has_next) {
+        write_lds(buf ^ 1);
+        __syncthreads();
+      }
+    } else if (has_next) {
+      __syncthreads();
+      load_regs(s0 + TILE);
+      write_lds(0);
+      __syncthreads();
+    }
   }
 
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int qg = q0 + wq + fq * 4 + r;
 #pragma unroll
-    for (int nd = 0; nd < DFRAGS; ++nd)
-      dQ[base + (long long)qg * D + nd * 16 + fr] = f2bf(dq_acc[nd][r]);
+    for (int nd = 0; nd < DFRAGS; nd += 2) {
+      ushort_t a, b;
+      f2bf2(dq_acc[nd][r], dq_acc[nd + 1][r], a, b);
+      dQ[base + (long long)qg * D + nd * 16 + fr] = a;
+      dQ[base + (long long)qg * D + (nd + 1) * 16 + fr] = b;
+    }
   }
 }
 
