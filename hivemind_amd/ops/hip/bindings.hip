@@ -21,6 +21,7 @@ extern "C" void launch_flash_bwd_dkdv(const void*, const void*, const void*, con
 extern "C" void launch_flash_bwd_dq(const void*, const void*, const void*, const void*,
                                     const void*, const void*, const void*, void*,
                                     int, int, int, int, float, int, void*);
+extern "C" void launch_transpose_bhsd(const void*, void*, long long, int, int, void*);
 
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
@@ -476,6 +477,17 @@ static void check_flash_shapes(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.dim() == 4, name, " must be [B, H, S, D]");
 }
 
+torch::Tensor transpose_bhsd(torch::Tensor t) {
+  // [B,H,S,D] -> [B,H,D,S] via the in-register 8x8 transpose kernel
+  check_flash_shapes(t, "transpose input");
+  int B = t.size(0), H = t.size(1), S = t.size(2), D = t.size(3);
+  TORCH_CHECK(S % 64 == 0 && D % 64 == 0, "transpose_bhsd needs S, D multiples of 64");
+  auto out = torch::empty({B, H, D, S}, t.options());
+  launch_transpose_bhsd(t.data_ptr(), out.data_ptr(), (long long)B * H, S, D,
+                        (void*)current_stream());
+  return out;
+}
+
 std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
                                           bool causal, double scale) {
   check_flash_shapes(q, "q"); check_flash_shapes(k, "k"); check_flash_shapes(v, "v");
@@ -489,7 +501,7 @@ std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torc
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   // one [B,H,S,D]->[B,H,D,S] copy so the PV operand stages with vector
   // LDS writes (in-kernel scalar transposes dominated the tile cost)
-  auto vt = v.transpose(-1, -2).contiguous();
+  auto vt = transpose_bhsd(v);
   launch_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(), lse.data_ptr(),
                    B, H, Hkv, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
   return {o, lse};
@@ -508,9 +520,9 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor d_out, torch::Tensor q, 
                      (long long)B * H * S, D, (void*)current_stream());
   // pre-transposed operand copies: one pass each instead of per-tile scalar
   // LDS transposes inside every workgroup's kv/q loop
-  auto dot = d_out_c.transpose(-1, -2).contiguous();
-  auto qt = q.transpose(-1, -2).contiguous();
-  auto kt = k.transpose(-1, -2).contiguous();
+  auto dot = transpose_bhsd(d_out_c);
+  auto qt = transpose_bhsd(q);
+  auto kt = transpose_bhsd(k);
   auto dq = torch::empty_like(q);
   auto dk = torch::empty_like(k);
   auto dv = torch::empty_like(v);
@@ -549,4 +561,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("layernorm_bwd", &layernorm_bwd, "layernorm backward");
   m.def("flash_attn_fwd", &flash_attn_fwd, "CDNA4 flash attention forward -> (O, logsumexp)");
   m.def("flash_attn_bwd", &flash_attn_bwd, "CDNA4 flash attention backward -> (dQ, dK, dV)");
+  m.def("transpose_bhsd", &transpose_bhsd, "[B,H,S,D] -> [B,H,D,S] in-register 8x8 transpose");
 }

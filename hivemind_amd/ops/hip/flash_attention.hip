@@ -106,6 +106,79 @@ __device__ __forceinline__ float rowsum16(float v) {
 }
 
 // ======================================================================
+// fast [B,H,S,D] -> [B,H,D,S] transpose (feeds the flash kernels'
+// pre-transposed operands; torch's bf16 transpose copy runs at ~1 TB/s,
+// this targets the HBM roofline with in-register 8x8 transposes)
+// ======================================================================
+// One wave per 64x64 tile: each lane transposes one 8x8 bf16 block in
+// registers (32 v_perm for the 16-bit interleave; the u32 regrouping is
+// register re-labeling), parks it at the transposed block position in LDS,
+// then the wave streams the tile out row-major -- 16B vectors on both
+// global sides and both LDS sides.
+
+typedef __attribute__((ext_vector_type(4))) unsigned int u32x4;
+
+__device__ __forceinline__ void transpose8x8(const unsigned int R[8][4], unsigned int O[8][4]) {
+  unsigned int P[8][4];
+#pragma unroll
+  for (int i = 0; i < 4; ++i)
+#pragma unroll
+    for (int k = 0; k < 4; ++k) {
+      // (M[2i][2k], M[2i+1][2k]) and (M[2i][2k+1], M[2i+1][2k+1])
+      P[2 * i][k] = __builtin_amdgcn_perm(R[2 * i][k], R[2 * i + 1][k], 0x01000504u);
+      P[2 * i + 1][k] = __builtin_amdgcn_perm(R[2 * i][k], R[2 * i + 1][k], 0x03020706u);
+    }
+#pragma unroll
+  for (int c = 0; c < 8; ++c)
+#pragma unroll
+    for (int k = 0; k < 4; ++k) O[c][k] = P[2 * k + (c & 1)][c >> 1];
+}
+
+__global__ __launch_bounds__(64) void transpose_bhsd_kernel(
+    const ushort_t* __restrict__ in,  // [BH, S, D]
+    ushort_t* __restrict__ out,       // [BH, D, S]
+    int S, int D) {
+  __shared__ ushort_t tile[64 * 64];
+  const int lane = threadIdx.x;
+  const int tiles_d = D / 64;
+  const int tile_idx = blockIdx.x;
+  const int s0 = (tile_idx / tiles_d) * 64, d0 = (tile_idx % tiles_d) * 64;
+  const long long base = (long long)blockIdx.y * S * D;
+  const long long obase = (long long)blockIdx.y * S * D;
+
+  // lane -> 8x8 block (sb, db) within the 64x64 tile
+  const int sb = lane & 7, db = lane >> 3;
+  unsigned int R[8][4], O[8][4];
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    *reinterpret_cast<u32x4*>(R[j]) = *reinterpret_cast<const u32x4*>(
+        in + base + (long long)(s0 + sb * 8 + j) * D + d0 + db * 8);
+  transpose8x8(R, O);
+  // park transposed block at (db, sb) -> LDS is the transposed tile, row-major
+#pragma unroll
+  for (int j = 0; j < 8; ++j)
+    *reinterpret_cast<u32x4*>(reinterpret_cast<char*>(tile) +
+                              (((db * 8 + j) * 64 + sb * 8) * 2 ^ (((db * 8 + j) & 7) << 4))) =
+        *reinterpret_cast<u32x4*>(O[j]);
+  __syncthreads();
+  // stream out: 64 rows (d) x 64 cols (s); 8 segs per lane
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+    int row = (lane * 8 + i) / 8, seg = (lane * 8 + i) % 8;  // row = lane, seg = i
+    u32x4 v = *reinterpret_cast<const u32x4*>(
+        reinterpret_cast<const char*>(tile) + ((row * 64 + seg * 8) * 2 ^ ((row & 7) << 4)));
+    *reinterpret_cast<u32x4*>(out + obase + (long long)(d0 + row) * S + s0 + seg * 8) = v;
+  }
+}
+
+extern "C" void launch_transpose_bhsd(const void* in, void* out, long long BH, int S, int D,
+                                      void* stream) {
+  dim3 grid((S / 64) * (D / 64), (unsigned)BH);
+  hipLaunchKernelGGL(transpose_bhsd_kernel, grid, dim3(64), 0, (hipStream_t)stream,
+                     (const ushort_t*)in, (ushort_t*)out, S, D);
+}
+
+// ======================================================================
 // forward
 // ======================================================================
 // grid (S/64, B*H); saves O and logsumexp (LSE = m + log(l), natural units

@@ -471,3 +471,14 @@ def test_flash_attention_gqa_forward():
             q.float(), k.float(), v.float(), is_causal=True, enable_gqa=True
         )
     assert (out.float() - ref).abs().max().item() < 0.08
+
+
+def test_transpose_bhsd_matches_torch():
+    from hivemind_amd.ops import hip_ops
+
+    torch.manual_seed(3)
+    for B, H, S, D in [(2, 3, 128, 64), (1, 2, 64, 128), (2, 1, 512, 64)]:
+        x = torch.randn(B, H, S, D, device="cuda", dtype=torch.bfloat16)
+        got = hip_ops().transpose_bhsd(x)
+        ref = x.transpose(-1, -2).contiguous()
+        assert torch.equal(got, ref), (B, H, S, D)
