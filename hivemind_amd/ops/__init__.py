@@ -513,6 +513,10 @@ class FusedAdamW(torch.optim.Optimizer):
         loss = closure() if closure is not None else None
         for group in self.param_groups:
             beta1, beta2 = group["betas"]
+            # one multi-tensor launch per (step-count) bucket instead of one
+            # kernel per parameter (round-1 profile: per-param launches were
+            # a visible tail)
+            mt_buckets = {}  # step -> ([p], [g], [m], [v], [mirror])
             for p in group["params"]:
                 if p.grad is None:
                     continue
@@ -523,7 +527,23 @@ class FusedAdamW(torch.optim.Optimizer):
                     state["exp_avg_sq"] = torch.zeros_like(p, dtype=torch.float32)
                 state["step"] += 1
                 mirror = self._mirrors.get(p)
-                if p.is_cuda and p.dtype == torch.float32:
+                if (
+                    p.is_cuda
+                    and p.dtype == torch.float32
+                    and p.is_contiguous()
+                    and p.grad.is_contiguous()
+                    and p.grad.dtype in (torch.float32, torch.bfloat16)
+                ):
+                    kernel_mirror = mirror if (mirror is not None and mirror.dtype == torch.bfloat16
+                                               and mirror.is_contiguous()) else None
+                    bucket = mt_buckets.setdefault(state["step"], ([], [], [], [], [], []))
+                    bucket[0].append(p.data)
+                    bucket[1].append(p.grad)
+                    bucket[2].append(state["exp_avg"])
+                    bucket[3].append(state["exp_avg_sq"])
+                    bucket[4].append(kernel_mirror)
+                    bucket[5].append((p, mirror, kernel_mirror))
+                elif p.is_cuda and p.dtype == torch.float32:
                     kernel_mirror = mirror if (mirror is not None and mirror.dtype == torch.bfloat16) else None
                     hip_ops().fused_adamw_(
                         p.data, p.grad.contiguous(), state["exp_avg"], state["exp_avg_sq"],
@@ -541,5 +561,13 @@ class FusedAdamW(torch.optim.Optimizer):
                     update = state["exp_avg"] / bias_corr1 / denom + group["weight_decay"] * p.float()
                     p.data.add_(-group["lr"] * update.to(p.dtype))
                     if mirror is not None:
+                        mirror.copy_(p.data.to(mirror.dtype))
+            for step_count, bucket in mt_buckets.items():
+                hip_ops().multi_adamw_(
+                    bucket[0], bucket[1], bucket[2], bucket[3], bucket[4],
+                    group["lr"], beta1, beta2, group["eps"], group["weight_decay"], step_count,
+                )
+                for p, mirror, kernel_mirror in bucket[5]:
+                    if mirror is not None and kernel_mirror is None:
                         mirror.copy_(p.data.to(mirror.dtype))
         return loss

@@ -23,6 +23,8 @@ extern "C" void launch_flash_bwd_dq(const void*, const void*, const void*, const
                                     int, int, int, int, float, int, void*);
 extern "C" void launch_transpose_bhsd(const void*, void*, long long, int, int, void*);
 
+#include "multi_tensor.h"
+
 #define CHECK_GPU(x) TORCH_CHECK(x.is_cuda(), #x " must be on the GPU")
 #define CHECK_CONTIG(x) TORCH_CHECK(x.is_contiguous(), #x " must be contiguous")
 
@@ -469,6 +471,72 @@ torch::Tensor stage_probe(torch::Tensor A) {
   return out;
 }
 
+// ----------------------------------------------------------- multi-tensor
+
+void multi_adamw_(std::vector<torch::Tensor> params, std::vector<torch::Tensor> grads,
+                  std::vector<torch::Tensor> exp_avgs, std::vector<torch::Tensor> exp_avg_sqs,
+                  std::vector<c10::optional<torch::Tensor>> mirrors,
+                  double lr, double beta1, double beta2, double eps, double weight_decay,
+                  int64_t step) {
+  size_t n = params.size();
+  TORCH_CHECK(grads.size() == n && exp_avgs.size() == n && exp_avg_sqs.size() == n &&
+              mirrors.size() == n, "multi_adamw_: list length mismatch");
+  float bias_corr1 = 1.0f - powf((float)beta1, (float)step);
+  float bias_corr2 = 1.0f - powf((float)beta2, (float)step);
+  for (size_t start = 0; start < n; start += MT_CHUNK) {
+    MTAdamArgs args{};
+    args.n = 0;
+    args.cum[0] = 0;
+    args.g_bf16_mask = 0;
+    for (size_t t = start; t < n && args.n < MT_CHUNK; ++t, ++args.n) {
+      auto& p = params[t];
+      CHECK_GPU(p); CHECK_CONTIG(p);
+      TORCH_CHECK(p.scalar_type() == torch::kFloat32, "multi_adamw_ params must be fp32 masters");
+      auto& g = grads[t];
+      TORCH_CHECK(g.is_contiguous() && g.numel() == p.numel(), "bad grad ", t);
+      int i = args.n;
+      args.p[i] = p.data_ptr<float>();
+      args.g[i] = g.data_ptr();
+      if (g.scalar_type() == torch::kBFloat16) args.g_bf16_mask |= (1ull << i);
+      else TORCH_CHECK(g.scalar_type() == torch::kFloat32, "grads must be fp32 or bf16");
+      args.m[i] = exp_avgs[t].data_ptr<float>();
+      args.v[i] = exp_avg_sqs[t].data_ptr<float>();
+      args.mirror[i] = mirrors[t].has_value()
+                           ? (mt_ushort*)mirrors[t]->data_ptr() : nullptr;
+      args.cum[i + 1] = args.cum[i] + p.numel();
+    }
+    launch_multi_tensor_adamw(&args, (float)lr, (float)beta1, (float)beta2, (float)eps,
+                              (float)weight_decay, bias_corr1, bias_corr2,
+                              (void*)current_stream());
+  }
+}
+
+void multi_accumulate_(std::vector<torch::Tensor> accs, std::vector<torch::Tensor> xs,
+                       double alpha) {
+  size_t n = accs.size();
+  TORCH_CHECK(xs.size() == n, "multi_accumulate_: list length mismatch");
+  for (size_t start = 0; start < n; start += MT_CHUNK) {
+    MTAccArgs args{};
+    args.n = 0;
+    args.cum[0] = 0;
+    args.x_bf16_mask = 0;
+    for (size_t t = start; t < n && args.n < MT_CHUNK; ++t, ++args.n) {
+      auto& a = accs[t];
+      CHECK_GPU(a); CHECK_CONTIG(a);
+      TORCH_CHECK(a.scalar_type() == torch::kFloat32, "accumulators must be fp32");
+      auto& x = xs[t];
+      TORCH_CHECK(x.is_contiguous() && x.numel() == a.numel(), "bad addend ", t);
+      int i = args.n;
+      args.acc[i] = a.data_ptr<float>();
+      args.x[i] = x.data_ptr();
+      if (x.scalar_type() == torch::kBFloat16) args.x_bf16_mask |= (1ull << i);
+      else TORCH_CHECK(x.scalar_type() == torch::kFloat32, "addends must be fp32 or bf16");
+      args.cum[i + 1] = args.cum[i] + a.numel();
+    }
+    launch_multi_tensor_accumulate(&args, (float)alpha, (void*)current_stream());
+  }
+}
+
 // ------------------------------------------------------------- attention
 
 static void check_flash_shapes(const torch::Tensor& t, const char* name) {
@@ -562,4 +630,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("flash_attn_fwd", &flash_attn_fwd, "CDNA4 flash attention forward -> (O, logsumexp)");
   m.def("flash_attn_bwd", &flash_attn_bwd, "CDNA4 flash attention backward -> (dQ, dK, dV)");
   m.def("transpose_bhsd", &transpose_bhsd, "[B,H,S,D] -> [B,H,D,S] in-register 8x8 transpose");
+  m.def("multi_adamw_", &multi_adamw_, "one-launch AdamW step over a parameter list");
+  m.def("multi_accumulate_", &multi_accumulate_, "one-launch acc += alpha*x over a tensor list");
 }

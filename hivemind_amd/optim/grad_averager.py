@@ -98,7 +98,27 @@ class GradientAverager(DecentralizedAverager):
         if not self.reuse_grad_buffers:
             # re-scale so batches of different sizes contribute proportionally
             alpha = float(batch_size) / self._anchor_batch_size
+            mt_accs, mt_bufs, fallback = [], [], []
             for grad_buf, grad_acc in zip(self._grads_from_parameters(), self._grad_accumulators()):
+                if (
+                    grad_acc.is_cuda
+                    and grad_acc.dtype == torch.float32
+                    and grad_acc.is_contiguous()
+                    and grad_buf.device == grad_acc.device
+                    and grad_buf.is_contiguous()
+                    and grad_buf.dtype in (torch.float32, torch.bfloat16)
+                ):
+                    mt_accs.append(grad_acc)
+                    mt_bufs.append(grad_buf)
+                else:
+                    fallback.append((grad_buf, grad_acc))
+            if mt_accs:
+                from ..ops import hip_ops
+
+                # one launch for the whole accumulator list (round-1 profile:
+                # per-tensor adds were 6.2% of the step)
+                hip_ops().multi_accumulate_(mt_accs, mt_bufs, alpha)
+            for grad_buf, grad_acc in fallback:
                 grad_acc.add_(grad_buf.to(grad_acc.device, grad_acc.dtype), alpha=alpha)
 
     def schedule_step(self, scheduled_time: Optional[DHTExpiration] = None, **kwargs) -> StepControl:

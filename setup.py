@@ -22,6 +22,7 @@ ext = CUDAExtension(
         "hivemind_amd/ops/hip/bindings.hip",
         "hivemind_amd/ops/hip/mfma_gemm_impl.hip",
         "hivemind_amd/ops/hip/flash_attention.hip",
+        "hivemind_amd/ops/hip/multi_tensor.hip",
     ],
     extra_compile_args={
         "cxx": ["-O3", "-std=c++17"],

@@ -482,3 +482,55 @@ def test_transpose_bhsd_matches_torch():
         got = hip_ops().transpose_bhsd(x)
         ref = x.transpose(-1, -2).contiguous()
         assert torch.equal(got, ref), (B, H, S, D)
+
+
+def test_multi_tensor_adamw_matches_torch():
+    """One-launch multi-tensor AdamW vs torch reference over a ragged list of
+    fp32 masters with mixed fp32/bf16 grads and optional bf16 mirrors."""
+    from hivemind_amd.ops import hip_ops
+
+    torch.manual_seed(11)
+    sizes = [7, 4096, 333, 64 * 1024 + 5, 1]
+    lr, b1, b2, eps, wd = 1e-2, 0.9, 0.99, 1e-8, 0.01
+    params = [torch.randn(s, device="cuda", dtype=torch.float32) for s in sizes]
+    grads = [
+        torch.randn(s, device="cuda", dtype=torch.bfloat16 if i % 2 else torch.float32)
+        for i, s in enumerate(sizes)
+    ]
+    ms = [torch.rand(s, device="cuda") * 0.1 for s in sizes]
+    vs = [torch.rand(s, device="cuda") * 0.1 for s in sizes]
+    mirrors = [torch.zeros(s, device="cuda", dtype=torch.bfloat16) if i == 2 else None
+               for i, s in enumerate(sizes)]
+    refs = [(p.clone(), m.clone(), v.clone()) for p, m, v in zip(params, ms, vs)]
+
+    step = 3
+    hip_ops().multi_adamw_(params, grads, ms, vs, mirrors, lr, b1, b2, eps, wd, step)
+    torch.cuda.synchronize()
+
+    for i, (s, g) in enumerate(zip(sizes, grads)):
+        p0, m0, v0 = refs[i]
+        gf = g.float()
+        m_ref = b1 * m0 + (1 - b1) * gf
+        v_ref = b2 * v0 + (1 - b2) * gf * gf
+        denom = (v_ref / (1 - b2 ** step)).sqrt() + eps
+        p_ref = p0 - lr * (m_ref / (1 - b1 ** step) / denom + wd * p0)
+        assert torch.allclose(params[i], p_ref, atol=1e-5, rtol=1e-5), f"tensor {i} param"
+        assert torch.allclose(ms[i], m_ref, atol=1e-6), f"tensor {i} m"
+        assert torch.allclose(vs[i], v_ref, atol=1e-6), f"tensor {i} v"
+        if mirrors[i] is not None:
+            assert torch.allclose(mirrors[i].float(), p_ref, atol=0.01), f"tensor {i} mirror"
+
+
+def test_multi_tensor_accumulate_matches_torch():
+    from hivemind_amd.ops import hip_ops
+
+    torch.manual_seed(12)
+    sizes = [5, 10000, 63]
+    accs = [torch.randn(s, device="cuda") for s in sizes]
+    xs = [torch.randn(s, device="cuda", dtype=torch.bfloat16 if i == 1 else torch.float32)
+          for i, s in enumerate(sizes)]
+    refs = [a + 0.25 * x.float() for a, x in zip(accs, xs)]
+    hip_ops().multi_accumulate_(accs, xs, 0.25)
+    torch.cuda.synchronize()
+    for a, r in zip(accs, refs):
+        assert torch.allclose(a, r, atol=1e-6)
