@@ -135,6 +135,13 @@ class AllReduceRunner:
                 self.tensor_part_container.finalize()
                 return
             my_mode = self.modes[self.ordered_peer_ids.index(self.peer_id)]
+            my_fraction = self.peer_fractions[self.ordered_peer_ids.index(self.peer_id)]
+            if my_fraction > 0:
+                # reducer side: a sender that dies after matchmaking but BEFORE
+                # opening its stream would stall our parts forever -- ban every
+                # sender that has not shown up within sender_timeout (reference
+                # allreduce.py:192-199)
+                pending_tasks.add(asyncio.create_task(self._ban_missing_senders()))
             if my_mode != AveragingMode.AUX:
                 for peer_index, (peer_id, fraction) in enumerate(zip(self.ordered_peer_ids, self.peer_fractions)):
                     if fraction > 0:
@@ -262,6 +269,21 @@ class AllReduceRunner:
             if sender_index is not None:
                 await self._ban_sender(sender_peer_id, sender_index)
             yield AveragingData(code=DataCode.ERROR, group_id=self.group_id)
+
+    async def _ban_missing_senders(self):
+        """Ban senders that never opened their rpc_aggregate_part stream."""
+        try:
+            # if every part reduces before the deadline, nobody is missing
+            await asyncio.wait_for(self.tensor_part_reducer.finished.wait(), timeout=self.sender_timeout)
+            return
+        except asyncio.TimeoutError:
+            pass
+        for sender_index, sender_peer_id in enumerate(self.sender_peer_ids):
+            if sender_peer_id == self.peer_id:
+                continue  # local parts accumulate synchronously
+            if sender_peer_id not in self.active_senders:
+                logger.debug(f"{self}: sender {sender_peer_id} never showed up; banning")
+                await self._ban_sender(sender_peer_id, sender_index)
 
     async def _ban_sender(self, sender_peer_id: PeerID, sender_index: int):
         """Exclude a failed sender from the rest of this round (reference allreduce.py:317-321)."""
