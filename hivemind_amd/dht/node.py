@@ -130,6 +130,7 @@ class DHTNode:
         cache_on_store: bool = True,
         cache_nearest: int = 1,
         cache_size: int = 10000,
+        cache_refresh_before_expiry: float = 5.0,
         reuse_get_requests: bool = True,
         listen_host: str = "127.0.0.1",
         port: int = 0,
@@ -144,6 +145,13 @@ class DHTNode:
         self.num_replicas, self.num_workers = num_replicas, num_workers
         self.beam_size = beam_size if beam_size is not None else bucket_size
         self.cache_locally, self.cache_on_store, self.cache_nearest = cache_locally, cache_on_store, cache_nearest
+        # proactive cache refresh (reference node.py cache_refresh_before_expiry):
+        # serving a cached value that is about to expire schedules a background
+        # re-fetch so hot keys stay warm
+        self.cache_refresh_before_expiry = cache_refresh_before_expiry
+        self.cache_refresh_queue: Dict[DHTID, DHTExpiration] = {}
+        self._cache_refresh_task: Optional[asyncio.Task] = None
+        self._cache_refresh_event = asyncio.Event()
         self.reuse_get_requests = reuse_get_requests
         self.blacklist = Blacklist(blacklist_time, backoff_rate)
         if p2p is None:
@@ -406,6 +414,7 @@ class DHTNode:
         num_workers: Optional[int] = None,
         beam_size: Optional[int] = None,
         return_futures: bool = False,
+        _is_refresh: bool = False,
     ) -> Dict[DHTID, Union[Optional[ValueWithExpiration], asyncio.Future]]:
         """Traverse the DHT for each key, returning the freshest value found
         (reference node.py:569-812)."""
@@ -430,7 +439,15 @@ class DHTNode:
         # check local storage and cache first
         for key_id, search in search_results.items():
             search.add_candidate(self.protocol.storage.get(key_id), source_node_id=self.node_id)
-            search.add_candidate(self.protocol.cache.get(key_id), source_node_id=self.node_id)
+            cached = self.protocol.cache.get(key_id)
+            search.add_candidate(cached, source_node_id=self.node_id)
+            if (
+                not _is_refresh  # a refresh fetch must not re-schedule itself
+                and cached is not None
+                and self.cache_refresh_before_expiry > 0
+                and cached.expiration_time - get_dht_time() < self.cache_refresh_before_expiry
+            ):
+                self._schedule_cache_refresh(key_id, cached.expiration_time)
 
         unfinished_ids = [key_id for key_id, s in search_results.items() if not s.finished]
         node_to_peer: Dict[DHTID, PeerInfo] = {}
@@ -493,6 +510,45 @@ class DHTNode:
             if traverse_task is not None:
                 traverse_task.cancel()
             raise
+
+    def _schedule_cache_refresh(self, key_id: DHTID, expiration_time: DHTExpiration):
+        """Queue a background re-fetch of a soon-to-expire cached key
+        (reference node.py cache_refresh_queue)."""
+        if key_id not in self.cache_refresh_queue:
+            self.cache_refresh_queue[key_id] = expiration_time
+            self._cache_refresh_event.set()
+            if self._cache_refresh_task is None or self._cache_refresh_task.done():
+                self._cache_refresh_task = asyncio.create_task(self._refresh_stale_cache_entries())
+
+    async def _refresh_stale_cache_entries(self):
+        while self.is_alive:
+            if not self.cache_refresh_queue:
+                self._cache_refresh_event.clear()
+                try:
+                    await asyncio.wait_for(self._cache_refresh_event.wait(), timeout=60)
+                except asyncio.TimeoutError:
+                    return  # idle: let the task die; next hit restarts it
+                continue
+            key_id, expiration = min(self.cache_refresh_queue.items(), key=lambda kv: kv[1])
+            wait = (expiration - get_dht_time()) - self.cache_refresh_before_expiry
+            if wait > 0:
+                try:
+                    await asyncio.wait_for(self._cache_refresh_event.wait(), timeout=wait)
+                    continue  # new entries may be more urgent
+                except asyncio.TimeoutError:
+                    pass
+            self.cache_refresh_queue.pop(key_id, None)
+            try:
+                # re-fetch, demanding something fresher than the cached copy so
+                # the search goes to the network; the search path re-populates
+                # the cache via cache_locally (reference node.py:711-761)
+                await self.get_many_by_id(
+                    [key_id],
+                    sufficient_expiration_time=max(get_dht_time(), expiration) + self.cache_refresh_before_expiry,
+                    _is_refresh=True,
+                )
+            except Exception as e:
+                logger.debug(f"cache refresh for {key_id} failed: {e!r}")
 
     def _cache_new_result(self, search: _SearchState, nearest_nodes: List[DHTID], node_to_peer: Dict[DHTID, PeerInfo]):
         """After a search: cache locally and/or on the nearest nodes (reference node.py:763-794)."""

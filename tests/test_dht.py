@@ -259,3 +259,30 @@ def test_dht_survives_peer_failures():
     assert res is not None and res.value == 42
     for d in dhts[:5]:
         d.shutdown()
+
+
+def test_cache_refresh_before_expiry():
+    """Reading a cached value close to expiry schedules a background re-fetch
+    that keeps the cache warm (reference test_dht_node.py:187 caching)."""
+    root = DHT(start=True)
+    peer = DHT(initial_peers=[root.endpoint], start=True)
+    time.sleep(0.3)
+    now = get_dht_time()
+    assert root.store("hot_key", 42, now + 4.0)
+    time.sleep(0.3)
+    # the get caches hot_key on `peer`; expiration (in ~3.7s) is inside the
+    # default refresh window (5s), so a background refresh gets scheduled
+    res = peer.get("hot_key", latest=True)
+    assert res is not None and res.value == 42
+    # owner extends the record's life; the refresher should pick the new copy up
+    assert root.store("hot_key", 43, now + 60)
+    deadline = time.monotonic() + 10
+    refreshed = None
+    while time.monotonic() < deadline:
+        refreshed = peer.get("hot_key")  # non-latest: cache is allowed
+        if refreshed is not None and refreshed.value == 43:
+            break
+        time.sleep(0.5)
+    assert refreshed is not None and refreshed.value == 43, refreshed
+    peer.shutdown()
+    root.shutdown()
