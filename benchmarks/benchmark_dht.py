@@ -2,9 +2,15 @@
 """DHT store/get latency benchmark (reference benchmarks/benchmark_dht.py).
 
 Reference headline: mean store 14.87 ms / mean get 6.64 ms at
-1024 peers, 16384 keys, batch 64 on one machine.
+1024 peers, 16384 keys, batch 64 on one machine — with every peer its own OS
+process. This benchmark shards peers across ``--host-processes`` worker
+processes (each hosting peers on in-process event-loop threads) so the swarm
+gets real CPU parallelism like the reference's per-process peers; with
+``--host-processes 1`` everything runs in one process (GIL-serialized swarm —
+only useful for small sanity runs).
 
-  python benchmarks/benchmark_dht.py --num_peers 32 --num_experts 256 --expert_batch_size 32
+  python benchmarks/benchmark_dht.py --num_peers 1024 --num_experts 16384 \
+      --expert_batch_size 64 --increase_file_limit
 """
 
 import argparse
@@ -17,13 +23,57 @@ import time
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 
+def _host_main(rank: int, n_local: int, seed_endpoints, conn, wait_timeout: float):
+    """Worker process: hosts `n_local` DHT peers and serves store/get commands."""
+    random.seed(1000 + rank)
+    from hivemind_amd import DHT
+    from hivemind_amd.moe.server.dht_handler import declare_experts, get_expert_infos
+    from hivemind_amd.utils.timed_storage import get_dht_time
+
+    peers = []
+    local_endpoints = []
+    try:
+        for i in range(n_local):
+            pool = seed_endpoints + local_endpoints
+            kwargs = {"initial_peers": random.sample(pool, min(2, len(pool)))} if pool else {}
+            peers.append(DHT(start=True, wait_timeout=wait_timeout, **kwargs))
+            local_endpoints.append(peers[-1].endpoint)
+        conn.send(("ready", local_endpoints))
+        while True:
+            cmd = conn.recv()
+            if cmd[0] == "stop":
+                break
+            if cmd[0] == "store":
+                _, uids, expiration = cmd
+                peer = random.choice(peers)
+                t0 = time.perf_counter()
+                result = declare_experts(peer, uids, expiration_time=get_dht_time() + expiration)
+                dt = time.perf_counter() - t0
+                conn.send((dt, sum(bool(v) for v in result.values()), len(result)))
+            elif cmd[0] == "get":
+                _, uids = cmd
+                peer = random.choice(peers)
+                t0 = time.perf_counter()
+                infos = get_expert_infos(peer, uids)
+                dt = time.perf_counter() - t0
+                conn.send((dt, sum(info is not None for info in infos), len(uids)))
+            elif cmd[0] == "alive":
+                conn.send(sum(p.is_alive for p in peers))
+    finally:
+        for p in peers:
+            try:
+                p.shutdown()
+            except Exception:
+                pass
+
+
 def main():
     parser = argparse.ArgumentParser()
     parser.add_argument("--num_peers", type=int, default=32)
-    parser.add_argument("--initial_peers", type=int, default=2)
+    parser.add_argument("--host-processes", type=int, default=min(8, os.cpu_count() or 1))
     parser.add_argument("--num_experts", type=int, default=256)
     parser.add_argument("--expert_batch_size", type=int, default=32)
-    parser.add_argument("--expiration", type=float, default=300.0)
+    parser.add_argument("--expiration", type=float, default=9999.0)
     parser.add_argument("--wait_timeout", type=float, default=5.0)
     parser.add_argument("--increase_file_limit", action="store_true")
     args = parser.parse_args()
@@ -33,68 +83,103 @@ def main():
 
         increase_file_limit()
 
-    from hivemind_amd import DHT
-    from hivemind_amd.moe.server.dht_handler import declare_experts, get_expert_infos
-    from hivemind_amd.utils.timed_storage import get_dht_time
+    import multiprocessing as mp
 
-    print(f"spawning {args.num_peers} DHT peers...", file=sys.stderr)
-    peers = [DHT(start=True, wait_timeout=args.wait_timeout)]
-    for _ in range(args.num_peers - 1):
-        initial = random.sample([p.endpoint for p in peers], min(args.initial_peers, len(peers)))
-        peers.append(DHT(initial_peers=initial, start=True, wait_timeout=args.wait_timeout))
+    ctx = mp.get_context("spawn")
+    n_hosts = max(1, min(args.host_processes, args.num_peers))
+    per_host = [args.num_peers // n_hosts + (1 if r < args.num_peers % n_hosts else 0)
+                for r in range(n_hosts)]
 
+    t_spawn = time.perf_counter()
+    # the first host provides seed endpoints for the rest
+    conns, procs = [], []
+    c0, child0 = ctx.Pipe()
+    p0 = ctx.Process(target=_host_main, args=(0, per_host[0], [], child0, args.wait_timeout), daemon=True)
+    p0.start()
+    conns.append(c0)
+    procs.append(p0)
+    msg = c0.recv()
+    assert msg[0] == "ready"
+    endpoints = list(msg[1])
+    for rank in range(1, n_hosts):
+        c, child = ctx.Pipe()
+        seeds = random.sample(endpoints, min(8, len(endpoints)))
+        p = ctx.Process(target=_host_main, args=(rank, per_host[rank], seeds, child, args.wait_timeout), daemon=True)
+        p.start()
+        conns.append(c)
+        procs.append(p)
+    for c in conns[1:]:
+        msg = c.recv()
+        assert msg[0] == "ready"
+        endpoints.extend(msg[1])
+    spawn_s = time.perf_counter() - t_spawn
+    print(f"spawned {len(endpoints)} peers across {n_hosts} processes in {spawn_s:.1f}s",
+          file=sys.stderr, flush=True)
+
+    random.seed(42)
     expert_uids = [f"expert.{i}.{random.randint(0, 255)}" for i in range(args.num_experts)]
     random.shuffle(expert_uids)
 
-    # stores
-    store_times, successes = [], 0
-    total_stores = 0
+    store_times, successes, total_stores = [], 0, 0
+    t_phase = time.perf_counter()
     for start in range(0, args.num_experts, args.expert_batch_size):
         batch = expert_uids[start : start + args.expert_batch_size]
-        store_peer = random.choice(peers)
-        t0 = time.perf_counter()
-        result = declare_experts(store_peer, batch, expiration_time=get_dht_time() + args.expiration)
-        dt = time.perf_counter() - t0
+        conn = random.choice(conns)
+        conn.send(("store", batch, args.expiration))
+        dt, ok, total = conn.recv()
         store_times.append(dt / max(len(batch), 1))
-        successes += sum(bool(v) for v in result.values())
-        total_stores += len(result)
+        successes += ok
+        total_stores += total
+    store_phase_s = time.perf_counter() - t_phase
+    print(f"store phase: {store_phase_s:.1f}s", file=sys.stderr, flush=True)
 
-    # gets
     get_times, get_successes, total_gets = [], 0, 0
+    t_phase = time.perf_counter()
     for start in range(0, args.num_experts, args.expert_batch_size):
         batch = expert_uids[start : start + args.expert_batch_size]
-        get_peer = random.choice(peers)
-        t0 = time.perf_counter()
-        infos = get_expert_infos(get_peer, batch)
-        dt = time.perf_counter() - t0
+        conn = random.choice(conns)
+        conn.send(("get", batch))
+        dt, ok, total = conn.recv()
         get_times.append(dt / max(len(batch), 1))
-        get_successes += sum(info is not None for info in infos)
-        total_gets += len(batch)
+        get_successes += ok
+        total_gets += total
+    get_phase_s = time.perf_counter() - t_phase
 
-    alive = sum(p.is_alive for p in peers)
+    alive = 0
+    for c in conns:
+        c.send(("alive",))
+        alive += c.recv()
+
     result = {
         "metric": "DHT store/get latency",
         "num_peers": args.num_peers,
+        "host_processes": n_hosts,
         "num_experts": args.num_experts,
+        "expert_batch_size": args.expert_batch_size,
         "mean_store_ms": round(1000 * sum(store_times) / len(store_times), 2),
         "mean_get_ms": round(1000 * sum(get_times) / len(get_times), 2),
         "store_success_rate": round(successes / max(total_stores, 1), 4),
         "get_success_rate": round(get_successes / max(total_gets, 1), 4),
         "node_survival_rate": round(alive / args.num_peers, 4),
+        "spawn_s": round(spawn_s, 1),
+        "store_phase_s": round(store_phase_s, 1),
+        "get_phase_s": round(get_phase_s, 1),
         "vs_baseline_store_14.87ms": round(14.87 / (1000 * sum(store_times) / len(store_times)), 2),
         "vs_baseline_get_6.64ms": round(6.64 / (1000 * sum(get_times) / len(get_times)), 2),
     }
     print(json.dumps(result), flush=True)
-    for p in peers:
-        p.shutdown()
+    for c in conns:
+        try:
+            c.send(("stop",))
+        except Exception:
+            pass
+    deadline = time.monotonic() + 15
+    for p in procs:
+        p.join(timeout=max(0.1, deadline - time.monotonic()))
+    for p in procs:
+        if p.is_alive():
+            p.terminate()
 
 
 if __name__ == "__main__":
-    try:
-        main()
-    except BaseException:
-        import traceback
-
-        traceback.print_exc()
-        os._exit(1)
-    os._exit(0)
+    main()

@@ -98,14 +98,20 @@ class RpcContext:
 def _tune_socket_buffers(writer: asyncio.StreamWriter, size: int = 4 * 1024 * 1024):
     """Unix sockets default to ~200 KB buffers; multi-MB tensor frames then pay
     many extra wakeups. Loopback TCP autotunes, AF_UNIX does not -- set both
-    directions explicitly (best effort)."""
+    directions explicitly (best effort). TCP also gets TCP_NODELAY: the RPC
+    protocol is small header+payload frame pairs, and Nagle + delayed-ACK
+    stalls sequential DHT lookup chains."""
     try:
         import socket as _socket
 
         sock = writer.get_extra_info("socket")
-        if sock is not None and sock.family == _socket.AF_UNIX:
+        if sock is None:
+            return
+        if sock.family == _socket.AF_UNIX:
             sock.setsockopt(_socket.SOL_SOCKET, _socket.SO_SNDBUF, size)
             sock.setsockopt(_socket.SOL_SOCKET, _socket.SO_RCVBUF, size)
+        elif sock.family in (_socket.AF_INET, _socket.AF_INET6):
+            sock.setsockopt(_socket.IPPROTO_TCP, _socket.TCP_NODELAY, 1)
     except Exception:
         pass
 

@@ -289,3 +289,23 @@ def test_dial_by_id_rejects_wrong_peer():
         await server.shutdown()
 
     run(main())
+
+
+def test_ed25519_backends_interop():
+    """The libcrypto fast path and the pure-Python reference implementation
+    must produce interchangeable keys and signatures (deterministic Ed25519:
+    identical bytes)."""
+    from hivemind_amd.utils import crypto as c
+
+    secret = bytes(range(32))
+    backend = c._LibCrypto.get()
+    if backend is None:
+        pytest.skip("libcrypto unavailable; pure-Python backend active")
+    assert backend.public_key(secret) == c.ed25519_public_key(secret)
+    msg = b"interop-check"
+    assert backend.sign(secret, msg) == c.ed25519_sign(secret, msg)
+    sig = backend.sign(secret, msg)
+    pub = backend.public_key(secret)
+    assert c.ed25519_verify(pub, msg, sig)
+    assert backend.verify(pub, msg, c.ed25519_sign(secret, msg))
+    assert not backend.verify(pub, msg + b"x", sig)
