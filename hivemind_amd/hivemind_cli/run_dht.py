@@ -35,7 +35,9 @@ def main():
     parser.add_argument("--port", type=int, default=0)
     parser.add_argument("--initial_peers", nargs="*", default=[], help="host:port of existing peers")
     parser.add_argument("--refresh_period", type=float, default=30.0)
-    args = parser.parse_args()
+    from ..utils.config import parse_args_with_config
+
+    args = parse_args_with_config(parser)
 
     dht = DHT(
         initial_peers=args.initial_peers,

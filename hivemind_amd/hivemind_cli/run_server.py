@@ -38,7 +38,9 @@ def main():
     parser.add_argument("--stats_report_interval", type=float, default=60.0)
     parser.add_argument("--relay_endpoint", type=str, default=None,
                         help="host:port of a public relay peer; serve through it when behind NAT (no inbound sockets)")
-    args = parser.parse_args()
+    from ..utils.config import parse_args_with_config
+
+    args = parse_args_with_config(parser)
 
     optim_cls = {"adam": torch.optim.Adam, "sgd": torch.optim.SGD, "none": None}[args.optimizer]
     server = Server.create(

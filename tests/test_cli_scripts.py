@@ -76,3 +76,34 @@ def test_run_server_starts():
     finally:
         proc.terminate()
         proc.wait(10)
+
+
+def test_config_yaml_defaults_and_precedence(tmp_path):
+    """--config config.yml provides defaults; explicit CLI flags win
+    (reference run_server.py:21-22 configargparse semantics)."""
+    import argparse
+
+    from hivemind_amd.utils.config import parse_args_with_config
+
+    cfg = tmp_path / "config.yml"
+    cfg.write_text("hidden_dim: 2048\nexpert_cls: transformer\n")
+
+    def build():
+        p = argparse.ArgumentParser()
+        p.add_argument("--hidden_dim", type=int, default=1024)
+        p.add_argument("--expert_cls", type=str, default="ffn")
+        return p
+
+    args = parse_args_with_config(build(), ["--config", str(cfg)])
+    assert args.hidden_dim == 2048 and args.expert_cls == "transformer"
+    args = parse_args_with_config(build(), ["--config", str(cfg), "--hidden_dim", "4096"])
+    assert args.hidden_dim == 4096 and args.expert_cls == "transformer"
+    args = parse_args_with_config(build(), [])
+    assert args.hidden_dim == 1024
+    # unknown keys are rejected loudly
+    bad = tmp_path / "bad.yml"
+    bad.write_text("no_such_flag: 1\n")
+    import pytest as _pytest
+
+    with _pytest.raises(ValueError):
+        parse_args_with_config(build(), ["--config", str(bad)])
