@@ -134,10 +134,20 @@ class _Connection:
         self.serving_tasks: Dict[bytes, asyncio.Task] = {}
         self.reader_task: Optional[asyncio.Task] = None
         self.closed = asyncio.Event()
+        import time as _time
+
+        self.last_used = _time.monotonic()  # for LRU eviction (connection manager)
+
+    def is_busy(self) -> bool:
+        """True if calls are in flight (must not be evicted)."""
+        return bool(self.pending_unary or self.inbound_streams or self.serving_streams or self.serving_tasks)
 
     async def send_frame(self, ftype: int, call_id: bytes, handler: str = "", payload: bytes = b""):
         # header and payload travel as separate buffers: a multi-MB tensor part
         # is never re-wrapped through msgpack (zero extra copies on the hot path)
+        import time as _time
+
+        self.last_used = _time.monotonic()
         header = MSGPackSerializer.dumps((ftype, call_id, handler))
         async with self.send_lock:
             self.writer.write(struct.pack(">II", len(header), len(payload)))
@@ -217,10 +227,17 @@ class P2P:
         identity: Optional[PrivateKey] = None,
         listen: bool = True,
         relay_endpoint: Optional[str] = None,
+        max_connections: Optional[int] = None,
     ) -> "P2P":
+        if max_connections is None:
+            max_connections = int(os.environ.get("HIVEMIND_MAX_CONNECTIONS", 512))
         self = cls()
         self._identity = identity if identity is not None else PrivateKey()
         self.peer_id = PeerID.from_identity(self._identity)
+        # connection manager (reference: p2pd's connManager high-water mark):
+        # beyond this many cached connections, idle LRU ones are closed --
+        # without it a large DHT swarm exhausts file descriptors
+        self._max_connections = max_connections
         self._listen_host = listen_host
         self._listen = listen
         if listen:
@@ -551,6 +568,28 @@ class P2P:
         existing = self._connections.get(conn.remote_id)
         if existing is None or existing.closed.is_set():
             self._connections[conn.remote_id] = conn
+        self._enforce_connection_limit()
+
+    def _enforce_connection_limit(self):
+        """Close idle least-recently-used connections beyond the high-water
+        mark (the reference's libp2p connection manager)."""
+        if len(self._connections) <= self._max_connections:
+            return
+        evictable = sorted(
+            (c for c in self._connections.values() if not c.is_busy() and not c.closed.is_set()),
+            key=lambda c: c.last_used,
+        )
+        excess = len(self._connections) - self._max_connections
+        for conn in evictable[:excess]:
+            peer = conn.remote_id
+            logger.debug(f"connection manager: evicting idle connection to {peer}")
+            self._connections.pop(peer, None)
+            task = asyncio.ensure_future(conn.close())
+            self._relay_splices.add(task)  # anchor until done
+            task.add_done_callback(self._relay_splices.discard)
+        # drop closed leftovers from the table while we're here
+        for peer in [p for p, c in self._connections.items() if c.closed.is_set()]:
+            self._connections.pop(peer, None)
 
     async def _connect(self, peer: Union[PeerInfo, PeerID], endpoint: Optional[str] = None) -> _Connection:
         if isinstance(peer, PeerInfo):

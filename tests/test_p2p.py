@@ -309,3 +309,29 @@ def test_ed25519_backends_interop():
     assert c.ed25519_verify(pub, msg, sig)
     assert backend.verify(pub, msg, c.ed25519_sign(secret, msg))
     assert not backend.verify(pub, msg + b"x", sig)
+
+
+def test_connection_manager_evicts_idle_lru():
+    """Beyond max_connections, idle least-recently-used connections close
+    (reference p2pd connManager; a 1024-peer swarm exhausted fds without it)."""
+
+    async def main():
+        server = await P2P.create(max_connections=3)
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return payload
+
+        server.add_unary_handler("echo", echo)
+        clients = [await P2P.create() for _ in range(6)]
+        for i, c in enumerate(clients):
+            assert await c.call_unary(server.peer_info, "echo", bytes([i]), timeout=5) == bytes([i])
+        await asyncio.sleep(0.2)
+        live = [c for c in server._connections.values() if not c.closed.is_set()]
+        assert len(live) <= 3, f"{len(live)} live connections cached, cap is 3"
+        # evicted clients can still call again (fresh dial + handshake)
+        assert await clients[0].call_unary(server.peer_info, "echo", b"again", timeout=5) == b"again"
+        for c in clients:
+            await c.shutdown()
+        await server.shutdown()
+
+    run(main())
