@@ -554,7 +554,14 @@ class P2P:
             host, port = split_endpoint(endpoint)
             reader, writer = await asyncio.wait_for(asyncio.open_connection(host, port, limit=STREAM_BUFFER_LIMIT), timeout=10)
         conn = _Connection(self, reader, writer)
-        await self._handshake_outbound(conn)
+        try:
+            await self._handshake_outbound(conn)
+        except Exception:
+            try:
+                writer.close()
+            except Exception:
+                pass
+            raise
         self._endpoint_book[conn.remote_id] = endpoint
         existing = self._connections.get(conn.remote_id)
         if existing is not None and not existing.closed.is_set():
@@ -636,6 +643,12 @@ class P2P:
                     return conn
                 except Exception as e:
                     last_exc = e
+                    if conn is not None:  # failed dial/handshake: release the fd
+                        try:
+                            conn.writer.close()
+                        except Exception:
+                            pass
+                        conn = None
                     continue
         raise P2PDaemonError(f"could not connect to {peer_id} via {endpoints}: {last_exc}")
 
@@ -698,6 +711,10 @@ class P2P:
             if self._connections.get(conn.remote_id) is conn:
                 self._connections.pop(conn.remote_id, None)
             conn.fail_all(P2PDaemonError(f"connection to {conn.remote_id} lost"))
+            try:
+                conn.writer.close()  # fail_all only fails futures; release the fd
+            except Exception:
+                pass
 
     # -------------------------------------------------------------- serving
 
