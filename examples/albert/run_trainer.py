@@ -43,6 +43,10 @@ def main():
     parser.add_argument("--batch_size", type=int, default=128)
     parser.add_argument("--seq_len", type=int, default=512)
     parser.add_argument("--lr", type=float, default=0.00176)
+    parser.add_argument("--optimizer", choices=["lamb", "adamw"], default="lamb",
+                        help="lamb = the reference recipe (LAMB + gradient clipping, "
+                             "examples/albert/run_trainer.py:266); adamw = FusedAdamW")
+    parser.add_argument("--clip_grad_norm", type=float, default=1.0)
     parser.add_argument("--max_epochs", type=int, default=10**9)
     parser.add_argument("--statistics_every", type=int, default=10)
     parser.add_argument("--backup_every_epochs", type=int, default=10)
@@ -76,12 +80,24 @@ def main():
         config.dtype = torch.float32
     model = AlbertForMaskedLM(config).to(device)
 
+    # the reference collaborative-ALBERT recipe trains with LAMB + gradient
+    # clipping (reference run_trainer.py:266 Lamb + ClippingWrapper)
+    if args.optimizer == "lamb":
+        from hivemind_amd.moe.server.layers.optim import ClippingWrapper
+        from hivemind_amd.ops import Lamb
+
+        opt_factory = ClippingWrapper.create(
+            Lamb, lr=args.lr, weight_decay=0.01, clip_grad_norm=args.clip_grad_norm
+        )
+    else:
+        opt_factory = lambda pg: FusedAdamW(pg, lr=args.lr, weight_decay=0.01)
+
     opt = Optimizer(
         dht=dht,
         run_id=args.run_id,
         target_batch_size=args.target_batch_size,
         batch_size_per_step=args.batch_size,
-        optimizer=lambda pg: FusedAdamW(pg, lr=args.lr, weight_decay=0.01),
+        optimizer=opt_factory,
         params=[{"params": list(model.parameters())}],
         offload_optimizer=True,
         delay_optimizer_step=True,

@@ -1,5 +1,6 @@
 """CLI smoke tests (reference tests/test_cli_scripts.py, test_start_server.py)."""
 
+import os
 import re
 import subprocess
 import sys
@@ -107,3 +108,25 @@ def test_config_yaml_defaults_and_precedence(tmp_path):
 
     with _pytest.raises(ValueError):
         parse_args_with_config(build(), ["--config", str(bad)])
+
+
+def test_albert_trainer_lamb_recipe_smoke(tmp_path):
+    """The collaborative-ALBERT example runs the reference recipe: LAMB with
+    gradient clipping through hivemind_amd.Optimizer (reference
+    examples/albert/run_trainer.py:266). 1-peer swarm, tiny shapes, 2 epochs."""
+    import subprocess
+    import sys
+
+    proc = subprocess.run(
+        [
+            sys.executable, "examples/albert/run_trainer.py",
+            "--optimizer", "lamb", "--clip_grad_norm", "1.0",
+            "--batch_size", "4", "--seq_len", "32", "--target_batch_size", "8",
+            "--max_epochs", "2", "--state_path", str(tmp_path / "state.pt"),
+            "--backup_every_epochs", "0",
+        ],
+        capture_output=True, text=True, timeout=420,
+        cwd=os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+    )
+    assert proc.returncode == 0, f"trainer failed:\n{proc.stderr[-2000:]}"
+    assert "epoch 2" in proc.stderr or "epoch 2" in proc.stdout, proc.stderr[-1500:]
