@@ -358,7 +358,10 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if q.is_cuda:
-        return _FlashAttention.apply(q.contiguous(), k.contiguous(), v.contiguous(), causal, float(scale))
+        # strided [B,H,S,D] views (e.g. the model's qkv-GEMM slices) feed the
+        # kernels directly; the binding falls back to .contiguous() only when
+        # the last dim is non-contiguous or rows are misaligned
+        return _FlashAttention.apply(q, k, v, causal, float(scale))
     return torch.nn.functional.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale)
 
 

@@ -13,15 +13,19 @@ extern "C" void launch_gemm_bt_bf16(const void*, const void*, const void*, void*
 extern "C" void launch_mfma_probe(const void*, const void*, void*, void*);
 extern "C" void launch_stage_probe(const void*, void*, int, void*);
 extern "C" void launch_flash_fwd(const void*, const void*, const void*, void*, void*,
-                                 int, int, int, int, int, float, int, void*);
+                                 int, int, int, int, int, float, int,
+                                 const long long*, const long long*, void*);
 extern "C" void launch_flash_delta(const void*, const void*, void*, long long, int, void*);
 extern "C" void launch_flash_bwd_dkdv(const void*, const void*, const void*, const void*,
                                       const void*, const void*, const void*, const void*,
-                                      void*, void*, int, int, int, int, float, int, void*);
+                                      void*, void*, int, int, int, int, float, int,
+                                      const long long*, const long long*, const long long*, void*);
 extern "C" void launch_flash_bwd_dq(const void*, const void*, const void*, const void*,
                                     const void*, const void*, const void*, void*,
-                                    int, int, int, int, float, int, void*);
-extern "C" void launch_transpose_bhsd(const void*, void*, long long, int, int, void*);
+                                    int, int, int, int, float, int,
+                                    const long long*, const long long*, const long long*, void*);
+extern "C" void launch_transpose_bhsd(const void*, void*, long long, int, int, int,
+                                      long long, long long, long long, void*);
 
 #include "multi_tensor.h"
 
@@ -545,40 +549,65 @@ static void check_flash_shapes(const torch::Tensor& t, const char* name) {
   TORCH_CHECK(t.dim() == 4, name, " must be [B, H, S, D]");
 }
 
+// strided [B,H,S,D] views are fine as long as the last dim is contiguous and
+// every row start stays 16B-aligned (the staging loads are 16B vectors); the
+// model's qkv-GEMM views qualify, so q/k/v arrive with ZERO copies
+static bool flash_strided_ok(const torch::Tensor& t) {
+  return t.is_cuda() && t.dim() == 4 && t.scalar_type() == torch::kBFloat16 &&
+         t.stride(3) == 1 && t.stride(2) % 8 == 0 && t.stride(1) % 8 == 0 &&
+         t.stride(0) % 8 == 0 && (reinterpret_cast<uintptr_t>(t.data_ptr()) % 16 == 0);
+}
+
+static torch::Tensor flash_input(const torch::Tensor& t) {
+  return flash_strided_ok(t) ? t : t.contiguous();
+}
+
+static void flash_strides(const torch::Tensor& t, long long out[3]) {
+  out[0] = t.stride(0);
+  out[1] = t.stride(1);
+  out[2] = t.stride(2);
+}
+
 torch::Tensor transpose_bhsd(torch::Tensor t) {
   // [B,H,S,D] -> [B,H,D,S] via the in-register 8x8 transpose kernel
-  check_flash_shapes(t, "transpose input");
-  int B = t.size(0), H = t.size(1), S = t.size(2), D = t.size(3);
+  auto tc = flash_input(t);
+  TORCH_CHECK(tc.is_cuda() && tc.dim() == 4 && tc.scalar_type() == torch::kBFloat16,
+              "transpose input must be CUDA bf16 [B,H,S,D]");
+  int B = tc.size(0), H = tc.size(1), S = tc.size(2), D = tc.size(3);
   TORCH_CHECK(S % 64 == 0 && D % 64 == 0, "transpose_bhsd needs S, D multiples of 64");
-  auto out = torch::empty({B, H, D, S}, t.options());
-  launch_transpose_bhsd(t.data_ptr(), out.data_ptr(), (long long)B * H, S, D,
-                        (void*)current_stream());
+  auto out = torch::empty({B, H, D, S}, tc.options());
+  launch_transpose_bhsd(tc.data_ptr(), out.data_ptr(), (long long)B * H, H, S, D,
+                        tc.stride(0), tc.stride(1), tc.stride(2), (void*)current_stream());
   return out;
 }
 
-std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q, torch::Tensor k, torch::Tensor v,
-                                          bool causal, double scale) {
-  check_flash_shapes(q, "q"); check_flash_shapes(k, "k"); check_flash_shapes(v, "v");
+std::vector<torch::Tensor> flash_attn_fwd(torch::Tensor q_in, torch::Tensor k_in,
+                                          torch::Tensor v_in, bool causal, double scale) {
+  auto q = flash_input(q_in), k = flash_input(k_in), v = flash_input(v_in);
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   int Hkv = k.size(1);
   TORCH_CHECK(D == 64 || D == 128, "flash attention requires head_dim 64 or 128, got ", D);
   TORCH_CHECK(S % 64 == 0, "flash attention requires seq_len % 64 == 0, got ", S);
   TORCH_CHECK(k.size(2) == S && v.size(2) == S, "kv seq_len mismatch");
   TORCH_CHECK(H % Hkv == 0, "H must be a multiple of Hkv");
-  auto o = torch::empty_like(q);
+  auto o = torch::empty({B, H, S, D}, q.options());
   auto lse = torch::empty({B, H, S}, q.options().dtype(torch::kFloat32));
   // one [B,H,S,D]->[B,H,D,S] copy so the PV operand stages with vector
   // LDS writes (in-kernel scalar transposes dominated the tile cost)
   auto vt = transpose_bhsd(v);
+  long long qs[3], ks[3];
+  flash_strides(q, qs); flash_strides(k, ks);
   launch_flash_fwd(q.data_ptr(), k.data_ptr(), vt.data_ptr(), o.data_ptr(), lse.data_ptr(),
-                   B, H, Hkv, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
+                   B, H, Hkv, S, D, (float)scale, causal ? 1 : 0, qs, ks,
+                   (void*)current_stream());
   return {o, lse};
 }
 
-std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor d_out, torch::Tensor q, torch::Tensor k,
-                                          torch::Tensor v, torch::Tensor o, torch::Tensor lse,
+std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor d_out, torch::Tensor q_in,
+                                          torch::Tensor k_in, torch::Tensor v_in,
+                                          torch::Tensor o, torch::Tensor lse,
                                           bool causal, double scale) {
-  check_flash_shapes(q, "q");
+  auto q = flash_input(q_in), k = flash_input(k_in), v = flash_input(v_in);
   int B = q.size(0), H = q.size(1), S = q.size(2), D = q.size(3);
   TORCH_CHECK(k.size(1) == H, "flash attention backward requires H == Hkv (expand kv first)");
   auto d_out_c = d_out.contiguous();
@@ -591,16 +620,20 @@ std::vector<torch::Tensor> flash_attn_bwd(torch::Tensor d_out, torch::Tensor q, 
   auto dot = transpose_bhsd(d_out_c);
   auto qt = transpose_bhsd(q);
   auto kt = transpose_bhsd(k);
-  auto dq = torch::empty_like(q);
-  auto dk = torch::empty_like(k);
-  auto dv = torch::empty_like(v);
+  auto dq = torch::empty({B, H, S, D}, q.options());
+  auto dk = torch::empty({B, H, S, D}, q.options());
+  auto dv = torch::empty({B, H, S, D}, q.options());
+  long long qs[3], ks[3], vs[3];
+  flash_strides(q, qs); flash_strides(k, ks); flash_strides(v, vs);
   launch_flash_bwd_dkdv(d_out_c.data_ptr(), dot.data_ptr(), q.data_ptr(), qt.data_ptr(),
                         k.data_ptr(), v.data_ptr(),
                         lse.data_ptr(), delta.data_ptr(), dk.data_ptr(), dv.data_ptr(),
-                        B, H, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
+                        B, H, S, D, (float)scale, causal ? 1 : 0, qs, ks, vs,
+                        (void*)current_stream());
   launch_flash_bwd_dq(d_out_c.data_ptr(), q.data_ptr(), k.data_ptr(), kt.data_ptr(), v.data_ptr(),
                       lse.data_ptr(), delta.data_ptr(), dq.data_ptr(),
-                      B, H, S, D, (float)scale, causal ? 1 : 0, (void*)current_stream());
+                      B, H, S, D, (float)scale, causal ? 1 : 0, qs, ks, vs,
+                      (void*)current_stream());
   return {dq, dk, dv};
 }
 

@@ -134,17 +134,22 @@ __device__ __forceinline__ void transpose8x8(const unsigned int R[8][4], unsigne
     for (int k = 0; k < 4; ++k) O[c][k] = P[2 * k + (c & 1)][c >> 1];
 }
 
+struct TStrides {
+  long long sb, sh, ss;  // element strides of the [B,H,S,D] input
+};
+
 __global__ __launch_bounds__(64) void transpose_bhsd_kernel(
-    const ushort_t* __restrict__ in,  // [BH, S, D]
-    ushort_t* __restrict__ out,       // [BH, D, S]
-    int S, int D) {
+    const ushort_t* __restrict__ in,  // [B, H, S, D], possibly strided
+    ushort_t* __restrict__ out,       // [BH, D, S] contiguous
+    int H, int S, int D, TStrides istr) {
   __shared__ ushort_t tile[64 * 64];
   const int lane = threadIdx.x;
   const int tiles_d = D / 64;
   const int tile_idx = blockIdx.x;
   const int s0 = (tile_idx / tiles_d) * 64, d0 = (tile_idx % tiles_d) * 64;
-  const long long base = (long long)blockIdx.y * S * D;
-  const long long obase = (long long)blockIdx.y * S * D;
+  const long long bh = blockIdx.y;
+  const long long base = (bh / H) * istr.sb + (bh % H) * istr.sh;
+  const long long obase = bh * S * D;
 
   // lane -> 8x8 block (sb, db) within the 64x64 tile
   const int sb = lane & 7, db = lane >> 3;
@@ -152,7 +157,7 @@ __global__ __launch_bounds__(64) void transpose_bhsd_kernel(
 #pragma unroll
   for (int j = 0; j < 8; ++j)
     *reinterpret_cast<u32x4*>(R[j]) = *reinterpret_cast<const u32x4*>(
-        in + base + (long long)(s0 + sb * 8 + j) * D + d0 + db * 8);
+        in + base + (long long)(s0 + sb * 8 + j) * istr.ss + d0 + db * 8);
   transpose8x8(R, O);
   // park transposed block at (db, sb) -> LDS is the transposed tile, row-major
 #pragma unroll
@@ -171,11 +176,12 @@ __global__ __launch_bounds__(64) void transpose_bhsd_kernel(
   }
 }
 
-extern "C" void launch_transpose_bhsd(const void* in, void* out, long long BH, int S, int D,
+extern "C" void launch_transpose_bhsd(const void* in, void* out, long long BH, int H, int S,
+                                      int D, long long sb, long long sh, long long ss,
                                       void* stream) {
   dim3 grid((S / 64) * (D / 64), (unsigned)BH);
   hipLaunchKernelGGL(transpose_bhsd_kernel, grid, dim3(64), 0, (hipStream_t)stream,
-                     (const ushort_t*)in, (ushort_t*)out, S, D);
+                     (const ushort_t*)in, (ushort_t*)out, H, S, D, TStrides{sb, sh, ss});
 }
 
 // ======================================================================
@@ -184,12 +190,21 @@ extern "C" void launch_transpose_bhsd(const void* in, void* out, long long BH, i
 // grid (S/64, B*H); saves O and logsumexp (LSE = m + log(l), natural units
 // of the scaled scores).
 
+// Strides: q/k are strided [B,H,S,D] views (last dim contiguous) so the
+// model's qkv GEMM output feeds the kernel with ZERO copies; sb/sh/ss are
+// element strides for batch/head/seq. VT is always our own contiguous
+// [B,Hkv,D,S] tensor.
+struct Strides3 {
+  long long sb, sh, ss;
+};
+
 template <int D, bool CAUSAL>
 __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ VT,  // [B,Hkv,D,S] pre-transposed
     ushort_t* __restrict__ O,
-    float* __restrict__ LSE, int B, int H, int Hkv, int S, float scale) {
+    float* __restrict__ LSE, int B, int H, int Hkv, int S, float scale,
+    Strides3 qstr, Strides3 kstr) {
   constexpr int KSTEPS = D / 32;       // MFMA K-steps over the head dim
   constexpr int DFRAGS = D / 16;       // output d-blocks per wave
   constexpr int KITERS = TILE * (D / 8) / NTHREADS;  // staging 16B segs/thread
@@ -204,8 +219,11 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
   const int q_tile = blockIdx.x;
   const long long bh = blockIdx.y;
   const long long b = bh / H, h = bh % H;
-  const long long q_base = (bh * S) * D;
-  const long long kv_base = ((b * Hkv + h / (H / Hkv)) * S) * D;
+  const long long hkv = h / (H / Hkv);
+  const long long q_base = b * qstr.sb + h * qstr.sh;    // strided Q input
+  const long long k_base = b * kstr.sb + hkv * kstr.sh;  // strided K input
+  const long long o_base = (bh * S) * D;                 // contiguous outputs
+  const long long vt_base = ((b * Hkv + hkv) * S) * D;   // contiguous VT
 
   const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
   const int q0 = q_tile * TILE;            // tile's first q row
@@ -215,7 +233,7 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
   // Q fragments in registers for the whole kernel, pre-scaled (Q-hoist)
   bf16x8 q_frag[KSTEPS];
   {
-    const ushort_t* qrow = Q + q_base + (long long)(q0 + wq + fr) * D;
+    const ushort_t* qrow = Q + q_base + (long long)(q0 + wq + fr) * qstr.ss;
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(qrow + ks * 32 + fq * 8);
@@ -241,14 +259,14 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
   // per-thread staging geometry (constant): 16B segment (row, seg) pairs
   const int tid_ = tid;
   bf16x8 k_st[KITERS], v_st[VITERS];
-  const ushort_t* k_src = K + kv_base;
-  const ushort_t* vt_src = VT + kv_base;
+  const ushort_t* k_src = K + k_base;
+  const ushort_t* vt_src = VT + vt_base;
 
   auto load_tile_regs = [&](int s0) {
 #pragma unroll
     for (int i = 0; i < KITERS; ++i) {
       int idx = tid_ + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
-      k_st[i] = *reinterpret_cast<const bf16x8*>(k_src + (long long)(s0 + row) * D + seg * 8);
+      k_st[i] = *reinterpret_cast<const bf16x8*>(k_src + (long long)(s0 + row) * kstr.ss + seg * 8);
     }
 #pragma unroll
     for (int i = 0; i < VITERS; ++i) {
@@ -363,8 +381,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     for (int nd = 0; nd < DFRAGS; nd += 2) {
       ushort_t a, b;
       f2bf2(o_acc[nd][r] * inv_l, o_acc[nd + 1][r] * inv_l, a, b);
-      O[q_base + (long long)qg * D + nd * 16 + fr] = a;
-      O[q_base + (long long)qg * D + (nd + 1) * 16 + fr] = b;
+      O[o_base + (long long)qg * D + nd * 16 + fr] = a;
+      O[o_base + (long long)qg * D + (nd + 1) * 16 + fr] = b;
     }
     if (fr == 0) LSE[bh * S + qg] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
   }
@@ -409,7 +427,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     const ushort_t* __restrict__ K, const ushort_t* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ delta,
     ushort_t* __restrict__ dK, ushort_t* __restrict__ dV,
-    int B, int H, int S, float scale) {
+    int B, int H, int S, float scale,
+    Strides3 qstr, Strides3 kstr, Strides3 vstr) {
   constexpr int KSTEPS = D / 32;
   constexpr int DFRAGS = D / 16;
   // double-buffer at D=64 (single barrier/iter); D=128's four 16 KB tiles
@@ -426,7 +445,11 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
 
   const int kv_tile = blockIdx.x;
   const long long bh = blockIdx.y;
-  const long long base = (bh * S) * D;
+  const long long b_ = bh / H, h_ = bh % H;
+  const long long base = (bh * S) * D;  // contiguous dO/QT/dOT/dK/dV
+  const long long q_base = b_ * qstr.sb + h_ * qstr.sh;
+  const long long k_base = b_ * kstr.sb + h_ * kstr.sh;
+  const long long v_base = b_ * vstr.sb + h_ * vstr.sh;
 
   const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
   const int s0 = kv_tile * TILE;
@@ -436,8 +459,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
   // K (pre-scaled) and V fragments in registers for the whole kernel
   bf16x8 k_frag[KSTEPS], v_frag[KSTEPS];
   {
-    const ushort_t* krow = K + base + (long long)(s0 + ws + fr) * D;
-    const ushort_t* vrow = V + base + (long long)(s0 + ws + fr) * D;
+    const ushort_t* krow = K + k_base + (long long)(s0 + ws + fr) * kstr.ss;
+    const ushort_t* vrow = V + v_base + (long long)(s0 + ws + fr) * vstr.ss;
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
       bf16x8 raw = *reinterpret_cast<const bf16x8*>(krow + ks * 32 + fq * 8);
@@ -464,7 +487,7 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
 #pragma unroll
     for (int i = 0; i < TITERS; ++i) {
       int idx = tid + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
-      q_st[i] = *reinterpret_cast<const bf16x8*>(Q + base + (long long)(qq0 + row) * D + seg * 8);
+      q_st[i] = *reinterpret_cast<const bf16x8*>(Q + q_base + (long long)(qq0 + row) * qstr.ss + seg * 8);
       do_st[i] = *reinterpret_cast<const bf16x8*>(dO + base + (long long)(qq0 + row) * D + seg * 8);
     }
 #pragma unroll
@@ -613,7 +636,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
     const ushort_t* __restrict__ K, const ushort_t* __restrict__ KT,
     const ushort_t* __restrict__ V,
     const float* __restrict__ LSE, const float* __restrict__ delta,
-    ushort_t* __restrict__ dQ, int B, int H, int S, float scale) {
+    ushort_t* __restrict__ dQ, int B, int H, int S, float scale,
+    Strides3 qstr, Strides3 kstr, Strides3 vstr) {
   constexpr int KSTEPS = D / 32;
   constexpr int DFRAGS = D / 16;
   constexpr int NBUF = (D == 64) ? 2 : 1;
@@ -627,7 +651,11 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 
   const int q_tile = blockIdx.x;
   const long long bh = blockIdx.y;
-  const long long base = (bh * S) * D;
+  const long long b_ = bh / H, h_ = bh % H;
+  const long long base = (bh * S) * D;  // contiguous dO/KT/LSE/delta/dQ
+  const long long q_base = b_ * qstr.sb + h_ * qstr.sh;
+  const long long k_base = b_ * kstr.sb + h_ * kstr.sh;
+  const long long v_base = b_ * vstr.sb + h_ * vstr.sh;
 
   const int tid = threadIdx.x, lane = tid & 63, wave = tid >> 6;
   const int q0 = q_tile * TILE;
@@ -636,7 +664,7 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 
   bf16x8 q_frag[KSTEPS], do_frag[KSTEPS];
   {
-    const ushort_t* qrow = Q + base + (long long)(q0 + wq + fr) * D;
+    const ushort_t* qrow = Q + q_base + (long long)(q0 + wq + fr) * qstr.ss;
     const ushort_t* dorow = dO + base + (long long)(q0 + wq + fr) * D;
 #pragma unroll
     for (int ks = 0; ks < KSTEPS; ++ks) {
@@ -667,8 +695,8 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 #pragma unroll
     for (int i = 0; i < TITERS; ++i) {
       int idx = tid + i * NTHREADS, row = idx / (D / 8), seg = idx % (D / 8);
-      k_st[i] = *reinterpret_cast<const bf16x8*>(K + base + (long long)(s0 + row) * D + seg * 8);
-      v_st[i] = *reinterpret_cast<const bf16x8*>(V + base + (long long)(s0 + row) * D + seg * 8);
+      k_st[i] = *reinterpret_cast<const bf16x8*>(K + k_base + (long long)(s0 + row) * kstr.ss + seg * 8);
+      v_st[i] = *reinterpret_cast<const bf16x8*>(V + v_base + (long long)(s0 + row) * vstr.ss + seg * 8);
     }
 #pragma unroll
     for (int i = 0; i < TTITERS; ++i) {
@@ -773,10 +801,12 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
 #define DISPATCH_FWD(DV, CV) \
   hipLaunchKernelGGL((flash_fwd_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
                      (hipStream_t)stream, (const ushort_t*)Q, (const ushort_t*)K, (const ushort_t*)VT, \
-                     (ushort_t*)O, (float*)LSE, B, H, Hkv, S, scale)
+                     (ushort_t*)O, (float*)LSE, B, H, Hkv, S, scale, \
+                     Strides3{qstr[0], qstr[1], qstr[2]}, Strides3{kstr[0], kstr[1], kstr[2]})
 
 extern "C" void launch_flash_fwd(const void* Q, const void* K, const void* VT, void* O, void* LSE,
                                  int B, int H, int Hkv, int S, int D, float scale, int causal,
+                                 const long long* qstr, const long long* kstr,
                                  void* stream) {
   if (D == 64) { if (causal) DISPATCH_FWD(64, true); else DISPATCH_FWD(64, false); }
   else         { if (causal) DISPATCH_FWD(128, true); else DISPATCH_FWD(128, false); }
@@ -799,12 +829,16 @@ extern "C" void launch_flash_delta(const void* dO, const void* O, void* delta,
                      (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)dOT, \
                      (const ushort_t*)Q, (const ushort_t*)QT, (const ushort_t*)K, \
                      (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
-                     (ushort_t*)dK, (ushort_t*)dV, B, H, S, scale)
+                     (ushort_t*)dK, (ushort_t*)dV, B, H, S, scale, \
+                     Strides3{qstr[0], qstr[1], qstr[2]}, Strides3{kstr[0], kstr[1], kstr[2]}, \
+                     Strides3{vstr[0], vstr[1], vstr[2]})
 
 extern "C" void launch_flash_bwd_dkdv(const void* dO, const void* dOT, const void* Q, const void* QT,
                                       const void* K, const void* V,
                                       const void* LSE, const void* delta, void* dK, void* dV,
-                                      int B, int H, int S, int D, float scale, int causal, void* stream) {
+                                      int B, int H, int S, int D, float scale, int causal,
+                                      const long long* qstr, const long long* kstr,
+                                      const long long* vstr, void* stream) {
   if (D == 64) { if (causal) DISPATCH_DKDV(64, true); else DISPATCH_DKDV(64, false); }
   else         { if (causal) DISPATCH_DKDV(128, true); else DISPATCH_DKDV(128, false); }
 }
@@ -813,12 +847,16 @@ extern "C" void launch_flash_bwd_dkdv(const void* dO, const void* dOT, const voi
   hipLaunchKernelGGL((flash_bwd_dq_kernel<DV, CV>), dim3(S / TILE, B * H), dim3(NTHREADS), 0, \
                      (hipStream_t)stream, (const ushort_t*)dO, (const ushort_t*)Q, (const ushort_t*)K, \
                      (const ushort_t*)KT, (const ushort_t*)V, (const float*)LSE, (const float*)delta, \
-                     (ushort_t*)dQ, B, H, S, scale)
+                     (ushort_t*)dQ, B, H, S, scale, \
+                     Strides3{qstr[0], qstr[1], qstr[2]}, Strides3{kstr[0], kstr[1], kstr[2]}, \
+                     Strides3{vstr[0], vstr[1], vstr[2]})
 
 extern "C" void launch_flash_bwd_dq(const void* dO, const void* Q, const void* K, const void* KT,
                                     const void* V,
                                     const void* LSE, const void* delta, void* dQ,
-                                    int B, int H, int S, int D, float scale, int causal, void* stream) {
+                                    int B, int H, int S, int D, float scale, int causal,
+                                    const long long* qstr, const long long* kstr,
+                                    const long long* vstr, void* stream) {
   if (D == 64) { if (causal) DISPATCH_DQ(64, true); else DISPATCH_DQ(64, false); }
   else         { if (causal) DISPATCH_DQ(128, true); else DISPATCH_DQ(128, false); }
 }

@@ -534,3 +534,25 @@ def test_multi_tensor_accumulate_matches_torch():
     torch.cuda.synchronize()
     for a, r in zip(accs, refs):
         assert torch.allclose(a, r, atol=1e-6)
+
+
+def test_flash_attention_strided_views_zero_copy():
+    """q/k/v as strided views of one qkv buffer (the model's GEMM output
+    layout) must produce identical results to contiguous inputs."""
+    from hivemind_amd.ops import flash_attention
+
+    torch.manual_seed(21)
+    B, H, S, D = 2, 4, 128, 64
+    qkv = torch.randn(B, S, 3, H, D, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))  # strided views
+    out = flash_attention(q, k, v)
+    out.sum().backward()
+    grad_strided = qkv.grad.clone()
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    qc, kc, vc = (qkv2[:, :, i].transpose(1, 2).contiguous() for i in range(3))
+    out2 = flash_attention(qc, kc, vc)
+    assert torch.equal(out, out2), "strided vs contiguous forward mismatch"
+    # backward through the contiguous copies, mapped back by autograd
+    out2.sum().backward()
+    assert torch.allclose(grad_strided, qkv2.grad, atol=1e-5), "strided vs contiguous backward"
