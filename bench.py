@@ -89,7 +89,10 @@ def main():
                 raise RuntimeError("disabled via PYTORCH_TUNABLEOP_ENABLED=0")
             tunable.enable(True)
             tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
-                                 "profiles", "tunableop_albert_b128.csv")
+                                 "profiles", "tunableop_albert.csv")
+            if not os.path.exists(tuned):
+                tuned = os.path.join(os.path.dirname(os.path.abspath(__file__)),
+                                     "profiles", "tunableop_albert_b128.csv")
             if os.path.exists(tuned):
                 try:
                     tunable.read_file(tuned)
