@@ -103,7 +103,8 @@ def main():
             # table just use hipBLASLt's default algorithm unless the caller
             # explicitly opts in to tuning
             tunable.tuning_enable(os.environ.get("PYTORCH_TUNABLEOP_TUNING", "0") == "1")
-            tunable.write_file_on_exit(False)
+            if hasattr(tunable, "write_file_on_exit"):  # removed in torch 2.10
+                tunable.write_file_on_exit(False)
         except Exception as e:
             log(f"TunableOp unavailable: {e}")
         device = torch.device("cuda", local_rank)
