@@ -252,9 +252,11 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
   f32x4 o_acc[DFRAGS];
 #pragma unroll
   for (int i = 0; i < DFRAGS; ++i) o_acc[i] = {0.f, 0.f, 0.f, 0.f};
-  float m_run[4], l_run[4];
-#pragma unroll
-  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+  // swapped QK^T (S^T = mfma(K, Q)) makes each lane own ONE q column
+  // (q = fr): the running max/sum are lane-local SCALARS, the row-max
+  // reduction is 2 shuffles instead of 16, and P writes become 8-byte
+  // vectors (guide T12's "make the reduction axis lane-local" idea)
+  float m_run = -1e30f, l_run = 0.f;
 
   // per-thread staging geometry (constant): 16B segment (row, seg) pairs
   const int tid_ = tid;
@@ -296,61 +298,72 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     const bool has_next = s0 + TILE < s_end;
     if (has_next) load_tile_regs(s0 + TILE);  // issue early; lands after MFMA
 
-    // S-tile: wave computes [16 q][64 s] as 4 col-block fragments
-    f32x4 sacc[4];
+    // S^T tile: wave computes [64 s][16 q] as 4 s-block fragments; lane
+    // holds 16 s-values of its own q column (q = fr)
+    f32x4 st[4];
 #pragma unroll
     for (int ns = 0; ns < 4; ++ns) {
-      sacc[ns] = {0.f, 0.f, 0.f, 0.f};
+      st[ns] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
         bf16x8 kf = lds_read8(k_l[buf], ns * 16 + fr, ks * 32 + fq * 8, D);
-        sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, sacc[ns], 0, 0, 0);
+        st[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, q_frag[ks], st[ns], 0, 0, 0);
       }
     }
     if (CAUSAL && s0 + TILE > q0) {
+      const int qg = q0 + wq + fr;
 #pragma unroll
       for (int ns = 0; ns < 4; ++ns)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          int qg = q0 + wq + fq * 4 + r, sg = s0 + ns * 16 + fr;
-          if (sg > qg) sacc[ns][r] = -1e30f;
-        }
+        for (int r = 0; r < 4; ++r)
+          if (s0 + ns * 16 + fq * 4 + r > qg) st[ns][r] = -1e30f;
     }
 
-    // online softmax: row max replicated across each 16-lane group; the row
-    // SUM stays a per-lane partial (it scales linearly with the rescale
-    // factor) and is butterfly-reduced once in the epilogue
-    float p[4][4];  // [ns][reg]
+    // online softmax for my q column: local 16-way max + 2 shuffles
+    float pm = fmaxf(fmaxf(fmaxf(st[0][0], st[0][1]), fmaxf(st[0][2], st[0][3])),
+                     fmaxf(fmaxf(st[1][0], st[1][1]), fmaxf(st[1][2], st[1][3])));
+    pm = fmaxf(pm, fmaxf(fmaxf(fmaxf(st[2][0], st[2][1]), fmaxf(st[2][2], st[2][3])),
+                         fmaxf(fmaxf(st[3][0], st[3][1]), fmaxf(st[3][2], st[3][3]))));
+    pm = fmaxf(pm, __shfl_xor(pm, 16, 64));
+    pm = fmaxf(pm, __shfl_xor(pm, 32, 64));
+    const float m_new = fmaxf(m_run, pm);
+    const float corr = __expf(m_run - m_new);
+    m_run = m_new;
+    float p[4][4];
+    float psum = 0.f;
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float pm = fmaxf(fmaxf(sacc[0][r], sacc[1][r]), fmaxf(sacc[2][r], sacc[3][r]));
-      pm = rowmax16(pm);
-      float m_new = fmaxf(m_run[r], pm);
-      float corr = __expf(m_run[r] - m_new);
-      m_run[r] = m_new;
-      float psum = 0.f;
+    for (int ns = 0; ns < 4; ++ns)
 #pragma unroll
-      for (int ns = 0; ns < 4; ++ns) {
-        float e = (sacc[ns][r] <= -1e29f) ? 0.f : __expf(sacc[ns][r] - m_new);
+      for (int r = 0; r < 4; ++r) {
+        float e = (st[ns][r] <= -1e29f) ? 0.f : __expf(st[ns][r] - m_new);
         p[ns][r] = e;
         psum += e;
       }
-      l_run[r] = l_run[r] * corr + psum;  // per-lane partial
+    l_run = l_run * corr + psum;  // per-lane partial over my 16 s-values
+
+    // rescale O: its rows are q_local = fq*4+r; lane (q_local) of the first
+    // 16 holds the final corr for that column
 #pragma unroll
-      for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr;
+    for (int r = 0; r < 4; ++r) {
+      float corr_row = __shfl(corr, fq * 4 + r, 64);
+#pragma unroll
+      for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr_row;
     }
 
-    // P -> wave-private LDS (bf16, paired conversions), then O += P @ V^T
+    // P -> wave-private LDS as [16 q][64 s]: my row q = fr, 4 consecutive
+    // s per fragment reg-group -> one 8-byte vector write per s-block
     ushort_t* pw = p_l[wave];
 #pragma unroll
-    for (int ns = 0; ns < 4; ns += 2)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        ushort_t a, b;
-        f2bf2(p[ns][r], p[ns + 1][r], a, b);
-        lds_write1(pw, fq * 4 + r, ns * 16 + fr, TILE, a);
-        lds_write1(pw, fq * 4 + r, (ns + 1) * 16 + fr, TILE, b);
-      }
+    for (int ns = 0; ns < 4; ++ns) {
+      ushort_t a, b, c, d2;
+      f2bf2(p[ns][0], p[ns][1], a, b);
+      f2bf2(p[ns][2], p[ns][3], c, d2);
+      unsigned int packed[2] = {(unsigned)a | ((unsigned)b << 16),
+                                (unsigned)c | ((unsigned)d2 << 16)};
+      *reinterpret_cast<ulonglong1*>(
+          reinterpret_cast<char*>(pw) + swz_off(fr, ns * 16 + fq * 4, TILE)) =
+          ulonglong1{((unsigned long long)packed[1] << 32) | packed[0]};
+    }
     // wave-private region: in-wave ds ordering suffices, no barrier
     bf16x8 pa[2];
 #pragma unroll
@@ -370,13 +383,15 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
     }
   }
 
-  // epilogue: reduce the per-lane partial row sums, then O /= l, save LSE
-#pragma unroll
-  for (int r = 0; r < 4; ++r) l_run[r] = rowsum16(l_run[r]);
+  // epilogue: finish my column's sum (2 shuffles), save LSE, then O /= l
+  l_run += __shfl_xor(l_run, 16, 64);
+  l_run += __shfl_xor(l_run, 32, 64);
+  if (fq == 0) LSE[bh * S + q0 + wq + fr] = m_run + __logf(fmaxf(l_run, 1e-30f));
+  const float inv_mine = (l_run > 0.f) ? 1.f / l_run : 0.f;
 #pragma unroll
   for (int r = 0; r < 4; ++r) {
     int qg = q0 + wq + fq * 4 + r;
-    float inv_l = (l_run[r] > 0.f) ? 1.f / l_run[r] : 0.f;
+    float inv_l = __shfl(inv_mine, fq * 4 + r, 64);
 #pragma unroll
     for (int nd = 0; nd < DFRAGS; nd += 2) {
       ushort_t a, b;
@@ -384,7 +399,6 @@ __global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
       O[o_base + (long long)qg * D + nd * 16 + fr] = a;
       O[o_base + (long long)qg * D + (nd + 1) * 16 + fr] = b;
     }
-    if (fr == 0) LSE[bh * S + qg] = m_run[r] + __logf(fmaxf(l_run[r], 1e-30f));
   }
 }
 
