@@ -741,35 +741,50 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dq_kernel(
     const bool has_next = s0 + TILE < s_end;
     if (NBUF == 2 && has_next) load_regs(s0 + TILE);
 
-    f32x4 sacc[4], dp[4];
+    // swapped operands: S^T/dP^T = [64 s][16 q]; lane owns q column fr, so
+    // the LSE/delta lookups hoist to scalars and the dS write packs 4
+    // consecutive s into one 8-byte vector (same trick as the forward)
+    const float lse_mine = lse_l[wq + fr];
+    const float dlt_mine = dlt_l[wq + fr];
+    const int qg = q0 + wq + fr;
+    f32x4 st[4], dpt[4];
 #pragma unroll
     for (int ns = 0; ns < 4; ++ns) {
-      sacc[ns] = {0.f, 0.f, 0.f, 0.f};
-      dp[ns] = {0.f, 0.f, 0.f, 0.f};
+      st[ns] = {0.f, 0.f, 0.f, 0.f};
+      dpt[ns] = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll
       for (int ks = 0; ks < KSTEPS; ++ks) {
         bf16x8 kf = lds_read8(k_l[buf], ns * 16 + fr, ks * 32 + fq * 8, D);
         bf16x8 vf = lds_read8(v_l[buf], ns * 16 + fr, ks * 32 + fq * 8, D);
-        sacc[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(q_frag[ks], kf, sacc[ns], 0, 0, 0);
-        dp[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(do_frag[ks], vf, dp[ns], 0, 0, 0);
+        st[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(kf, q_frag[ks], st[ns], 0, 0, 0);
+        dpt[ns] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(vf, do_frag[ks], dpt[ns], 0, 0, 0);
       }
     }
 
-    // dS = scale * P * (dP - delta[q]); write to wave-private T rows
+    // dS = scale * P * (dP - delta); write my q row of T, 8 bytes per block
 #pragma unroll
-    for (int ns = 0; ns < 4; ns += 2)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        int qrow = wq + fq * 4 + r;
-        int qg = q0 + qrow;
-        float p0 = (CAUSAL && s0 + ns * 16 + fr > qg) ? 0.f : __expf(sacc[ns][r] - lse_l[qrow]);
-        float p1 = (CAUSAL && s0 + (ns + 1) * 16 + fr > qg) ? 0.f : __expf(sacc[ns + 1][r] - lse_l[qrow]);
-        ushort_t a, b;
-        f2bf2(scale * p0 * (dp[ns][r] - dlt_l[qrow]),
-              scale * p1 * (dp[ns + 1][r] - dlt_l[qrow]), a, b);
-        lds_write1(t_l, qrow, ns * 16 + fr, TILE, a);
-        lds_write1(t_l, qrow, (ns + 1) * 16 + fr, TILE, b);
+    for (int ns = 0; ns < 4; ++ns) {
+      float v0 = 0.f, v1 = 0.f, v2 = 0.f, v3 = 0.f;
+      {
+        int sg = s0 + ns * 16 + fq * 4;
+        float p0 = (CAUSAL && sg + 0 > qg) ? 0.f : __expf(st[ns][0] - lse_mine);
+        float p1 = (CAUSAL && sg + 1 > qg) ? 0.f : __expf(st[ns][1] - lse_mine);
+        float p2 = (CAUSAL && sg + 2 > qg) ? 0.f : __expf(st[ns][2] - lse_mine);
+        float p3 = (CAUSAL && sg + 3 > qg) ? 0.f : __expf(st[ns][3] - lse_mine);
+        v0 = scale * p0 * (dpt[ns][0] - dlt_mine);
+        v1 = scale * p1 * (dpt[ns][1] - dlt_mine);
+        v2 = scale * p2 * (dpt[ns][2] - dlt_mine);
+        v3 = scale * p3 * (dpt[ns][3] - dlt_mine);
       }
+      ushort_t a, b, c, d2;
+      f2bf2(v0, v1, a, b);
+      f2bf2(v2, v3, c, d2);
+      unsigned int lo = (unsigned)a | ((unsigned)b << 16);
+      unsigned int hi = (unsigned)c | ((unsigned)d2 << 16);
+      *reinterpret_cast<ulonglong1*>(
+          reinterpret_cast<char*>(t_l) + swz_off(wq + fr, ns * 16 + fq * 4, TILE)) =
+          ulonglong1{((unsigned long long)hi << 32) | lo};
+    }
     // dQ += dS @ K   (b: K^T rows over s)
     bf16x8 da[2];
 #pragma unroll
