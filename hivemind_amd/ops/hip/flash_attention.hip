@@ -543,7 +543,10 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
     const bool has_next = qq0 + TILE < S;
     if (NBUF == 2 && has_next) load_regs(qq0 + TILE);
 
-    // S^T[s][q] (pre-scaled via k_frag) and dP^T[s][q]
+    // swapped operands: S[q][s] / dP[q][s] with lane owning s column
+    // (s = ws + fr, this wave's kv rows) -- P^T/dS^T then write as 8-byte
+    // vectors into their [s][q] rows of T (same trick as fwd/dq)
+    const int sg = s0 + ws + fr;
     f32x4 st[4], dpt[4];
 #pragma unroll
     for (int nq = 0; nq < 4; ++nq) {
@@ -553,31 +556,29 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
       for (int ks = 0; ks < KSTEPS; ++ks) {
         bf16x8 qf = lds_read8(q_l[buf], nq * 16 + fr, ks * 32 + fq * 8, D);
         bf16x8 dof = lds_read8(do_l[buf], nq * 16 + fr, ks * 32 + fq * 8, D);
-        st[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(k_frag[ks], qf, st[nq], 0, 0, 0);
-        dpt[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(v_frag[ks], dof, dpt[nq], 0, 0, 0);
+        st[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(qf, k_frag[ks], st[nq], 0, 0, 0);
+        dpt[nq] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dof, v_frag[ks], dpt[nq], 0, 0, 0);
       }
     }
 
-    // P^T = exp(S^T - LSE[q]); write to wave-private T rows (paired cvt)
+    // P^T = exp(S - LSE[q]) elementwise; packed write to my s row of T
     float pt[4][4];
 #pragma unroll
     for (int nq = 0; nq < 4; ++nq) {
-      float lse_q = lse_l[buf][nq * 16 + fr];
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        int sg = s0 + ws + fq * 4 + r, qg = qq0 + nq * 16 + fr;
-        pt[nq][r] = (CAUSAL && sg > qg) ? 0.f : __expf(st[nq][r] - lse_q);
+        int qg = qq0 + nq * 16 + fq * 4 + r;
+        pt[nq][r] = (CAUSAL && sg > qg) ? 0.f : __expf(st[nq][r] - lse_l[buf][nq * 16 + fq * 4 + r]);
       }
+      ushort_t a, b, c, d2;
+      f2bf2(pt[nq][0], pt[nq][1], a, b);
+      f2bf2(pt[nq][2], pt[nq][3], c, d2);
+      unsigned int lo = (unsigned)a | ((unsigned)b << 16);
+      unsigned int hi = (unsigned)c | ((unsigned)d2 << 16);
+      *reinterpret_cast<ulonglong1*>(
+          reinterpret_cast<char*>(t_l) + swz_off(ws + fr, nq * 16 + fq * 4, TILE)) =
+          ulonglong1{((unsigned long long)hi << 32) | lo};
     }
-#pragma unroll
-    for (int nq = 0; nq < 4; nq += 2)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        ushort_t a, b;
-        f2bf2(pt[nq][r], pt[nq + 1][r], a, b);
-        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, a);
-        lds_write1(t_l, ws + fq * 4 + r, (nq + 1) * 16 + fr, TILE, b);
-      }
     // dV += P^T @ dO   (a: own T rows over q; b: dO^T rows over q)
     bf16x8 ta[2];
 #pragma unroll
@@ -590,18 +591,21 @@ __global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
         dv_acc[nd] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ta[ks], db, dv_acc[nd], 0, 0, 0);
       }
 
-    // dS^T = scale * P^T * (dP^T - delta[q]); overwrite own T rows
+    // dS^T = scale * P^T * (dP - delta[q]); overwrite my s row of T
 #pragma unroll
-    for (int nq = 0; nq < 4; nq += 2) {
-      float d0 = dlt_l[buf][nq * 16 + fr], d1 = dlt_l[buf][(nq + 1) * 16 + fr];
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        ushort_t a, b;
-        f2bf2(scale * pt[nq][r] * (dpt[nq][r] - d0),
-              scale * pt[nq + 1][r] * (dpt[nq + 1][r] - d1), a, b);
-        lds_write1(t_l, ws + fq * 4 + r, nq * 16 + fr, TILE, a);
-        lds_write1(t_l, ws + fq * 4 + r, (nq + 1) * 16 + fr, TILE, b);
-      }
+    for (int nq = 0; nq < 4; ++nq) {
+      float v0 = scale * pt[nq][0] * (dpt[nq][0] - dlt_l[buf][nq * 16 + fq * 4 + 0]);
+      float v1 = scale * pt[nq][1] * (dpt[nq][1] - dlt_l[buf][nq * 16 + fq * 4 + 1]);
+      float v2 = scale * pt[nq][2] * (dpt[nq][2] - dlt_l[buf][nq * 16 + fq * 4 + 2]);
+      float v3 = scale * pt[nq][3] * (dpt[nq][3] - dlt_l[buf][nq * 16 + fq * 4 + 3]);
+      ushort_t a, b, c, d2;
+      f2bf2(v0, v1, a, b);
+      f2bf2(v2, v3, c, d2);
+      unsigned int lo = (unsigned)a | ((unsigned)b << 16);
+      unsigned int hi = (unsigned)c | ((unsigned)d2 << 16);
+      *reinterpret_cast<ulonglong1*>(
+          reinterpret_cast<char*>(t_l) + swz_off(ws + fr, nq * 16 + fq * 4, TILE)) =
+          ulonglong1{((unsigned long long)hi << 32) | lo};
     }
     // dK += dS^T @ Q   (b: Q^T rows over q)
 #pragma unroll
