@@ -433,23 +433,33 @@ class DecentralizedAverager(ServicerBase):
         lock_ctx = enter_asynchronously(self.lock_averaged_tensors) if take_lock else _null_actx()
         rccl_ranks = self._rccl_group_ranks(meta)
         if rccl_ranks is not None:
-            self.last_data_plane = "rccl"
-            # full node and Moshpit subgroups alike: cached member-only communicator
-            process_group = get_process_group_for_ranks(rccl_ranks)
-            async with lock_ctx:
-                await asyncio.get_event_loop().run_in_executor(
-                    None,
-                    DistributedAllReduceRunner(
-                        tensors,
-                        weight,
-                        process_group=process_group,
-                        wire_dtype=self.allreduce_wire_dtype,
-                        codec=self.allreduce_codec,
-                        averaging_alpha=self.averaging_alpha,
-                        ticket=meta.rccl_ticket,
-                    ).run,
-                )
-            return
+            try:
+                self.last_data_plane = "rccl"
+                # full node and Moshpit subgroups alike: cached member-only communicator
+                process_group = get_process_group_for_ranks(rccl_ranks)
+                async with lock_ctx:
+                    await asyncio.get_event_loop().run_in_executor(
+                        None,
+                        DistributedAllReduceRunner(
+                            tensors,
+                            weight,
+                            process_group=process_group,
+                            wire_dtype=self.allreduce_wire_dtype,
+                            codec=self.allreduce_codec,
+                            averaging_alpha=self.averaging_alpha,
+                            ticket=meta.rccl_ticket,
+                        ).run,
+                    )
+                return
+            except RuntimeError as e:
+                # communicator creation / collective failure: the matched group
+                # is still valid, so fall through to the RPC butterfly rather
+                # than failing the whole round. NOTE: this branch is a
+                # same-decision-on-every-rank situation only for communicator
+                # SETUP failures; a mid-collective failure already poisons the
+                # communicator on every member, so they all land here.
+                logger.warning(f"RCCL data plane failed ({e!r}); falling back to the RPC butterfly")
+                self.last_data_plane = "rpc"
         self.last_data_plane = "rpc"
         download_bandwidths = [
             (0.0 if mode == AveragingMode.CLIENT else bw) for mode, bw in zip(meta.modes, meta.bandwidths)
