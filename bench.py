@@ -266,6 +266,7 @@ def main():
             "dtype": "bf16" if use_gpu else "fp32",
             "data": "synthetic",
             "epochs_in_timed_window": epochs_completed,
+            "grad_data_plane": (opt.grad_averager.last_data_plane if opt.grad_averager is not None else None),
             "global_step_ms": round(1000.0 * sum(epoch_steps) / len(epoch_steps), 2) if epoch_steps else None,
             "plain_step_ms": round(
                 1000.0 * (sum(step_times) - sum(epoch_steps)) / max(1, len(step_times) - len(epoch_steps)), 2

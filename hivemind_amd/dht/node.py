@@ -320,8 +320,15 @@ class DHTNode:
 
         unfinished = set(binary_values.keys())
         store_ok: Dict[Any, bool] = {id_to_original[c]: False for c in unfinished}
+        # store lookups only need num_replicas owners; a full bucket-size beam
+        # (20) visited ~2x more peers per key for placement accuracy the
+        # replication factor doesn't use. Reads keep the full beam, so a
+        # record stored on the 5-nearest-of-a-12-beam is still found.
         nearest_per_query = await self.find_nearest_nodes(
-            list(dict.fromkeys(key_ids)), k_nearest=self.num_replicas, exclude_self=exclude_self
+            list(dict.fromkeys(key_ids)),
+            k_nearest=self.num_replicas,
+            beam_size=max(2 * self.num_replicas + 2, self.num_replicas),
+            exclude_self=exclude_self,
         )
 
         # group work by DESTINATION PEER: one bulk call_store per peer carrying
