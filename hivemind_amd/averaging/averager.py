@@ -529,7 +529,10 @@ class DecentralizedAverager(ServicerBase):
             index = 0
             async for delta in aiter_with_timeout(iter_results, self.reducer_timeout):
                 if runner.modes[runner.ordered_peer_ids.index(self.peer_id)] != AveragingMode.AUX:
-                    tensors[index].detach().add_(delta.to(tensors[index].dtype), alpha=self.averaging_alpha)
+                    from ..ops import apply_delta_
+
+                    # fused in-place delta apply (HIP kernel on GPU)
+                    apply_delta_(tensors[index].detach(), delta.to(tensors[index].device), self.averaging_alpha)
                 index += 1
             self._state_updated.set()
         finally:

@@ -256,7 +256,11 @@ class TensorPartReducer:
 
         current_part_future = self.current_part_future
         effective_weight = weight * self.weights[sender_index]
-        self.accumulator.add_(tensor_part.to(self.accumulator.dtype), alpha=effective_weight)
+        # the fused weighted-accumulate HIP kernel on GPU parts
+        # (bf16/f32 into the fp32 accumulator in one pass); torch add_ on CPU
+        from ..ops import weighted_accumulate_
+
+        weighted_accumulate_(self.accumulator, tensor_part, effective_weight)
         self.denominator += effective_weight
         self.current_part_accumulated_from += 1
 
