@@ -47,6 +47,9 @@ def main():
                         choices=["none", "bf16", "int8"],
                         help="gradient wire format on the RCCL/xGMI plane: bf16 cast (default) or "
                              "blockwise-int8 quantized butterfly (baseline config 2)")
+    parser.add_argument("--hipgraph", type=str, default="auto", choices=["auto", "on", "off"],
+                        help="capture the fwd+bwd region in a hipGraph and replay it per step "
+                             "(auto: try, fall back to eager on capture failure)")
     args = parser.parse_args()
 
     world_size = int(os.environ.get("WORLD_SIZE", "1"))
@@ -196,13 +199,51 @@ def main():
             labels[mask] = -100
         return input_ids, labels
 
+    graph_state = {}
+
+    def try_capture_hipgraph():
+        """Capture fwd+bwd once; per-step we copy fresh data into the static
+        buffers and replay (the north star's "capture launch-bound inner
+        loops in hipGraphs"). The optimizer/averaging logic stays eager."""
+        static_ids, static_labels = make_batch()
+        # grads must be stable buffers for replayed accumulation
+        loss, _ = model(static_ids, labels=static_labels)
+        loss.backward()
+        opt.zero_grad(set_to_none=False)
+        torch.cuda.synchronize(device)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static_loss, _ = model(static_ids, labels=static_labels)
+            static_loss.backward()
+        graph_state.update(graph=graph, ids=static_ids, labels=static_labels, loss=static_loss)
+        log("hipGraph capture of fwd+bwd succeeded")
+
     def one_step():
+        if graph_state:
+            input_ids, labels = make_batch()
+            graph_state["ids"].copy_(input_ids)
+            graph_state["labels"].copy_(labels)
+            graph_state["graph"].replay()
+            loss = graph_state["loss"]
+            opt.step()
+            opt.zero_grad(set_to_none=False)  # keep the graph's grad buffers
+            return loss
         input_ids, labels = make_batch()
         loss, _ = model(input_ids, labels=labels)
         loss.backward()
         opt.step()
         opt.zero_grad()
         return loss
+
+    if use_gpu and args.hipgraph in ("auto", "on"):
+        try:
+            try_capture_hipgraph()
+        except Exception as e:
+            if args.hipgraph == "on":
+                raise
+            log(f"hipGraph capture failed ({e!r}); running eager")
+            graph_state.clear()
+            opt.zero_grad()
 
     def sync():
         if use_gpu:
