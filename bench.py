@@ -206,9 +206,17 @@ def main():
         buffers and replay (the north star's "capture launch-bound inner
         loops in hipGraphs"). The optimizer/averaging logic stays eager."""
         static_ids, static_labels = make_batch()
-        # grads must be stable buffers for replayed accumulation
-        loss, _ = model(static_ids, labels=static_labels)
-        loss.backward()
+        # canonical capture recipe: warm up on a SIDE stream so AccumulateGrad
+        # nodes bind to a non-default stream (a default-stream warmup made the
+        # later capture core-dump on ROCm), then drop the autograd graph refs
+        side = torch.cuda.Stream(device)
+        side.wait_stream(torch.cuda.current_stream(device))
+        with torch.cuda.stream(side):
+            for _ in range(2):
+                loss, _ = model(static_ids, labels=static_labels)
+                loss.backward()
+        torch.cuda.current_stream(device).wait_stream(side)
+        del loss
         opt.zero_grad(set_to_none=False)
         torch.cuda.synchronize(device)
         graph = torch.cuda.CUDAGraph()
@@ -250,6 +258,15 @@ def main():
             torch.cuda.synchronize(device)
         if world_size > 1:
             dist.barrier(group=bench_pg)
+
+    if world_size > 1:
+        # measure the steady swarm, not the discovery transient: wait until the
+        # progress tracker on every rank sees the whole world before warmup
+        deadline = time.perf_counter() + 30.0
+        while opt.tracker.global_progress.num_peers < world_size and time.perf_counter() < deadline:
+            time.sleep(0.1)
+        log(f"tracker sees {opt.tracker.global_progress.num_peers}/{world_size} peers")
+        sync()
 
     log(f"warmup: {args.warmup} steps")
     for i in range(args.warmup):
