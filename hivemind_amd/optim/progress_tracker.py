@@ -206,6 +206,10 @@ class ProgressTracker(threading.Thread):
     async def _progress_reporter(self):
         """Publish local progress whenever it changes (reference progress_tracker.py:195-233)."""
         last_report_time = -float("inf")
+        # announce ourselves immediately: peers must count us BEFORE our first
+        # step (waiting for the first report made fresh swarms invisible to
+        # each other until training had already begun)
+        self.should_report_progress.set()
         while not self.shutdown_triggered.is_set():
             wait_timeout = max(0.0, last_report_time + self.metadata_expiration / 2 - get_dht_time())
             await asyncio.get_event_loop().run_in_executor(
