@@ -199,7 +199,9 @@ struct Strides3 {
 };
 
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS) void flash_fwd_kernel(
+// min 4 waves/SIMD: the fwd kernel is latency-bound (WAIT:BUSY ~7 in PMC)
+// and sat at 140 VGPR = 3 waves; capping the allocator at 128 buys a 4th
+__global__ __launch_bounds__(NTHREADS, D == 64 ? 4 : 2) void flash_fwd_kernel(
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ K,
     const ushort_t* __restrict__ VT,  // [B,Hkv,D,S] pre-transposed
     ushort_t* __restrict__ O,
