@@ -133,3 +133,27 @@ def test_two_process_dispatch_matches_local_reference():
         assert torch.allclose(out, ref_outs[rank], atol=1e-5), f"rank {rank} output"
         assert torch.allclose(xg, ref_xg[rank], atol=1e-5), f"rank {rank} input grad"
         assert torch.allclose(eg, ref.experts[rank].up.weight.grad, atol=1e-5), f"rank {rank} expert grad"
+
+
+def test_rccl_moe_trains_to_high_accuracy():
+    """Reference test_training.py-style convergence: a tiny classifier built
+    on RcclMixtureOfExperts must fit a separable problem (world=1)."""
+    from hivemind_amd.moe.rccl_dispatch import RcclMixtureOfExperts
+
+    torch.manual_seed(42)
+    n, dim, classes = 256, 16, 4
+    x = torch.randn(n, dim)
+    w_true = torch.randn(dim, classes)
+    y = (x @ w_true).argmax(-1)
+
+    moe = RcclMixtureOfExperts(dim, num_local_experts=4, k=2)
+    head = torch.nn.Linear(dim, classes)
+    opt = torch.optim.Adam(list(moe.parameters()) + list(head.parameters()), lr=5e-3)
+    for step in range(500):
+        logits = head(moe(x))
+        loss = torch.nn.functional.cross_entropy(logits, y)
+        opt.zero_grad()
+        loss.backward()
+        opt.step()
+    accuracy = (logits.argmax(-1) == y).float().mean().item()
+    assert accuracy >= 0.9, f"accuracy {accuracy}"
