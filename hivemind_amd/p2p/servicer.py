@@ -47,11 +47,23 @@ def _to_wire(obj: Any) -> Any:
     return obj
 
 
+_HINTS_CACHE: dict = {}
+
+
+def _cached_hints(cls: Type) -> dict:
+    # get_type_hints evaluates string annotations with compile() -- doing
+    # that per MESSAGE was a measurable share of DHT RPC cost
+    hints = _HINTS_CACHE.get(cls)
+    if hints is None:
+        hints = _HINTS_CACHE[cls] = get_type_hints(cls)
+    return hints
+
+
 def _from_wire(cls: Type, data: Any) -> Any:
     if data is None:
         return None
     if is_dataclass(cls):
-        hints = get_type_hints(cls)
+        hints = _cached_hints(cls)
         kwargs = {}
         for f in fields(cls):
             if f.name in data:
