@@ -150,9 +150,13 @@ class _Connection:
         self.last_used = _time.monotonic()
         header = MSGPackSerializer.dumps((ftype, call_id, handler))
         async with self.send_lock:
-            self.writer.write(struct.pack(">II", len(header), len(payload)))
-            self.writer.write(header)
-            if payload:
+            if len(payload) < 65536:
+                # one write -> one send syscall for control-plane frames
+                self.writer.write(struct.pack(">II", len(header), len(payload)) + header + payload)
+            else:
+                # keep multi-MB tensor payloads zero-copy
+                self.writer.write(struct.pack(">II", len(header), len(payload)))
+                self.writer.write(header)
                 self.writer.write(payload)
             await self.writer.drain()
 
