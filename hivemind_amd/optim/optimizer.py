@@ -349,6 +349,7 @@ class Optimizer(torch.optim.Optimizer):
                 # never wait on a control that was not (and will not be) triggered
                 self.scheduled_grads.cancel()
                 self.scheduled_grads = None
+        _t_grads = time.perf_counter()
 
         should_perform_optimizer_step = not self.auxiliary and not self.use_local_updates
         should_average_state = (
@@ -378,6 +379,7 @@ class Optimizer(torch.optim.Optimizer):
         if self.use_gradient_averaging:
             if not self.delay_grad_averaging:
                 self._average_gradients_and_load_into_optimizer(self.scheduled_grads)
+        _t_avg_load = time.perf_counter()
 
         self.state_averager.step(
             increment_epoch=True,
@@ -399,6 +401,7 @@ class Optimizer(torch.optim.Optimizer):
             zero_grad=False,
         )
         self.scheduled_state = None
+        _t_state = time.perf_counter()
 
         self.tracker.update_epoch(new_epoch=self.state_averager.local_epoch)
         self._should_check_synchronization_on_update = True
@@ -407,6 +410,12 @@ class Optimizer(torch.optim.Optimizer):
             self.state_averager.allow_state_sharing = True
         logger.log(self.status_loglevel, f"transitioning to epoch {self.local_epoch} "
                    f"({time.perf_counter() - _epoch_start:.2f}s)")
+        if logger.isEnabledFor(logging.DEBUG):
+            logger.debug(
+                f"epoch timings: grads={_t_grads - _epoch_start:.3f}s "
+                f"avg_load={_t_avg_load - _t_grads:.3f}s state_step={_t_state - _t_avg_load:.3f}s "
+                f"tracker={time.perf_counter() - _t_state:.3f}s"
+            )
 
     def _begin_averaging_gradients(self) -> bool:
         """Start (or join) this epoch's gradient averaging round
