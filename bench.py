@@ -260,6 +260,16 @@ def main():
             dist.barrier(group=bench_pg)
 
     if world_size > 1:
+        # pre-warm the full-world communicator the averaging rounds will use
+        # (cached member-only group): a communicator problem then fails at
+        # startup instead of hanging the first global step
+        from hivemind_amd.averaging.rccl import get_process_group_for_ranks
+
+        warm_pg = get_process_group_for_ranks(list(range(world_size)))
+        warm = torch.ones(1, device=device if use_gpu else "cpu")
+        dist.all_reduce(warm, group=warm_pg)
+        log(f"averaging communicator warm: {warm.item():.0f} == {world_size}")
+
         # measure the steady swarm, not the discovery transient: wait until the
         # progress tracker on every rank sees the whole world before warmup
         deadline = time.perf_counter() + 30.0
