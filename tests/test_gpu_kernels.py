@@ -556,3 +556,42 @@ def test_flash_attention_strided_views_zero_copy():
     # backward through the contiguous copies, mapped back by autograd
     out2.sum().backward()
     assert torch.allclose(grad_strided, qkv2.grad, atol=1e-5), "strided vs contiguous backward"
+
+
+def test_hipgraph_capture_fwd_bwd():
+    """bench.py's hipGraph path: capture fwd+bwd once (side-stream warmup),
+    replay with fresh data, and check grads accumulate across replays."""
+    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM
+
+    torch.manual_seed(5)
+    config = AlbertConfig.base()
+    model = AlbertForMaskedLM(config).cuda()
+    ids = torch.randint(0, config.vocab_size, (2, 128), device="cuda")
+    labels = ids.clone()
+
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        for _ in range(2):
+            loss, _ = model(ids, labels=labels)
+            loss.backward()
+    torch.cuda.current_stream().wait_stream(side)
+    del loss
+    for p in model.parameters():
+        if p.grad is not None:
+            p.grad.zero_()
+    torch.cuda.synchronize()
+
+    graph = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(graph):
+        static_loss, _ = model(ids, labels=labels)
+        static_loss.backward()
+
+    graph.replay()
+    torch.cuda.synchronize()
+    some_param = model.albert.layer.qkv.weight
+    g1 = some_param.grad.clone()
+    assert torch.isfinite(static_loss).item() and g1.abs().sum().item() > 0
+    graph.replay()
+    torch.cuda.synchronize()
+    assert torch.allclose(some_param.grad, 2 * g1, rtol=1e-2, atol=1e-4), "grads must accumulate across replays"
