@@ -93,3 +93,24 @@ def test_bench_torchrun_two_ranks_int8_compression():
     assert result["epochs_in_timed_window"] >= 1
     assert result["grad_data_plane"] == "rccl", result
     assert result["config"]["grad_compression"] == "int8"
+
+
+def test_bench_torchrun_two_ranks_powersgd():
+    """BASELINE config 3 shape: rank-r PowerSGD gradient averaging through
+    the full Optimizer on 2 gloo ranks (two chained rounds per global step)."""
+    from hivemind_amd.utils.networking import get_free_port
+
+    port = str(get_free_port())
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1", "--nproc-per-node", "2",
+         "--master-addr", "127.0.0.1", "--master-port", port, "bench.py",
+         "--gpus", "2", "--steps", "6", "--warmup", "2", "--model", "tiny",
+         "--batch", "4", "--seq-len", "32", "--target-batch-size", "16",
+         "--powersgd-rank", "2"],
+        cwd=REPO, capture_output=True, text=True, timeout=900,
+    )
+    assert proc.returncode == 0, proc.stderr[-3000:]
+    result = _last_json_line(proc.stdout)
+    assert result["epochs_in_timed_window"] >= 1
+    assert result["config"]["powersgd_rank"] == 2
+    assert result["grad_data_plane"] == "rccl", result
