@@ -328,6 +328,9 @@ __global__ __launch_bounds__(NTHREADS, D == 64 ? 4 : 2) void flash_fwd_kernel(
                          fmaxf(fmaxf(st[3][0], st[3][1]), fmaxf(st[3][2], st[3][3]))));
     pm = fmaxf(pm, __shfl_xor(pm, 16, 64));
     pm = fmaxf(pm, __shfl_xor(pm, 32, 64));
+    // defer-max (guide T13): when no lane's max grew, corr == 1 exactly --
+    // skip the rescale shuffles and multiplies (most tiles after the first)
+    const bool need_rescale = !__all(pm <= m_run);
     const float m_new = fmaxf(m_run, pm);
     const float corr = __expf(m_run - m_new);
     m_run = m_new;
@@ -343,13 +346,15 @@ __global__ __launch_bounds__(NTHREADS, D == 64 ? 4 : 2) void flash_fwd_kernel(
       }
     l_run = l_run * corr + psum;  // per-lane partial over my 16 s-values
 
-    // rescale O: its rows are q_local = fq*4+r; lane (q_local) of the first
-    // 16 holds the final corr for that column
+    if (need_rescale) {
+      // rescale O: its rows are q_local = fq*4+r; lane (q_local) of the
+      // first 16 holds the final corr for that column
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      float corr_row = __shfl(corr, fq * 4 + r, 64);
+      for (int r = 0; r < 4; ++r) {
+        float corr_row = __shfl(corr, fq * 4 + r, 64);
 #pragma unroll
-      for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr_row;
+        for (int nd = 0; nd < DFRAGS; ++nd) o_acc[nd][r] *= corr_row;
+      }
     }
 
     // P -> wave-private LDS as [16 q][64 s]: my row q = fr, 4 consecutive
