@@ -373,7 +373,12 @@ def flash_attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
 @torch.no_grad()
 def apply_delta_(tensor: torch.Tensor, delta: torch.Tensor, alpha: float = 1.0):
     """tensor += alpha * delta."""
-    if tensor.is_cuda and tensor.dtype == delta.dtype and tensor.dtype in (torch.bfloat16, torch.float32):
+    if (
+        tensor.is_cuda
+        and delta.device == tensor.device  # never hand the kernel a host pointer
+        and tensor.dtype == delta.dtype
+        and tensor.dtype in (torch.bfloat16, torch.float32)
+    ):
         hip_ops().apply_delta_(tensor, delta.contiguous(), float(alpha))
     else:
         tensor.add_(delta.to(tensor.device, tensor.dtype), alpha=alpha)
@@ -383,7 +388,12 @@ def apply_delta_(tensor: torch.Tensor, delta: torch.Tensor, alpha: float = 1.0):
 @torch.no_grad()
 def weighted_accumulate_(acc: torch.Tensor, x: torch.Tensor, w: float = 1.0):
     """acc += w * x."""
-    if acc.is_cuda and acc.dtype == torch.float32 and x.dtype in (torch.float32, torch.bfloat16):
+    if (
+        acc.is_cuda
+        and x.device == acc.device  # never hand the kernel a host pointer
+        and acc.dtype == torch.float32
+        and x.dtype in (torch.float32, torch.bfloat16)
+    ):
         hip_ops().weighted_accumulate_(acc, x.contiguous(), float(w))
     else:
         acc.add_(x.to(acc.device, acc.dtype), alpha=w)

@@ -50,7 +50,7 @@ inline int grid_1d(long long n, int block = 256, int cap = 2048) {
 // ---------------------------------------------------------------- averaging
 
 void apply_delta_(torch::Tensor tensor, torch::Tensor delta, double alpha) {
-  CHECK_GPU(tensor); CHECK_CONTIG(tensor); CHECK_CONTIG(delta);
+  CHECK_GPU(tensor); CHECK_GPU(delta); CHECK_CONTIG(tensor); CHECK_CONTIG(delta);
   TORCH_CHECK(tensor.numel() == delta.numel(), "size mismatch");
   long long n = tensor.numel();
   if (tensor.scalar_type() == torch::kBFloat16 && delta.scalar_type() == torch::kBFloat16) {
@@ -66,7 +66,7 @@ void apply_delta_(torch::Tensor tensor, torch::Tensor delta, double alpha) {
 }
 
 void weighted_accumulate_(torch::Tensor acc, torch::Tensor x, double w) {
-  CHECK_GPU(acc); CHECK_CONTIG(acc); CHECK_CONTIG(x);
+  CHECK_GPU(acc); CHECK_GPU(x); CHECK_CONTIG(acc); CHECK_CONTIG(x);
   TORCH_CHECK(acc.scalar_type() == torch::kFloat32, "accumulator must be fp32");
   long long n = acc.numel();
   if (x.scalar_type() == torch::kFloat32) {
