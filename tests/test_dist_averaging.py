@@ -289,3 +289,74 @@ def test_collective_sequencer_abandoned_ticket_does_not_hang():
     assert seq.wait_turn(live) is False
     seq.release(stale)
     seq.release(live)
+
+
+def _weighted_q_peer_main(rank: int, world_size: int, port: int, result_queue):
+    import torch.distributed as dist
+
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world_size)
+    try:
+        from hivemind_amd.averaging import DecentralizedAverager
+        from hivemind_amd.dht import DHT
+
+        if rank == 0:
+            dht = DHT(start=True)
+            endpoint = [dht.endpoint]
+        else:
+            dht, endpoint = None, [None]
+        dist.broadcast_object_list(endpoint, src=0)
+        if rank != 0:
+            dht = DHT(initial_peers=[endpoint[0]], start=True)
+
+        base = torch.linspace(-1.0, 1.0, 4096)
+        weight = [1.0, 3.0][rank]
+        averager = DecentralizedAverager(
+            [base * (rank + 1)],
+            dht,
+            start=True,
+            prefix="wq",
+            target_group_size=world_size,
+            min_group_size=world_size,
+            min_matchmaking_time=1.0,
+            request_timeout=0.5,
+            allreduce_codec="blockwise_int8",
+        )
+        result = averager.step(weight=weight, timeout=60)
+        assert result is not None
+        # weighted mean: (1*x1 + 3*x2) / 4 with x_r = base*(r+1)
+        expected = base * (1 * 1 + 3 * 2) / 4
+        with averager.get_tensors() as ts:
+            err = (ts[0] - expected).abs().max().item()
+        result_queue.put((rank, err, averager.last_data_plane))
+        averager.shutdown()
+        dht.shutdown()
+    finally:
+        dist.barrier()
+        dist.destroy_process_group()
+
+
+def test_dist_quantized_weighted_average():
+    """Weighted averaging on the int8 plane: w=[1,3] must produce the
+    weighted mean within codec tolerance."""
+    from hivemind_amd.utils.networking import get_free_port
+
+    world_size = 2
+    port = get_free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_weighted_q_peer_main, args=(r, world_size, port, q)) for r in range(world_size)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+    for p in procs:
+        assert p.exitcode == 0, p.exitcode
+    results = []
+    while not q.empty():
+        results.append(q.get())
+    assert len(results) == world_size
+    for rank, err, plane in results:
+        assert plane == "rccl", plane
+        assert err < 0.1, f"rank {rank} weighted avg err {err}"
