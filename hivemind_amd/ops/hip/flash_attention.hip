@@ -442,7 +442,10 @@ __global__ __launch_bounds__(NTHREADS) void flash_delta_kernel(
 // ======================================================================
 
 template <int D, bool CAUSAL>
-__global__ __launch_bounds__(NTHREADS) void flash_bwd_dkdv_kernel(
+// min 2 waves/SIMD: at D=128 the register allocator drifted to 253 VGPR
+// (+64 AGPR accumulators) = ONE wave/SIMD, and the kernel ran 2.8x slower
+// than bwd_dq per unit work on the Llama-8B profile
+__global__ __launch_bounds__(NTHREADS, 2) void flash_bwd_dkdv_kernel(
     const ushort_t* __restrict__ dO, const ushort_t* __restrict__ dOT,
     const ushort_t* __restrict__ Q, const ushort_t* __restrict__ QT,
     const ushort_t* __restrict__ K, const ushort_t* __restrict__ V,
