@@ -16,6 +16,7 @@ import torch
 
 def main():
     ap = argparse.ArgumentParser()
+    ap.add_argument("--model", type=str, default="albert-base", choices=["albert-base", "llama-8b", "llama-1b"])
     ap.add_argument("--batch", type=int, default=410)
     ap.add_argument("--seq-len", type=int, default=512)
     ap.add_argument("--steps", type=int, default=8)
@@ -34,18 +35,24 @@ def main():
             print(f"base table not loaded: {e}", file=sys.stderr)
     tunable.tuning_enable(True)
 
-    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM
+    from hivemind_amd.models import AlbertConfig, AlbertForMaskedLM, LlamaConfig, LlamaForCausalLM
     from hivemind_amd.ops import FusedAdamW
 
     torch.manual_seed(0)
-    config = AlbertConfig.base()
-    model = AlbertForMaskedLM(config).to("cuda")
+    if args.model == "albert-base":
+        config, cls = AlbertConfig.base(), AlbertForMaskedLM
+    elif args.model == "llama-8b":
+        config, cls = LlamaConfig.llama_3_8b(), LlamaForCausalLM
+    else:
+        config, cls = LlamaConfig.llama_1b(), LlamaForCausalLM
+    model = cls(config).to("cuda")
     opt = FusedAdamW([{"params": [p for p in model.parameters() if p.dtype == torch.float32
                                   or p.dtype == torch.bfloat16]}], lr=1e-4)
     for i in range(args.steps):
         ids = torch.randint(0, config.vocab_size, (args.batch, args.seq_len), device="cuda")
         labels = ids.clone()
-        labels[torch.rand(labels.shape, device="cuda") > 0.15] = -100
+        if args.model == "albert-base":
+            labels[torch.rand(labels.shape, device="cuda") > 0.15] = -100
         loss, _ = model(ids, labels=labels)
         loss.backward()
         opt.step()
