@@ -318,10 +318,12 @@ class DHTProtocol(ServicerBase):
                 self._update_routing(response.peer, peer_id)
             assert len(response.results) == len(keys)
             output = {}
+            _from_bytes = DHTID.from_bytes
             for key_id, result in zip(keys, response.results):
-                nearest = {}
-                for nid, pid, ep in zip(result.nearest_node_ids, result.nearest_peer_ids, result.nearest_endpoints):
-                    nearest[DHTID.from_bytes(nid)] = (PeerID(pid), ep)
+                nearest = {
+                    _from_bytes(nid): (PeerID(pid), ep)
+                    for nid, pid, ep in zip(result.nearest_node_ids, result.nearest_peer_ids, result.nearest_endpoints)
+                }
                 if result.type == NOT_FOUND:
                     output[key_id] = (None, nearest)
                 elif result.type == FOUND_REGULAR:
@@ -357,6 +359,7 @@ class DHTProtocol(ServicerBase):
         if request.peer is not None:
             self._update_routing(request.peer, context.remote_id)
         response = FindResponse(results=[], peer=self._my_info())
+        exclude = DHTID.from_bytes(request.peer.node_id) if request.peer and request.peer.node_id else None
         for key_bytes in request.keys:
             key_id = DHTID.from_bytes(key_bytes)
             result = FindResult()
@@ -372,12 +375,10 @@ class DHTProtocol(ServicerBase):
                     result.type = FOUND_REGULAR
                     result.value = maybe_item.value
                 result.expiration_time = maybe_item.expiration_time
-            for node_id, (peer_id, endpoint) in self.routing_table.get_nearest_neighbors(
-                key_id, k=self.bucket_size, exclude=DHTID.from_bytes(request.peer.node_id) if request.peer and request.peer.node_id else None
-            ):
-                result.nearest_node_ids.append(node_id.to_bytes())
-                result.nearest_peer_ids.append(peer_id.to_bytes())
-                result.nearest_endpoints.append(endpoint)
+            nn = self.routing_table.get_nearest_neighbors(key_id, k=self.bucket_size, exclude=exclude)
+            result.nearest_node_ids = [node_id.to_bytes() for node_id, _ in nn]
+            result.nearest_peer_ids = [peer[0].to_bytes() for _, peer in nn]
+            result.nearest_endpoints = [peer[1] for _, peer in nn]
             response.results.append(result)
         return response
 

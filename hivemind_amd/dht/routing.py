@@ -54,11 +54,20 @@ class DHTID(int):
         return min(map(len, ids_bits))
 
     def to_bytes(self, length: int = 32, byteorder: str = "big", *, signed: bool = False) -> bytes:
-        return super().to_bytes(length, byteorder, signed=signed)
+        # routing-table DHTIDs get re-serialized into every FindResponse that
+        # lists them as neighbors: cache the canonical 32-byte form
+        if length == 32 and byteorder == "big" and not signed:
+            b = getattr(self, "_b32", None)
+            if b is None:
+                b = self._b32 = int.to_bytes(self, 32, "big")
+            return b
+        return int.to_bytes(self, length, byteorder, signed=signed)
 
     @classmethod
     def from_bytes(cls, raw: bytes, byteorder: str = "big", *, signed: bool = False) -> "DHTID":
-        return DHTID(super().from_bytes(raw, byteorder, signed=signed))
+        # hot path (one call per peer entry in every FindResponse): a 32-byte
+        # input is in range by construction, so bypass __new__'s range assert
+        return int.__new__(cls, int.from_bytes(raw, byteorder, signed=signed))
 
     def __repr__(self):
         return f"{self.__class__.__name__}({hex(self)[:10]}…)"
@@ -141,12 +150,17 @@ class RoutingTable:
         self, query_id: DHTID, k: int, exclude: Optional[DHTID] = None
     ) -> List[Tuple[DHTID, Tuple[PeerID, str]]]:
         """k nodes nearest to query_id by XOR metric, excluding `exclude`."""
+        # hot path: called for every incoming rpc_find; DHTID is an int, so
+        # XOR directly and tuple-sort (distances are unique -- XOR metric with
+        # distinct uids -- so the comparison never falls through to field 2)
+        q = int(query_id)
         candidates: List[Tuple[int, DHTID, Tuple[PeerID, str]]] = []
+        append = candidates.append
         for bucket in self.buckets:
             for uid, peer in bucket.nodes_to_peers.items():
                 if uid != exclude:
-                    candidates.append((query_id.xor_distance(uid), uid, peer))
-        candidates.sort(key=lambda t: t[0])
+                    append((q ^ uid, uid, peer))
+        candidates.sort()
         return [(uid, peer) for _, uid, peer in candidates[:k]]
 
     def __repr__(self):
