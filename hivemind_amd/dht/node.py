@@ -255,7 +255,8 @@ class DHTNode:
             for nid, (peer_id, endpoint) in self.protocol.routing_table.get_nearest_neighbors(
                 q, beam_size, exclude=self.node_id
             ):
-                node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                if nid not in node_to_peer:
+                    node_to_peer[nid] = PeerInfo(peer_id, (endpoint,))
                 if nid not in initial:
                     initial.append(nid)
 
@@ -266,7 +267,8 @@ class DHTNode:
             output = {}
             for q, (_, nearest) in response.items():
                 for nid, (peer_id, endpoint) in nearest.items():
-                    node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                    if nid not in node_to_peer:  # setdefault would construct PeerInfo per entry
+                        node_to_peer[nid] = PeerInfo(peer_id, (endpoint,))
                 output[q] = (list(nearest.keys()), False)
             return output
 
@@ -463,7 +465,8 @@ class DHTNode:
             for nid, (peer_id, endpoint) in self.protocol.routing_table.get_nearest_neighbors(
                 q, beam_size, exclude=self.node_id
             ):
-                node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                if nid not in node_to_peer:
+                    node_to_peer[nid] = PeerInfo(peer_id, (endpoint,))
                 if nid not in initial:
                     initial.append(nid)
 
@@ -478,7 +481,8 @@ class DHTNode:
             for q, (maybe_value, nearest) in response.items():
                 search_results[q].add_candidate(maybe_value, source_node_id=node)
                 for nid, (peer_id, endpoint) in nearest.items():
-                    node_to_peer.setdefault(nid, PeerInfo(peer_id, (endpoint,)))
+                    if nid not in node_to_peer:  # setdefault would construct PeerInfo per entry
+                        node_to_peer[nid] = PeerInfo(peer_id, (endpoint,))
                 output[q] = (list(nearest.keys()), search_results[q].finished)
             return output
 

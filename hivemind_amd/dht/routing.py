@@ -63,10 +63,24 @@ class DHTID(int):
             return b
         return int.to_bytes(self, length, byteorder, signed=signed)
 
+    # the same ~swarm-size node ids get re-parsed from every FindResponse:
+    # intern by raw bytes so a repeat costs one dict lookup instead of an int
+    # parse + object alloc (cap bounds memory in long-running processes)
+    _INTERN: dict = {}
+    _INTERN_CAP = 65536
+
     @classmethod
     def from_bytes(cls, raw: bytes, byteorder: str = "big", *, signed: bool = False) -> "DHTID":
-        # hot path (one call per peer entry in every FindResponse): a 32-byte
-        # input is in range by construction, so bypass __new__'s range assert
+        if byteorder == "big" and not signed and cls is DHTID:
+            v = cls._INTERN.get(raw)
+            if v is None:
+                if len(cls._INTERN) >= cls._INTERN_CAP:
+                    cls._INTERN.clear()
+                # a 32-byte input is in range by construction: bypass __new__'s assert
+                v = cls._INTERN[raw] = int.__new__(cls, int.from_bytes(raw, "big"))
+                if len(raw) == 32:
+                    v._b32 = raw
+            return v
         return int.__new__(cls, int.from_bytes(raw, byteorder, signed=signed))
 
     def __repr__(self):

@@ -126,20 +126,22 @@ async def traverse_dht(
                 # pick the (query, candidate) pair with the smallest distance among active queries
                 best: Optional[Tuple[int, DHTID, DHTID]] = None  # (distance, query, peer)
                 idle_wait = False
+                _heappop = heapq.heappop
                 for q in queries:
                     if q in finished:
                         continue
-                    while candidates[q] and (
-                        candidates[q][ROOT][0] > upper_bound(q) or candidates[q][ROOT][1] in queried[q]
-                    ):
-                        heapq.heappop(candidates[q])
+                    cand_q = candidates[q]
+                    ub = upper_bound(q)  # constant within this pass (nearest[q] unchanged)
+                    queried_q = queried[q]
+                    while cand_q and (cand_q[ROOT][0] > ub or cand_q[ROOT][1] in queried_q):
+                        _heappop(cand_q)
                     maybe_finish(q)
                     if q in finished:
                         continue
-                    if not candidates[q]:
+                    if not cand_q:
                         idle_wait = True  # in-flight elsewhere may refill this query
                         continue
-                    d, uid = candidates[q][ROOT]
+                    d, uid = cand_q[ROOT]
                     if best is None or d < best[0]:
                         best = (d, q, uid)
                 if best is None:
