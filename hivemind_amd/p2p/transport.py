@@ -208,7 +208,11 @@ class P2P:
         self._balanced: Dict[str, List[Tuple[HandlerType, bool, bool]]] = {}
         self._balanced_rr: Dict[str, int] = {}
         self._connections: Dict[PeerID, _Connection] = {}
-        self._conn_lock = asyncio.Lock()
+        # per-peer dial locks: dedup concurrent dials to the SAME peer without
+        # serializing dials to different peers (a global lock here made every
+        # first-contact batch pay ~#peers x (connect + handshake) sequentially
+        # -- the dominant latency at the 1024-peer DHT benchmark config)
+        self._dial_locks: dict = {}  # peer_id -> [asyncio.Lock, refcount]
         self._endpoint_book: Dict[PeerID, str] = {}  # last known endpoint per peer
         self._alive = True
         self._listen = True
@@ -617,10 +621,23 @@ class P2P:
         # expand "unix:/path,host:port" advertisements into individual dials
         endpoints = [part for ep in endpoints for part in ep.split(",") if part]
         last_exc: Optional[Exception] = None
-        async with self._conn_lock:
-            conn = self._connections.get(peer_id)
-            if conn is not None and not conn.closed.is_set():
-                return conn
+        entry = self._dial_locks.get(peer_id)
+        if entry is None:
+            entry = self._dial_locks[peer_id] = [asyncio.Lock(), 0]
+        entry[1] += 1
+        try:
+            async with entry[0]:
+                return await self._connect_locked(peer_id, endpoints, last_exc)
+        finally:
+            entry[1] -= 1
+            if entry[1] == 0:
+                self._dial_locks.pop(peer_id, None)
+
+    async def _connect_locked(self, peer_id, endpoints, last_exc) -> "_Connection":
+        conn = self._connections.get(peer_id)
+        if conn is not None and not conn.closed.is_set():
+            return conn
+        if endpoints:
             for ep in endpoints:
                 try:
                     if ep.startswith(RELAY_SCHEME):
