@@ -620,23 +620,23 @@ class P2P:
             endpoints.append(self._endpoint_book[peer_id])
         # expand "unix:/path,host:port" advertisements into individual dials
         endpoints = [part for ep in endpoints for part in ep.split(",") if part]
-        last_exc: Optional[Exception] = None
         entry = self._dial_locks.get(peer_id)
         if entry is None:
             entry = self._dial_locks[peer_id] = [asyncio.Lock(), 0]
         entry[1] += 1
         try:
             async with entry[0]:
-                return await self._connect_locked(peer_id, endpoints, last_exc)
+                return await self._connect_locked(peer_id, endpoints)
         finally:
             entry[1] -= 1
             if entry[1] == 0:
                 self._dial_locks.pop(peer_id, None)
 
-    async def _connect_locked(self, peer_id, endpoints, last_exc) -> "_Connection":
+    async def _connect_locked(self, peer_id, endpoints) -> "_Connection":
         conn = self._connections.get(peer_id)
         if conn is not None and not conn.closed.is_set():
             return conn
+        last_exc: Optional[Exception] = None
         if endpoints:
             for ep in endpoints:
                 try:
