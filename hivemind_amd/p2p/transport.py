@@ -213,6 +213,11 @@ class P2P:
         # first-contact batch pay ~#peers x (connect + handshake) sequentially
         # -- the dominant latency at the 1024-peer DHT benchmark config)
         self._dial_locks: dict = {}  # peer_id -> [asyncio.Lock, refcount]
+        # cap concurrent outbound dials: full per-peer parallelism for normal
+        # fan-out, but no thundering herd against remote accept queues
+        self._dial_semaphore = asyncio.Semaphore(
+            int(os.environ.get("HIVEMIND_MAX_CONCURRENT_DIALS", "64"))
+        )
         self._endpoint_book: Dict[PeerID, str] = {}  # last known endpoint per peer
         self._alive = True
         self._listen = True
@@ -249,7 +254,12 @@ class P2P:
         self._listen_host = listen_host
         self._listen = listen
         if listen:
-            self._server = await asyncio.start_server(self._on_accept, listen_host, port, limit=STREAM_BUFFER_LIMIT)
+            # backlog: asyncio's default of 100 drops SYNs under the dial
+            # bursts concurrent per-peer dialing produces in large swarms;
+            # dropped SYNs retransmit after 1-3 s and poison handshake latency
+            self._server = await asyncio.start_server(
+                self._on_accept, listen_host, port, limit=STREAM_BUFFER_LIMIT, backlog=128
+            )
             self._port = self._server.sockets[0].getsockname()[1]
             # any listening peer can serve as a circuit relay for NATed peers
             self.add_unary_handler(RELAY_REGISTER_HANDLER, self._rpc_relay_register)
@@ -265,7 +275,7 @@ class P2P:
                     tempfile.gettempdir(), f"hivemind_p2p_{os.getpid()}_{self._port}.sock"
                 )
                 self._uds_server = await asyncio.start_unix_server(
-                    self._on_accept, self._uds_path, limit=STREAM_BUFFER_LIMIT
+                    self._on_accept, self._uds_path, limit=STREAM_BUFFER_LIMIT, backlog=128
                 )
             except Exception as e:
                 logger.debug(f"no unix-domain listener: {e!r}")
@@ -637,7 +647,7 @@ class P2P:
         if conn is not None and not conn.closed.is_set():
             return conn
         last_exc: Optional[Exception] = None
-        if endpoints:
+        async with self._dial_semaphore:
             for ep in endpoints:
                 try:
                     if ep.startswith(RELAY_SCHEME):
