@@ -59,6 +59,20 @@ def _host_main(rank: int, n_local: int, seed_endpoints, conn, wait_timeout: floa
                 conn.send((dt, sum(info is not None for info in infos), len(uids)))
             elif cmd[0] == "alive":
                 conn.send(sum(p.is_alive for p in peers))
+            elif cmd[0] == "stats":
+                async def _stats(_dht, node):
+                    out = dict(node.protocol.rpc_stats)
+                    for k, v in node.p2p.transport_stats.items():
+                        out[k] = v
+                    return out
+                agg = {}
+                for p in peers:
+                    try:
+                        for k, v in p.run_coroutine(_stats).items():
+                            agg[k] = max(agg.get(k, 0), v) if k.endswith("_max_s") else agg.get(k, 0) + v
+                    except Exception:
+                        pass
+                conn.send(agg)
     finally:
         for p in peers:
             try:
@@ -149,6 +163,16 @@ def main():
     for c in conns:
         c.send(("alive",))
         alive += c.recv()
+
+    # aggregate transport/RPC observability across all peers
+    stats = {}
+    for c in conns:
+        c.send(("stats",))
+        for k, v in c.recv().items():
+            stats[k] = max(stats.get(k, 0), v) if k.endswith("_max_s") else stats.get(k, 0) + v
+    if stats:
+        pretty = {k: (round(v, 2) if isinstance(v, float) else v) for k, v in sorted(stats.items())}
+        print(f"transport/rpc stats: {pretty}", file=sys.stderr, flush=True)
 
     result = {
         "metric": "DHT store/get latency",

@@ -10,6 +10,7 @@ nodes from the routing table; every request updates the routing table.
 from __future__ import annotations
 
 import asyncio
+import time
 from dataclasses import dataclass, field
 from typing import Collection, Dict, List, Optional, Sequence, Tuple, Union
 
@@ -109,6 +110,7 @@ class DHTProtocol(ServicerBase):
         self.wait_timeout = 5.0
         self.record_validator: Optional[RecordValidatorBase] = None
         self.client_mode = False
+        self.rpc_stats: dict = {}
 
     @classmethod
     async def create(
@@ -126,6 +128,8 @@ class DHTProtocol(ServicerBase):
         self = cls()
         self.p2p, self.node_id = p2p, node_id
         self.bucket_size, self.num_replicas, self.wait_timeout = bucket_size, num_replicas, wait_timeout
+        # per-method client-side in-flight time sums/counts (observability)
+        self.rpc_stats: dict = {}
         self.routing_table = RoutingTable(node_id, bucket_size, depth_modulo)
         self.cache = DHTLocalStorage(maxsize=cache_size)
         self.record_validator = record_validator
@@ -170,7 +174,13 @@ class DHTProtocol(ServicerBase):
         try:
             request = PingRequest(peer=self._my_info(), validate=validate)
             time_requested = get_dht_time()
+            _t0 = time.monotonic()
             raw = await self._get_stub(peer).rpc_ping(request, timeout=self.wait_timeout)
+            _dt = time.monotonic() - _t0
+            st = self.rpc_stats
+            st["rpc_ping_n"] = st.get("rpc_ping_n", 0) + 1
+            st["rpc_ping_s"] = st.get("rpc_ping_s", 0.0) + _dt
+            st["rpc_ping_max_s"] = max(st.get("rpc_ping_max_s", 0.0), _dt)
             response = PingResponse.loads(raw)
             time_responded = get_dht_time()
             if validate:
@@ -250,7 +260,13 @@ class DHTProtocol(ServicerBase):
                 in_cache=list(in_cache),
                 peer=self._my_info(),
             )
+            _t0 = time.monotonic()
             raw = await self._get_stub(peer).rpc_store(request, timeout=self.wait_timeout)
+            _dt = time.monotonic() - _t0
+            st = self.rpc_stats
+            st["rpc_store_n"] = st.get("rpc_store_n", 0) + 1
+            st["rpc_store_s"] = st.get("rpc_store_s", 0.0) + _dt
+            st["rpc_store_max_s"] = max(st.get("rpc_store_max_s", 0.0), _dt)
             response = StoreResponse.loads(raw)
             if response.peer is not None and response.peer.node_id:
                 peer_id = peer.peer_id if hasattr(peer, "peer_id") else peer
@@ -311,7 +327,13 @@ class DHTProtocol(ServicerBase):
         keys = list(keys)
         try:
             request = FindRequest(keys=[k.to_bytes() for k in keys], peer=self._my_info())
+            _t0 = time.monotonic()
             raw = await self._get_stub(peer).rpc_find(request, timeout=self.wait_timeout)
+            _dt = time.monotonic() - _t0
+            st = self.rpc_stats
+            st["rpc_find_n"] = st.get("rpc_find_n", 0) + 1
+            st["rpc_find_s"] = st.get("rpc_find_s", 0.0) + _dt
+            st["rpc_find_max_s"] = max(st.get("rpc_find_max_s", 0.0), _dt)
             response = FindResponse.loads(raw)
             if response.peer is not None and response.peer.node_id:
                 peer_id = peer.peer_id if hasattr(peer, "peer_id") else peer

@@ -218,6 +218,9 @@ class P2P:
         self._dial_semaphore = asyncio.Semaphore(
             int(os.environ.get("HIVEMIND_MAX_CONCURRENT_DIALS", "64"))
         )
+        # lightweight transport observability (always on; ~ns per event):
+        # counters + wall-time sums keyed by event name, read via stats()
+        self.transport_stats: Dict[str, float] = {}
         self._endpoint_book: Dict[PeerID, str] = {}  # last known endpoint per peer
         self._alive = True
         self._listen = True
@@ -605,6 +608,8 @@ class P2P:
             key=lambda c: c.last_used,
         )
         excess = len(self._connections) - self._max_connections
+        st = self.transport_stats
+        st["evictions"] = st.get("evictions", 0) + min(excess, len(evictable))
         for conn in evictable[:excess]:
             peer = conn.remote_id
             logger.debug(f"connection manager: evicting idle connection to {peer}")
@@ -647,7 +652,13 @@ class P2P:
         if conn is not None and not conn.closed.is_set():
             return conn
         last_exc: Optional[Exception] = None
+        import time as _time
+
+        _t_sem = _time.monotonic()
         async with self._dial_semaphore:
+            _t_dial = _time.monotonic()
+            st = self.transport_stats
+            st["dial_sem_wait_s"] = st.get("dial_sem_wait_s", 0.0) + (_t_dial - _t_sem)
             for ep in endpoints:
                 try:
                     if ep.startswith(RELAY_SCHEME):
@@ -671,8 +682,12 @@ class P2P:
                     self._endpoint_book[conn.remote_id] = ep
                     self._register_connection(conn)
                     conn.reader_task = asyncio.create_task(self._connection_loop(conn))
+                    st["dials_ok"] = st.get("dials_ok", 0) + 1
+                    st["dial_time_s"] = st.get("dial_time_s", 0.0) + (_time.monotonic() - _t_dial)
+                    st["dial_time_max_s"] = max(st.get("dial_time_max_s", 0.0), _time.monotonic() - _t_dial)
                     return conn
                 except Exception as e:
+                    st["dials_failed"] = st.get("dials_failed", 0) + 1
                     last_exc = e
                     if conn is not None:  # failed dial/handshake: release the fd
                         try:
