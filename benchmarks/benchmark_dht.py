@@ -147,6 +147,16 @@ def main():
     store_phase_s = time.perf_counter() - t_phase
     print(f"store phase: {store_phase_s:.1f}s", file=sys.stderr, flush=True)
 
+    def _collect_stats():
+        agg = {}
+        for c in conns:
+            c.send(("stats",))
+            for k, v in c.recv().items():
+                agg[k] = max(agg.get(k, 0), v) if k.endswith("_max_s") else agg.get(k, 0) + v
+        return agg
+
+    stats_after_store = _collect_stats()
+
     get_times, get_successes, total_gets = [], 0, 0
     t_phase = time.perf_counter()
     for start in range(0, args.num_experts, args.expert_batch_size):
@@ -164,15 +174,17 @@ def main():
         c.send(("alive",))
         alive += c.recv()
 
-    # aggregate transport/RPC observability across all peers
-    stats = {}
-    for c in conns:
-        c.send(("stats",))
-        for k, v in c.recv().items():
-            stats[k] = max(stats.get(k, 0), v) if k.endswith("_max_s") else stats.get(k, 0) + v
-    if stats:
-        pretty = {k: (round(v, 2) if isinstance(v, float) else v) for k, v in sorted(stats.items())}
-        print(f"transport/rpc stats: {pretty}", file=sys.stderr, flush=True)
+    # aggregate transport/RPC observability, split by phase
+    stats_total = _collect_stats()
+
+    def _pretty(d):
+        return {k: (round(v, 2) if isinstance(v, float) else v) for k, v in sorted(d.items())}
+
+    get_delta = {
+        k: (v if k.endswith("_max_s") else v - stats_after_store.get(k, 0)) for k, v in stats_total.items()
+    }
+    print(f"store-phase stats: {_pretty(stats_after_store)}", file=sys.stderr, flush=True)
+    print(f"get-phase stats: {_pretty(get_delta)}", file=sys.stderr, flush=True)
 
     result = {
         "metric": "DHT store/get latency",
