@@ -407,3 +407,37 @@ def test_dhtid_interning_and_bytes_cache():
     little = DHTID.from_bytes(raw[::-1], byteorder="little")
     assert little == a
     assert DHTID.from_bytes((2**255).to_bytes(32, "big")) == 2**255
+
+
+def test_concurrent_dials_same_peer_share_one_connection():
+    """Per-peer dial locks must dedup concurrent dials to one peer (one
+    physical connection, one handshake) while not serializing dials to
+    DIFFERENT peers (the old global dial lock did, and made large-swarm
+    first-contact batches pay #peers x handshake sequentially)."""
+    async def main():
+        servers = [await P2P.create() for _ in range(6)]
+        client = await P2P.create()
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return payload
+
+        for s in servers:
+            s.add_unary_handler("echo", echo)
+
+        # burst of calls to the SAME fresh peer: exactly one dial
+        await asyncio.gather(*(client.call_unary(servers[0].peer_info, "echo", b"x", timeout=10) for _ in range(8)))
+        assert client.transport_stats.get("dials_ok", 0) == 1
+
+        # burst across DIFFERENT fresh peers: one dial each, all succeed
+        await asyncio.gather(
+            *(client.call_unary(s.peer_info, "echo", b"y", timeout=10) for s in servers[1:] for _ in range(3))
+        )
+        assert client.transport_stats.get("dials_ok", 0) == len(servers)
+        assert client.transport_stats.get("dials_failed", 0) == 0
+        assert client.transport_stats.get("dial_time_s", 0) > 0
+
+        await client.shutdown()
+        for s in servers:
+            await s.shutdown()
+
+    run(main())
