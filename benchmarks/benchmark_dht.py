@@ -26,6 +26,18 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 def _host_main(rank: int, n_local: int, seed_endpoints, conn, wait_timeout: float):
     """Worker process: hosts `n_local` DHT peers and serves store/get commands."""
     random.seed(1000 + rank)
+    import os as _os
+    import resource as _resource
+
+    # hosting n_local peers in ONE process shares its fd budget among them:
+    # derive a per-peer connection cap from RLIMIT_NOFILE (each connection
+    # costs ~1 fd at each endpoint; /2 margin for sockets mid-handshake,
+    # listeners, pipes). Production nodes run one peer per host and use the
+    # transport's libp2p-parity default instead.
+    if "HIVEMIND_MAX_CONNECTIONS" not in _os.environ:
+        soft, _hard = _resource.getrlimit(_resource.RLIMIT_NOFILE)
+        per_peer = max(64, (soft // max(n_local, 1) - 8) // 2)
+        _os.environ["HIVEMIND_MAX_CONNECTIONS"] = str(min(896, per_peer))
     from hivemind_amd import DHT
     from hivemind_amd.moe.server.dht_handler import declare_experts, get_expert_infos
     from hivemind_amd.utils.timed_storage import get_dht_time
