@@ -34,10 +34,11 @@ def _host_main(rank: int, n_local: int, seed_endpoints, conn, wait_timeout: floa
     # costs ~1 fd at each endpoint; /2 margin for sockets mid-handshake,
     # listeners, pipes). Production nodes run one peer per host and use the
     # transport's libp2p-parity default instead.
-    if "HIVEMIND_MAX_CONNECTIONS" not in _os.environ:
-        soft, _hard = _resource.getrlimit(_resource.RLIMIT_NOFILE)
-        per_peer = max(64, (soft // max(n_local, 1) - 8) // 2)
-        _os.environ["HIVEMIND_MAX_CONNECTIONS"] = str(min(896, per_peer))
+    # a short idle sweep keeps the shared fd budget bounded by the ACTIVE
+    # working set (one-shot connections close after 15 s) while the full
+    # libp2p-parity cap avoids LRU churn during a peer's client bursts
+    _os.environ.setdefault("HIVEMIND_IDLE_CONN_TIMEOUT", "15")
+    _ = _resource.getrlimit(_resource.RLIMIT_NOFILE)  # raised by --increase_file_limit
     from hivemind_amd import DHT
     from hivemind_amd.moe.server.dht_handler import declare_experts, get_expert_infos
     from hivemind_amd.utils.timed_storage import get_dht_time

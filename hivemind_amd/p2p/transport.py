@@ -289,7 +289,36 @@ class P2P:
                 self._uds_path, self._uds_server = None, None
         if relay_endpoint is not None:
             await self.register_with_relay(relay_endpoint)
+        # libp2p-connmgr-style grace sweep: connections idle beyond the
+        # timeout are closed even below the cap, so long-lived processes
+        # (and many-peers-per-process benchmark hosts) do not accumulate
+        # one-shot connections toward the fd ceiling. The LRU cap handles
+        # bursts; the sweeper handles slow accumulation.
+        self._idle_timeout = float(os.environ.get("HIVEMIND_IDLE_CONN_TIMEOUT", "120"))
+        self._sweeper_task = (
+            asyncio.get_event_loop().create_task(self._idle_sweeper()) if self._idle_timeout > 0 else None
+        )
         return self
+
+    async def _idle_sweeper(self):
+        import time as _time
+
+        interval = min(15.0, max(1.0, self._idle_timeout / 4))
+        while self._alive:
+            await asyncio.sleep(interval)
+            now = _time.monotonic()
+            stale = [
+                c
+                for c in self._connections.values()
+                if not c.is_busy() and not c.closed.is_set() and now - c.last_used > self._idle_timeout
+            ]
+            for conn in stale:
+                self._connections.pop(conn.remote_id, None)
+                self.transport_stats["idle_closed"] = self.transport_stats.get("idle_closed", 0) + 1
+                try:
+                    await conn.close()
+                except Exception:
+                    pass
 
     @property
     def tcp_endpoint(self) -> str:
@@ -914,6 +943,8 @@ class P2P:
 
     async def shutdown(self):
         self._alive = False
+        if getattr(self, "_sweeper_task", None) is not None:
+            self._sweeper_task.cancel()
         for task in list(self._relay_splices):
             task.cancel()
         self._relay_splices.clear()
