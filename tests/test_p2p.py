@@ -441,3 +441,37 @@ def test_concurrent_dials_same_peer_share_one_connection():
             await s.shutdown()
 
     run(main())
+
+
+def test_idle_connection_sweeper():
+    """Connections idle beyond HIVEMIND_IDLE_CONN_TIMEOUT close even below the
+    LRU cap (libp2p connmgr grace sweep): long-lived processes must not
+    accumulate one-shot connections toward the fd ceiling."""
+    import os
+
+    async def main():
+        os.environ["HIVEMIND_IDLE_CONN_TIMEOUT"] = "1"
+        try:
+            server = await P2P.create()
+            client = await P2P.create()
+        finally:
+            del os.environ["HIVEMIND_IDLE_CONN_TIMEOUT"]
+
+        async def echo(payload: bytes, ctx) -> bytes:
+            return payload
+
+        server.add_unary_handler("echo", echo)
+        assert await client.call_unary(server.peer_info, "echo", b"x", timeout=5) == b"x"
+        assert server.peer_id in client._connections
+        for _ in range(40):
+            await asyncio.sleep(0.1)
+            if server.peer_id not in client._connections:
+                break
+        assert server.peer_id not in client._connections, "idle connection not swept"
+        assert client.transport_stats.get("idle_closed", 0) >= 1
+        # the path still works after the sweep (transparent re-dial)
+        assert await client.call_unary(server.peer_info, "echo", b"y", timeout=5) == b"y"
+        await client.shutdown()
+        await server.shutdown()
+
+    run(main())
